@@ -1,0 +1,325 @@
+#!/usr/bin/env python3
+"""Benchmark: MatrixOne ANN vector-index search hot path on MI355X.
+
+Workload (BASELINE.json configs[1], the metric's single-GPU operating point):
+  IVF-Flat L2, 10M x 768 f32, nlist=4096, nprobe=32, batch=1024, k=10,
+  recall@10 >= 0.95 verified against exact ground truth.
+
+One "step" = one batch of 1024 queries through the full hot path
+(centroid rank -> list scan -> top-k -> id gather), queries HBM-resident
+when the timed region starts (device-pointer search).
+
+Scaling (--gpus N via torch.distributed.run): WEAK — each rank owns its own
+10M-row shard (lists sharded, SHARDED mode per cgo/cuvs/index_base.hpp:104),
+every rank serves the same query batch, one RCCL all-gather of [B,k]
+(id,dist) blocks over xGMI + GPU merge per step (replaces the reference's
+CPU cpu_topk_merge_sharded, helper.h:137-167). value = queries/s of the
+whole job over the N x 10M corpus.
+
+Emits ONE JSON line from rank 0 including `roofline` (HIP-event time +
+algorithmic bytes of the list-scan kernel vs 8 TB/s HBM peak) and
+`cpu_baseline` (the oracle CPU port of the reference probe, OpenMP, timed on
+a bounded sample of the same workload).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK_GBS = 8000.0  # MI355X_MICROARCH.md: HBM3E 8.0 TB/s spec peak
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def kmeans_torch(data_gpu: torch.Tensor, nlist: int, iters: int, seed: int):
+    """Harness-side k-means (index BUILD — out of the judged search path,
+    SURVEY.md §2 kmeans row). f32 matmul Lloyd on the GPU."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    n = data_gpu.shape[0]
+    sample = data_gpu[torch.randperm(n, generator=g)[:min(n, 262144)].to(data_gpu.device)]
+    idx = torch.randperm(sample.shape[0], generator=g)[:nlist].to(data_gpu.device)
+    cent = sample[idx].clone()
+    for _ in range(iters):
+        # assignment by expanded distance (build-time only)
+        d2 = (sample * sample).sum(1, keepdim=True) \
+            - 2.0 * (sample @ cent.T) + (cent * cent).sum(1)[None, :]
+        a = d2.argmin(1)
+        cent_new = torch.zeros_like(cent)
+        counts = torch.zeros(nlist, device=cent.device, dtype=torch.float32)
+        cent_new.index_add_(0, a, sample)
+        counts.index_add_(0, a, torch.ones_like(a, dtype=torch.float32))
+        empty = counts == 0
+        counts = counts.clamp(min=1.0)
+        cent = cent_new / counts[:, None]
+        if empty.any():
+            # reseed empty lists from random sample rows
+            ridx = torch.randint(0, sample.shape[0], (int(empty.sum()),),
+                                 device=cent.device)
+            cent[empty] = sample[ridx]
+    return cent.contiguous()
+
+
+def assign_torch(data_gpu: torch.Tensor, cent: torch.Tensor,
+                 chunk: int = 131072) -> torch.Tensor:
+    """Nearest-centroid assignment, chunked f32 matmul (build-time)."""
+    n = data_gpu.shape[0]
+    out = torch.empty(n, dtype=torch.int32, device=data_gpu.device)
+    cnorm = (cent * cent).sum(1)
+    for c0 in range(0, n, chunk):
+        blk = data_gpu[c0:c0 + chunk]
+        d2 = (blk * blk).sum(1, keepdim=True) - 2.0 * (blk @ cent.T) + cnorm[None, :]
+        out[c0:c0 + chunk] = d2.argmin(1).to(torch.int32)
+    return out
+
+
+def exact_gt(data_gpu: torch.Tensor, queries: torch.Tensor, k: int,
+             chunk: int = 1 << 20):
+    """Exact L2 top-k ground truth (recall measurement only)."""
+    nq = queries.shape[0]
+    qnorm = (queries * queries).sum(1, keepdim=True)
+    best_d = torch.full((nq, k), float("inf"), device=data_gpu.device)
+    best_i = torch.full((nq, k), -1, dtype=torch.int64, device=data_gpu.device)
+    for c0 in range(0, data_gpu.shape[0], chunk):
+        blk = data_gpu[c0:c0 + chunk]
+        d2 = qnorm - 2.0 * (queries @ blk.T) + (blk * blk).sum(1)[None, :]
+        d, i = torch.topk(d2, min(k, blk.shape[0]), dim=1, largest=False)
+        cat_d = torch.cat([best_d, d], 1)
+        cat_i = torch.cat([best_i, i + c0], 1)
+        d, pos = torch.topk(cat_d, k, dim=1, largest=False)
+        best_d = d
+        best_i = torch.gather(cat_i, 1, pos)
+    return best_i
+
+
+def run_cpu_baseline(host_vecs: np.ndarray, cent: np.ndarray,
+                     assign: np.ndarray, queries: np.ndarray, nprobe: int,
+                     k: int, sample_q: int):
+    """The oracle (CPU port of the reference probe: rankCentroids + list scan
+    + top-k, OpenMP over queries) on a bounded sample of the same workload."""
+    from oracle import oracle as orc
+    idx = orc.IvfIndex(cent, host_vecs, assign.astype(np.int64))
+    qs = queries[:sample_q]
+    idx.search(orc.METRIC_L2SQ, qs[:2], nprobe, k)  # warm
+    t0 = time.perf_counter()
+    idx.search(orc.METRIC_L2SQ, qs, nprobe, k)
+    dt = time.perf_counter() - t0
+    cores = orc.lib().moann_oracle_num_threads()
+    return {
+        "value": round(sample_q / dt, 3),
+        "unit": "queries/s",
+        "cores": int(cores),
+        "kind": "port",
+        "sample": f"{sample_q} queries of the same 10M-shard probe "
+                  f"(nprobe={nprobe}, k={k}), {dt:.1f}s wall",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--rows", type=int, default=10_000_000,
+                    help="rows PER GPU (weak scaling)")
+    ap.add_argument("--dim", type=int, default=768)
+    ap.add_argument("--nlist", type=int, default=4096)
+    ap.add_argument("--nprobe", type=int, default=32)
+    ap.add_argument("--batch", type=int, default=1024)
+    ap.add_argument("--k", type=int, default=10)
+    ap.add_argument("--recall-queries", type=int, default=256)
+    ap.add_argument("--cpu-baseline-queries", type=int, default=48)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+    if distributed:
+        import torch.distributed as dist
+        dist.init_process_group("nccl")
+    torch.cuda.set_device(local_rank)
+    dev = torch.device("cuda", local_rank)
+
+    from matrixone_amd import engine
+    from matrixone_amd.distributed import merge_topk_torch
+
+    # ---- data + build (per-rank shard; seed differs per rank) -------------
+    t_build0 = time.perf_counter()
+    torch.manual_seed(42 + rank)
+    log(f"[bench] generating {args.rows}x{args.dim} f32 shard on GPU ...")
+    data = torch.randn(args.rows, args.dim, device=dev, dtype=torch.float32)
+    log("[bench] k-means (harness-side build) ...")
+    cent = kmeans_torch(data, args.nlist, iters=8, seed=123 + rank)
+    assign = assign_torch(data, cent)
+    # queries: held-out, same on every rank
+    qg = torch.Generator(device="cpu").manual_seed(4242)
+    queries_host = torch.randn(10000, args.dim, generator=qg,
+                               dtype=torch.float32)
+    queries = queries_host.to(dev)
+
+    log("[bench] ingest + pack into moann index ...")
+    host_vecs = data.cpu().numpy()  # host copy (oracle baseline + ingest)
+    cent_np = cent.cpu().numpy()
+    assign_np = assign.cpu().numpy()
+    ix = engine.IvfFlatIndex(args.dim, args.nlist, metric="l2sq",
+                             capacity=args.rows, device=local_rank)
+    CH = 1 << 21
+    for c0 in range(0, args.rows, CH):
+        ix.add(host_vecs[c0:c0 + CH])
+    ix.set_centroids(cent_np)
+    ix.set_assignments(assign_np)
+    ix.build()
+    log(f"[bench] build done in {time.perf_counter() - t_build0:.1f}s")
+
+    # ---- recall check ------------------------------------------------------
+    nrq = args.recall_queries
+    gt_local = exact_gt(data, queries[:nrq], args.k)  # local-shard GT
+    ids_r, dists_r = ix.search_device(queries[:nrq].contiguous(), args.k,
+                                      args.nprobe)
+    if not distributed:
+        gt_ids = gt_local.cpu().numpy()
+        recall = float(np.mean([
+            len(set(ids_r[i]).intersection(gt_ids[i])) / args.k
+            for i in range(nrq)]))
+    else:
+        import torch.distributed as dist
+        # merge IVF results across shards (ids are shard-local rows; offset)
+        off = rank * args.rows
+        ids_off = np.where(ids_r >= 0, ids_r + off, -1)
+        from matrixone_amd.distributed import allgather_merge
+        m_ids, _ = allgather_merge(ids_off, dists_r, args.k, device=dev)
+        # global GT: merge per-shard exact GT by distance
+        gt_dist = gt_exact_dists(data, queries[:nrq], gt_local)
+        g_ids, _ = allgather_merge(
+            np.where(gt_local.cpu().numpy() >= 0,
+                     gt_local.cpu().numpy() + off, -1),
+            gt_dist.cpu().numpy().astype(np.float32), args.k, device=dev)
+        recall = float(np.mean([
+            len(set(m_ids[i]).intersection(g_ids[i])) / args.k
+            for i in range(nrq)]))
+    log(f"[bench] recall@{args.k} = {recall:.4f} (target >= 0.95)")
+
+    # ---- timed region ------------------------------------------------------
+    nbatches = max(1, queries.shape[0] // args.batch)
+
+    def step(i):
+        b0 = (i % nbatches) * args.batch
+        q = queries[b0:b0 + args.batch].contiguous()
+        ids, dists = ix.search_device(q, args.k, args.nprobe)
+        if distributed:
+            from matrixone_amd.distributed import allgather_merge
+            off = rank * args.rows
+            ids = np.where(ids >= 0, ids + off, -1)
+            ids, dists = allgather_merge(ids, dists, args.k, device=dev)
+        return ids
+
+    for i in range(args.warmup):
+        step(i)
+    ix.perf_reset()
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    qps = args.steps * args.batch / elapsed
+    perf = ix.perf()
+
+    # ---- roofline (list-scan kernel, HIP events on the index stream) ------
+    scan_gbs = (perf["scan_bytes"] / 1e9) / max(perf["scan_ms"] / 1e3, 1e-9)
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(scan_gbs, 1),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(scan_gbs / HBM_PEAK_GBS, 4),
+        "traffic": None,  # PMC bytes come from the committed rocprof runs
+    }
+
+    # ---- CPU baseline (rank 0, N=1 only) -----------------------------------
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
+        log("[bench] CPU baseline (oracle port, OpenMP) ...")
+        try:
+            cpu_baseline = run_cpu_baseline(
+                host_vecs, cent_np, assign_np,
+                queries_host.numpy(), args.nprobe, args.k,
+                args.cpu_baseline_queries)
+        except Exception as e:  # report the miss, never fake a number
+            log(f"[bench] cpu_baseline failed: {e}")
+
+    if rank == 0:
+        line = {
+            "metric": "ANN queries/sec @ recall@10>=0.95, 10Mx768 f32",
+            "value": round(qps, 1),
+            "unit": "queries/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f32",
+            "data": "synthetic",
+            "config": {
+                "workload": f"ivfflat-l2-{args.rows}x{args.dim}-"
+                            f"nlist{args.nlist}-nprobe{args.nprobe}-"
+                            f"batch{args.batch}-k{args.k}",
+                "rows_per_gpu": args.rows,
+                "dim": args.dim,
+                "nlist": args.nlist,
+                "nprobe": args.nprobe,
+                "global_batch": args.batch,
+                "k": args.k,
+                "recall_at_k": round(recall, 4),
+                "parallelism": f"sharded-lists x{world}" if world > 1
+                               else "single",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line), flush=True)
+
+    ix.close()
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+def gt_exact_dists(data, queries, gt_ids):
+    """Exact squared L2 for each (query, gt id) pair."""
+    rows = data[gt_ids.reshape(-1)]
+    q = queries.repeat_interleave(gt_ids.shape[1], 0)
+    return ((rows - q) ** 2).sum(1).reshape(gt_ids.shape)
+
+
+exact_gt_dists = gt_exact_dists  # legacy alias
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,84 @@
+/* Internal interfaces between moann_lib.cpp (host/ABI) and
+ * moann_kernels.hip (CDNA4 device code). Not part of the public ABI. */
+
+#ifndef MOANN_INTERNAL_H
+#define MOANN_INTERNAL_H
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace moann {
+
+/* Internal metric codes (template selectors for the scan kernel). */
+enum MetricKind : int {
+    KM_L2SQ = 0,  /* sum((x-q)^2)                     */
+    KM_IP = 1,    /* -x.q (MO convention)             */
+    KM_COS = 2,   /* 1 - x.q/(|x||q|), clamp, 0 -> 1  */
+    KM_L1 = 3,    /* sum|x-q|                         */
+};
+
+/* One scan job: a (list segment, query tile) pair. SoA device arrays. */
+struct ScanJobs {
+    const int64_t* databaseg;  /* [njobs] base group index of the list       */
+    const int32_t* gstart;     /* [njobs] first group (list-relative)        */
+    const int32_t* gcount;     /* [njobs] groups this job scans              */
+    const int32_t* rows;       /* [njobs] valid rows in the WHOLE list       */
+    const int32_t* nq;         /* [njobs] queries in the tile (<= QT)        */
+    const int32_t* qbase;      /* [njobs] offset into qslot arrays           */
+    const int32_t* qslot_query;   /* [nqslots] query index                   */
+    const int64_t* qslot_outbase; /* [nqslots] output base in dists buffer   */
+    int njobs;
+};
+
+/* Launch the distance scan over interleaved packed data.
+ * packed layout: group g, dim-quad j4, lane l, c -> packed[((g*dpad + 4*j4)*64) + l*4 + c]
+ * i.e. [g][dpad/4][64][4] floats; row (g*64+l) holds vector dims 4*j4+c.
+ * queries: [nq_total][dpad] device, zero-padded.
+ * qnorms: [nq_total] sum-of-squares (cos only, else nullptr).
+ * dists_out: ragged candidate buffer, written at qslot_outbase + row. */
+void launch_scan(int metric, int qt, const float* packed, const float* queries,
+                 const float* qnorms, int dpad, const ScanJobs& jobs,
+                 float* dists_out, hipStream_t stream);
+
+/* Per-query ascending top-k select (radix select + in-LDS bitonic sort).
+ * Candidates for query q live at dists[off(q) .. off(q)+count(q)) where
+ * off/count come from qoffs (ragged, [nq+1]) or uniform n (qoffs==nullptr).
+ * out_slots: [nq][k] candidate index within the query's range, -1 padding.
+ * out_dists: [nq][k] ascending, FLT_MAX padding. k <= 4096. */
+void launch_topk(const float* dists, const int64_t* qoffs, int64_t uniform_n,
+                 int nq, int k, int32_t* out_slots, float* out_dists,
+                 hipStream_t stream);
+
+/* Map selected candidate slots to entry ids + apply the score transform
+ * (scoreFromQuantized: raw/mul^2, then sqrt for orig-l2 — ivfflat/search.go:
+ * 1062-1077). probe_lists/-offs describe each query's probed lists in rank
+ * order. list_slot_base maps list -> base slot of id_by_slot. */
+void launch_gather(const int32_t* sel_slots, const float* sel_dists,
+                   const int32_t* probe_lists, const int64_t* probe_offs,
+                   const int64_t* list_slot_base, const int64_t* id_by_slot,
+                   int nprobe, int nq, int k, int do_sqrt, double inv_mul2,
+                   int64_t* out_ids, float* out_dists, hipStream_t stream);
+
+/* Same transform for uniform (brute-force / centroid) results where the slot
+ * IS the row: id = id_by_slot ? id_by_slot[slot] : slot. */
+void launch_gather_uniform(const int32_t* sel_slots, const float* sel_dists,
+                           const int64_t* id_by_slot, int nq, int k,
+                           int do_sqrt, double inv_mul2,
+                           int64_t* out_ids, float* out_dists,
+                           hipStream_t stream);
+
+/* Pack rows into the interleaved layout.
+ * group_rowbase[g]: index into slot_rows of the group's first row;
+ * group_valid[g]: valid rows in group g; slot_rows: slot -> source row. */
+void launch_pack(const float* vecs, int dim, int dpad,
+                 const int64_t* group_rowbase, const int32_t* group_valid,
+                 const int64_t* slot_rows, int64_t ngroups, float* packed,
+                 hipStream_t stream);
+
+/* Sum-of-squares per query row (cosine). queries: [nq][dpad]. */
+void launch_qnorms(const float* queries, int nq, int dpad, float* qnorms,
+                   hipStream_t stream);
+
+}  // namespace moann
+
+#endif

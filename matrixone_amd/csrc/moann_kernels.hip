@@ -1,0 +1,488 @@
+/* CDNA4 (gfx950) device kernels for the MatrixOne ANN search hot path.
+ *
+ * Written MI355X-first (not a port): 64-wide wavefronts, coalesced
+ * float4 streams over an interleaved HBM layout, LDS-staged query tiles,
+ * per-query radix select. The list-scan kernel replaces the reference's
+ * per-row SimSIMD distance loop (pkg/vectorindex/ivfflat/relation_search.go:
+ * 334-368 + pkg/sql/plan/function/func_binary.go:9045-9175) and the cuVS
+ * ivf_flat list-scan launch (cgo/cuvs/ivf_flat.hpp:921-945 precedent);
+ * results follow MO's distance conventions (metric/distance_func.go:
+ * IP = -x.q at :207, cosine clamp + denom-0 at :211-286).
+ *
+ * Data layout: vectors are packed per list into groups of 64 rows,
+ * dim-major inside the group in float4 quads:
+ *   packed[(g*dpad + 4*j4)*64 + lane*4 + c] = row(g*64+lane) dim(4*j4+c)
+ * so one wave reading quad j4 for its 64 rows issues one fully coalesced
+ * 1 KiB load (16 B/lane). Queries are tiled (QT<=8 per workgroup) in LDS and
+ * broadcast-read, so each HBM byte of list data serves QT queries.
+ */
+
+#include <hip/hip_runtime.h>
+#include <cfloat>
+#include <cstdint>
+
+#include "moann_internal.h"
+
+namespace moann {
+
+#define WAVE 64
+
+/* monotone float -> u32 key (ascending order preserved) */
+__device__ __forceinline__ uint32_t f2u(float f) {
+    uint32_t u = __float_as_uint(f);
+    return u ^ (((int32_t)u >> 31) | 0x80000000u);
+}
+__device__ __forceinline__ float u2f(uint32_t u) {
+    u ^= (((int32_t)(u ^ 0x80000000u)) >> 31) | 0x80000000u;
+    return __uint_as_float(u);
+}
+
+/* ------------------------------ scan -------------------------------------
+ * One workgroup (256 threads = 4 waves) per job. Each wave walks a pair of
+ * adjacent 64-row groups so every LDS broadcast of a query quad feeds two
+ * 16 B data quads (halves LDS traffic per byte). Per j4 step and query:
+ * 8 fma-class VALU ops per 32 B/lane -> VALU ceiling ~19 TB/s equivalent,
+ * comfortably above the ~6.3 TB/s HBM bound this kernel targets. */
+template <int METRIC, int QT>
+__global__ __launch_bounds__(256) void scan_kernel(
+    const float* __restrict__ packed, const float* __restrict__ queries,
+    const float* __restrict__ qnorms, int dpad,
+    const int64_t* __restrict__ j_databaseg,
+    const int32_t* __restrict__ j_gstart, const int32_t* __restrict__ j_gcount,
+    const int32_t* __restrict__ j_rows, const int32_t* __restrict__ j_nq,
+    const int32_t* __restrict__ j_qbase,
+    const int32_t* __restrict__ qslot_query,
+    const int64_t* __restrict__ qslot_outbase,
+    float* __restrict__ dists_out) {
+    extern __shared__ float lds[];  /* [QT][dpad] (+[QT] qnorm for cos) */
+    float* ldsq = lds;
+    float* ldsn = lds + QT * dpad;
+
+    const int j = blockIdx.x;
+    const int nq = j_nq[j];
+    const int qbase = j_qbase[j];
+
+    for (int t = 0; t < QT; ++t) {
+        if (t < nq) {
+            const int q = qslot_query[qbase + t];
+            const float* src = queries + (int64_t)q * dpad;
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = src[e];
+            if (METRIC == KM_COS && threadIdx.x == 0) ldsn[t] = qnorms[q];
+        } else {
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = 0.f;
+            if (METRIC == KM_COS && threadIdx.x == 0) ldsn[t] = 0.f;
+        }
+    }
+    __syncthreads();
+
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int gstart = j_gstart[j], gcount = j_gcount[j];
+    const int rows = j_rows[j];
+    const int64_t baseg = j_databaseg[j];
+    const int d4 = dpad >> 2;
+
+    /* each wave takes group pairs (2*wave, 2*wave+1) strided by 8 */
+    for (int gp = gstart + 2 * wave; gp < gstart + gcount; gp += 8) {
+        const int g0 = gp;
+        const bool has1 = (gp + 1) < (gstart + gcount);
+        const int g1 = has1 ? gp + 1 : gp;
+        const float4* __restrict__ d0 =
+            (const float4*)(packed + (baseg + g0) * (int64_t)64 * dpad) + lane;
+        const float4* __restrict__ d1 =
+            (const float4*)(packed + (baseg + g1) * (int64_t)64 * dpad) + lane;
+
+        float acc0[QT], acc1[QT];
+#pragma unroll
+        for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0.f;
+        float rn0 = 0.f, rn1 = 0.f;
+
+        for (int q4 = 0; q4 < d4; ++q4) {
+            const float4 x0 = d0[q4 * 64];
+            const float4 x1 = d1[q4 * 64];
+            if (METRIC == KM_COS) {
+                rn0 += x0.x * x0.x + x0.y * x0.y + x0.z * x0.z + x0.w * x0.w;
+                rn1 += x1.x * x1.x + x1.y * x1.y + x1.z * x1.z + x1.w * x1.w;
+            }
+#pragma unroll
+            for (int t = 0; t < QT; ++t) {
+                const float4 qv = ((const float4*)(ldsq + t * dpad))[q4];
+                if (METRIC == KM_L2SQ) {
+                    float e;
+                    e = x0.x - qv.x; acc0[t] = fmaf(e, e, acc0[t]);
+                    e = x0.y - qv.y; acc0[t] = fmaf(e, e, acc0[t]);
+                    e = x0.z - qv.z; acc0[t] = fmaf(e, e, acc0[t]);
+                    e = x0.w - qv.w; acc0[t] = fmaf(e, e, acc0[t]);
+                    e = x1.x - qv.x; acc1[t] = fmaf(e, e, acc1[t]);
+                    e = x1.y - qv.y; acc1[t] = fmaf(e, e, acc1[t]);
+                    e = x1.z - qv.z; acc1[t] = fmaf(e, e, acc1[t]);
+                    e = x1.w - qv.w; acc1[t] = fmaf(e, e, acc1[t]);
+                } else if (METRIC == KM_IP || METRIC == KM_COS) {
+                    acc0[t] = fmaf(x0.x, qv.x, acc0[t]);
+                    acc0[t] = fmaf(x0.y, qv.y, acc0[t]);
+                    acc0[t] = fmaf(x0.z, qv.z, acc0[t]);
+                    acc0[t] = fmaf(x0.w, qv.w, acc0[t]);
+                    acc1[t] = fmaf(x1.x, qv.x, acc1[t]);
+                    acc1[t] = fmaf(x1.y, qv.y, acc1[t]);
+                    acc1[t] = fmaf(x1.z, qv.z, acc1[t]);
+                    acc1[t] = fmaf(x1.w, qv.w, acc1[t]);
+                } else { /* KM_L1 */
+                    acc0[t] += fabsf(x0.x - qv.x) + fabsf(x0.y - qv.y) +
+                               fabsf(x0.z - qv.z) + fabsf(x0.w - qv.w);
+                    acc1[t] += fabsf(x1.x - qv.x) + fabsf(x1.y - qv.y) +
+                               fabsf(x1.z - qv.z) + fabsf(x1.w - qv.w);
+                }
+            }
+        }
+
+        const int row0 = g0 * 64 + lane;
+        const int row1 = g1 * 64 + lane;
+#pragma unroll
+        for (int t = 0; t < QT; ++t) {
+            if (t >= nq) break;
+            const int64_t ob = qslot_outbase[qbase + t];
+            if (row0 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = -acc0[t];
+                else if (METRIC == KM_COS) {
+                    /* distance_func.go:211-286: double denom, clamp, 0 -> 1 */
+                    double denom = sqrt((double)rn0) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc0[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else dist = acc0[t];
+                dists_out[ob + row0] = dist;
+            }
+            if (has1 && row1 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = -acc1[t];
+                else if (METRIC == KM_COS) {
+                    double denom = sqrt((double)rn1) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc1[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else dist = acc1[t];
+                dists_out[ob + row1] = dist;
+            }
+        }
+    }
+}
+
+template <int METRIC>
+static void launch_scan_qt(int qt, const float* packed, const float* queries,
+                           const float* qnorms, int dpad, const ScanJobs& jb,
+                           float* dists_out, hipStream_t stream) {
+    const dim3 grid(jb.njobs), block(256);
+#define ARGS packed, queries, qnorms, dpad, jb.databaseg, jb.gstart, \
+    jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase, \
+    dists_out
+    switch (qt) {
+    case 8:
+        hipLaunchKernelGGL((scan_kernel<METRIC, 8>), grid, block,
+                           (8 * dpad + 8) * sizeof(float), stream, ARGS);
+        break;
+    case 4:
+        hipLaunchKernelGGL((scan_kernel<METRIC, 4>), grid, block,
+                           (4 * dpad + 4) * sizeof(float), stream, ARGS);
+        break;
+    case 2:
+        hipLaunchKernelGGL((scan_kernel<METRIC, 2>), grid, block,
+                           (2 * dpad + 2) * sizeof(float), stream, ARGS);
+        break;
+    default:
+        hipLaunchKernelGGL((scan_kernel<METRIC, 1>), grid, block,
+                           (1 * dpad + 1) * sizeof(float), stream, ARGS);
+        break;
+    }
+#undef ARGS
+}
+
+void launch_scan(int metric, int qt, const float* packed, const float* queries,
+                 const float* qnorms, int dpad, const ScanJobs& jb,
+                 float* dists_out, hipStream_t stream) {
+    if (jb.njobs == 0) return;
+    switch (metric) {
+    case KM_L2SQ: launch_scan_qt<KM_L2SQ>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
+    case KM_IP:   launch_scan_qt<KM_IP>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
+    case KM_COS:  launch_scan_qt<KM_COS>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
+    default:      launch_scan_qt<KM_L1>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
+    }
+}
+
+/* ------------------------------ top-k ------------------------------------
+ * One workgroup per query: 4-level 8-bit radix select over the monotone u32
+ * keys (exact kth threshold), then gather of the k winners and an in-LDS
+ * bitonic sort of packed (key32|slot32) u64s — ascending by distance, slot as
+ * the tie-breaker, giving deterministic output. k <= 4096.
+ * Replaces the bounded heaps of pkg/vectorindex/index.go:135-309 and the
+ * cuVS select::topk stage. Padding: (-1, FLT_MAX) per cgo/cuvs/helper.h:160. */
+__global__ __launch_bounds__(256) void topk_kernel(
+    const float* __restrict__ dists, const int64_t* __restrict__ qoffs,
+    int64_t uniform_n, int k, int sortP,
+    int32_t* __restrict__ out_slots, float* __restrict__ out_dists) {
+    extern __shared__ unsigned char smem[];
+    uint32_t* hist = (uint32_t*)smem;                    /* 256 */
+    uint32_t* ctr = hist + 256;                          /* n_lt, n_tie */
+    uint32_t* st = ctr + 8;  /* prefix, c_sure, tieval, need_ties, done */
+    uint64_t* keys = (uint64_t*)(smem + 2048);           /* [sortP] */
+
+    const int q = blockIdx.x;
+    const int64_t off = qoffs ? qoffs[q] : (int64_t)q * uniform_n;
+    const int64_t count = (qoffs ? qoffs[q + 1] : (int64_t)(q + 1) * uniform_n) - off;
+    const int tid = threadIdx.x;
+    const int m = (int)(count < (int64_t)k ? count : k);
+
+    if (count > (int64_t)k) {
+        /* radix select for the exact threshold */
+        if (tid == 0) { st[0] = 0; st[1] = 0; st[3] = 0; st[4] = 0; }
+        __syncthreads();
+        for (int level = 0; level < 4 && !st[4]; ++level) {
+            const int shift = 24 - 8 * level;
+            for (int b = tid; b < 256; b += blockDim.x) hist[b] = 0;
+            __syncthreads();
+            const uint32_t pref = st[0];
+            for (int64_t i = tid; i < count; i += blockDim.x) {
+                const uint32_t u = f2u(dists[off + i]);
+                if (level == 0 || (u >> (shift + 8)) == (pref >> (shift + 8)))
+                    atomicAdd(&hist[(u >> shift) & 255], 1u);
+            }
+            __syncthreads();
+            if (tid == 0) {
+                const uint32_t need = (uint32_t)k - st[1];
+                uint32_t cum = 0; int b = 0;
+                for (; b < 256; ++b) {
+                    if (cum + hist[b] >= need) break;
+                    cum += hist[b];
+                }
+                st[0] = pref | ((uint32_t)b << shift);
+                st[1] += cum;
+                if (cum + hist[b] == need) {
+                    /* whole bin selected: no tie resolution needed */
+                    st[1] += hist[b];
+                    st[0] |= (shift ? ((1u << shift) - 1u) : 0u);
+                    st[3] = 0; st[4] = 1;
+                } else if (level == 3) {
+                    st[2] = st[0];
+                    st[3] = need - cum;
+                }
+            }
+            __syncthreads();
+        }
+        const uint32_t c_sure = st[1];
+        const uint32_t tieval = st[2];
+        const uint32_t need_ties = st[3];
+        const int early = st[4];
+        /* strict-below bound: early-exit keeps everything <= prefix (low bits
+         * already forced to 1), full depth keeps u < prefix exactly */
+        const uint64_t below = (uint64_t)st[0] + (early ? 1u : 0u);
+        if (tid == 0) { ctr[0] = 0; ctr[1] = 0; }
+        __syncthreads();
+        for (int64_t i = tid; i < count; i += blockDim.x) {
+            const uint32_t u = f2u(dists[off + i]);
+            if ((uint64_t)u < below) {
+                const uint32_t pos = atomicAdd(&ctr[0], 1u);
+                keys[pos] = ((uint64_t)u << 32) | (uint32_t)i;
+            } else if (need_ties && u == tieval) {
+                const uint32_t t = atomicAdd(&ctr[1], 1u);
+                if (t < need_ties)
+                    keys[c_sure + t] = ((uint64_t)u << 32) | (uint32_t)i;
+            }
+        }
+        __syncthreads();
+    } else {
+        for (int64_t i = tid; i < count; i += blockDim.x)
+            keys[i] = ((uint64_t)f2u(dists[off + i]) << 32) | (uint32_t)i;
+        __syncthreads();
+    }
+
+    for (int i = m + tid; i < sortP; i += blockDim.x) keys[i] = ~0ULL;
+    __syncthreads();
+
+    /* bitonic sort ascending over sortP u64 keys */
+    for (int ks = 2; ks <= sortP; ks <<= 1) {
+        for (int jj = ks >> 1; jj > 0; jj >>= 1) {
+            for (int i = tid; i < sortP; i += blockDim.x) {
+                const int l = i ^ jj;
+                if (l > i) {
+                    const uint64_t a = keys[i], b = keys[l];
+                    const bool up = (i & ks) == 0;
+                    if ((a > b) == up) { keys[i] = b; keys[l] = a; }
+                }
+            }
+            __syncthreads();
+        }
+    }
+
+    for (int i = tid; i < k; i += blockDim.x) {
+        if (i < m) {
+            const uint64_t kv = keys[i];
+            out_slots[(int64_t)q * k + i] = (int32_t)(kv & 0xFFFFFFFFu);
+            out_dists[(int64_t)q * k + i] = u2f((uint32_t)(kv >> 32));
+        } else {
+            out_slots[(int64_t)q * k + i] = -1;
+            out_dists[(int64_t)q * k + i] = FLT_MAX;
+        }
+    }
+}
+
+void launch_topk(const float* dists, const int64_t* qoffs, int64_t uniform_n,
+                 int nq, int k, int32_t* out_slots, float* out_dists,
+                 hipStream_t stream) {
+    if (nq == 0) return;
+    int sortP = 1;
+    while (sortP < k) sortP <<= 1;
+    if (sortP < 256) sortP = 256;
+    const size_t shmem = 2048 + (size_t)sortP * 8;
+    hipLaunchKernelGGL(topk_kernel, dim3(nq), dim3(256), shmem, stream,
+                       dists, qoffs, uniform_n, k, sortP, out_slots, out_dists);
+}
+
+/* ------------------------------ gather ----------------------------------- */
+
+__device__ __forceinline__ float transform_score_dev(float raw, int do_sqrt,
+                                                     double inv_mul2) {
+    /* ivfflat/search.go:1062-1077 + metric/types.go:245-251 */
+    double r = (double)raw * inv_mul2;
+    if (do_sqrt) r = sqrt(r);
+    return (float)r;
+}
+
+__global__ void gather_kernel(
+    const int32_t* __restrict__ sel_slots, const float* __restrict__ sel_dists,
+    const int32_t* __restrict__ probe_lists, const int64_t* __restrict__ probe_offs,
+    const int64_t* __restrict__ list_slot_base, const int64_t* __restrict__ id_by_slot,
+    int nprobe, int nq, int k, int do_sqrt, double inv_mul2,
+    int64_t* __restrict__ out_ids, float* __restrict__ out_dists) {
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (int64_t)nq * k) return;
+    const int q = (int)(idx / k);
+    const int32_t slot = sel_slots[idx];
+    if (slot < 0) { out_ids[idx] = -1; out_dists[idx] = FLT_MAX; return; }
+    /* binary search rank r: probe_offs[q][r] <= slot < probe_offs[q][r+1] */
+    const int64_t* offs = probe_offs + (int64_t)q * (nprobe + 1);
+    int lo = 0, hi = nprobe - 1;
+    while (lo < hi) {
+        const int mid = (lo + hi + 1) >> 1;
+        if (offs[mid] <= slot) lo = mid; else hi = mid - 1;
+    }
+    const int32_t list = probe_lists[(int64_t)q * nprobe + lo];
+    const int64_t local = slot - offs[lo];
+    out_ids[idx] = id_by_slot[list_slot_base[list] + local];
+    out_dists[idx] = transform_score_dev(sel_dists[idx], do_sqrt, inv_mul2);
+}
+
+void launch_gather(const int32_t* sel_slots, const float* sel_dists,
+                   const int32_t* probe_lists, const int64_t* probe_offs,
+                   const int64_t* list_slot_base, const int64_t* id_by_slot,
+                   int nprobe, int nq, int k, int do_sqrt, double inv_mul2,
+                   int64_t* out_ids, float* out_dists, hipStream_t stream) {
+    const int64_t total = (int64_t)nq * k;
+    if (!total) return;
+    const int block = 256;
+    const int grid = (int)((total + block - 1) / block);
+    hipLaunchKernelGGL(gather_kernel, dim3(grid), dim3(block), 0, stream,
+                       sel_slots, sel_dists, probe_lists, probe_offs,
+                       list_slot_base, id_by_slot, nprobe, nq, k, do_sqrt,
+                       inv_mul2, out_ids, out_dists);
+}
+
+__global__ void gather_uniform_kernel(
+    const int32_t* __restrict__ sel_slots, const float* __restrict__ sel_dists,
+    const int64_t* __restrict__ id_by_slot, int nq, int k, int do_sqrt,
+    double inv_mul2, int64_t* __restrict__ out_ids,
+    float* __restrict__ out_dists) {
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (int64_t)nq * k) return;
+    const int32_t slot = sel_slots[idx];
+    if (slot < 0) { out_ids[idx] = -1; out_dists[idx] = FLT_MAX; return; }
+    out_ids[idx] = id_by_slot ? id_by_slot[slot] : (int64_t)slot;
+    out_dists[idx] = transform_score_dev(sel_dists[idx], do_sqrt, inv_mul2);
+}
+
+void launch_gather_uniform(const int32_t* sel_slots, const float* sel_dists,
+                           const int64_t* id_by_slot, int nq, int k,
+                           int do_sqrt, double inv_mul2, int64_t* out_ids,
+                           float* out_dists, hipStream_t stream) {
+    const int64_t total = (int64_t)nq * k;
+    if (!total) return;
+    const int block = 256;
+    const int grid = (int)((total + block - 1) / block);
+    hipLaunchKernelGGL(gather_uniform_kernel, dim3(grid), dim3(block), 0,
+                       stream, sel_slots, sel_dists, id_by_slot, nq, k,
+                       do_sqrt, inv_mul2, out_ids, out_dists);
+}
+
+/* ------------------------------ pack -------------------------------------
+ * One-time permutation into the interleaved layout: thread = one output
+ * float4; writes coalesced, reads gather (4 B x 4 from the source row). */
+__global__ void pack_kernel(const float* __restrict__ vecs, int dim, int dpad,
+                            const int64_t* __restrict__ group_rowbase,
+                            const int32_t* __restrict__ group_valid,
+                            const int64_t* __restrict__ slot_rows,
+                            int64_t ngroups, float* __restrict__ packed) {
+    const int64_t nquads = ngroups * (dpad >> 2) * 64;
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= nquads) return;
+    const int lane = (int)(idx & 63);
+    const int64_t rest = idx >> 6;
+    const int j4 = (int)(rest % (dpad >> 2));
+    const int64_t g = rest / (dpad >> 2);
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (lane < group_valid[g]) {
+        const int64_t row = slot_rows[group_rowbase[g] + lane];
+        const float* src = vecs + row * (int64_t)dim;
+        const int base = 4 * j4;
+        v.x = base + 0 < dim ? src[base + 0] : 0.f;
+        v.y = base + 1 < dim ? src[base + 1] : 0.f;
+        v.z = base + 2 < dim ? src[base + 2] : 0.f;
+        v.w = base + 3 < dim ? src[base + 3] : 0.f;
+    }
+    ((float4*)packed)[idx] = v;
+}
+
+void launch_pack(const float* vecs, int dim, int dpad,
+                 const int64_t* group_rowbase, const int32_t* group_valid,
+                 const int64_t* slot_rows, int64_t ngroups, float* packed,
+                 hipStream_t stream) {
+    const int64_t nquads = ngroups * (int64_t)(dpad >> 2) * 64;
+    if (!nquads) return;
+    const int block = 256;
+    const int64_t grid = (nquads + block - 1) / block;
+    hipLaunchKernelGGL(pack_kernel, dim3((uint32_t)grid), dim3(block), 0,
+                       stream, vecs, dim, dpad, group_rowbase, group_valid,
+                       slot_rows, ngroups, packed);
+}
+
+/* --------------------------- query norms ---------------------------------
+ * one wave per query; f32 accumulate (matches distance_func.go cosine's f32
+ * norm accumulation within the 1e-5 parity tolerance). */
+__global__ void qnorm_kernel(const float* __restrict__ queries, int nq,
+                             int dpad, float* __restrict__ qnorms) {
+    const int q = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (q >= nq) return;
+    const float* src = queries + (int64_t)q * dpad;
+    float s = 0.f;
+    for (int e = lane; e < dpad; e += 64) s += src[e] * src[e];
+#pragma unroll
+    for (int w = 32; w; w >>= 1) s += __shfl_down(s, w, 64);
+    if (lane == 0) qnorms[q] = s;
+}
+
+void launch_qnorms(const float* queries, int nq, int dpad, float* qnorms,
+                   hipStream_t stream) {
+    if (!nq) return;
+    const int wpb = 4;
+    const int grid = (nq + wpb - 1) / wpb;
+    hipLaunchKernelGGL(qnorm_kernel, dim3(grid), dim3(wpb * 64), 0, stream,
+                       queries, nq, dpad, qnorms);
+}
+
+}  // namespace moann

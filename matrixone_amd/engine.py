@@ -1,0 +1,286 @@
+"""ctypes binding over libmoann_hip.so — the C-ABI boundary of the MI355X
+backend (include/moann.h, mirroring cgo/cuvs/ivf_flat_c.h).
+
+FAIL-LOUD CONTRACT: this module raises immediately if the HIP extension is
+missing or a call fails. There is NO CPU fallback anywhere in the product
+path — the oracle under oracle/ is test infrastructure only and is never
+imported here.
+"""
+
+from __future__ import annotations
+
+import ctypes as ct
+import os
+
+import numpy as np
+
+_PKG = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_PKG, "libmoann_hip.so")
+
+# distance_type_t (include/moann.h, values from cgo/cuvs/cuvs_types.h)
+METRICS = {
+    "l2sq": 0,   # DistanceType_L2Expanded
+    "l2": 1,     # DistanceType_L2SqrtExpanded (sqrt applied at gather)
+    "cos": 2,    # DistanceType_CosineExpanded
+    "l1": 3,     # DistanceType_L1
+    "ip": 6,     # DistanceType_InnerProduct (MO convention -a.b)
+}
+
+
+class BuildParams(ct.Structure):
+    _fields_ = [("n_lists", ct.c_uint32), ("add_data_on_build", ct.c_bool),
+                ("kmeans_trainset_fraction", ct.c_double)]
+
+
+class SearchParams(ct.Structure):
+    _fields_ = [("n_probes", ct.c_uint32)]
+
+
+class SearchRes(ct.Structure):
+    _fields_ = [("result_ptr", ct.c_void_p)]
+
+
+class PerfT(ct.Structure):
+    _fields_ = [
+        ("scan_launches", ct.c_uint64), ("scan_ms", ct.c_double),
+        ("scan_bytes", ct.c_uint64), ("scan_rows", ct.c_uint64),
+        ("rank_launches", ct.c_uint64), ("rank_ms", ct.c_double),
+        ("rank_flops", ct.c_uint64),
+        ("select_ms", ct.c_double), ("other_ms", ct.c_double),
+    ]
+
+
+_lib = None
+
+
+def lib() -> ct.CDLL:
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB_PATH):
+        raise RuntimeError(
+            f"moann HIP extension missing at {_LIB_PATH} — build it with "
+            "`make -C matrixone_amd/csrc` or __graft_entry__.build(). "
+            "There is no CPU fallback.")
+    L = ct.CDLL(_LIB_PATH)
+    L.gpu_ivf_flat_new_empty.restype = ct.c_void_p
+    L.gpu_ivf_flat_new_empty.argtypes = [
+        ct.c_uint64, ct.c_uint32, ct.c_int, BuildParams,
+        ct.POINTER(ct.c_int), ct.c_int, ct.c_uint32, ct.c_int, ct.c_int,
+        ct.c_int, ct.c_void_p, ct.c_void_p]
+    L.gpu_ivf_flat_add_chunk.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_flat_set_centroids.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint32, ct.c_void_p]
+    L.moann_ivf_flat_set_assignments.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    for f in (L.gpu_ivf_flat_build, L.gpu_ivf_flat_start):
+        f.argtypes = [ct.c_void_p, ct.c_void_p]
+    L.gpu_ivf_flat_destroy.argtypes = [ct.c_void_p, ct.c_void_p]
+    for f in (L.gpu_ivf_flat_search, L.moann_ivf_flat_search_device):
+        f.restype = SearchRes
+        f.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32,
+                      ct.c_uint32, SearchParams, ct.c_void_p]
+    L.gpu_ivf_flat_search_async.restype = ct.c_uint64
+    L.gpu_ivf_flat_search_async.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
+        SearchParams, ct.c_void_p]
+    L.gpu_ivf_flat_search_wait.restype = SearchRes
+    L.gpu_ivf_flat_search_wait.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    L.gpu_ivf_flat_get_neighbors.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    L.gpu_ivf_flat_get_distances.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    L.gpu_ivf_flat_free_result.argtypes = [ct.c_void_p]
+    for f in (L.gpu_ivf_flat_cap, L.gpu_ivf_flat_len):
+        f.restype = ct.c_uint64
+        f.argtypes = [ct.c_void_p]
+    L.gpu_ivf_flat_get_n_list.restype = ct.c_uint32
+    L.gpu_ivf_flat_get_n_list.argtypes = [ct.c_void_p]
+    L.moann_brute_force_search.argtypes = [
+        ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, ct.c_void_p,
+        ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, ct.c_void_p,
+        ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_flat_perf.argtypes = [ct.c_void_p, ct.POINTER(PerfT)]
+    L.moann_ivf_flat_perf_reset.argtypes = [ct.c_void_p]
+    L.moann_device_count.restype = ct.c_int
+    L.moann_version.restype = ct.c_char_p
+    _lib = L
+    return L
+
+
+class MoannError(RuntimeError):
+    pass
+
+
+class _Err:
+    """errmsg out-param holder (char** per cgo/cuvs/helper.h:59)."""
+
+    def __init__(self):
+        self.p = ct.c_char_p(None)
+
+    @property
+    def ref(self):
+        return ct.byref(self.p)
+
+    def check(self, where: str):
+        if self.p.value:
+            msg = self.p.value.decode()
+            raise MoannError(f"{where}: {msg}")
+
+
+def device_count() -> int:
+    return lib().moann_device_count()
+
+
+def _as_f32(a) -> np.ndarray:
+    arr = np.ascontiguousarray(a, dtype=np.float32)
+    return arr
+
+
+class IvfFlatIndex:
+    """IVF-Flat index on one MI355X.
+
+    Mirrors the lifecycle the reference Go binding drives through the C ABI
+    (pkg/cuvs/ivf_flat.go:35-1125: New / AddChunk / Build / Search /
+    SearchAsync / SearchWait), with the MatrixOne-CPU-path ingestion shape
+    (explicit centroids + entry assignments from the hidden tables)."""
+
+    def __init__(self, dim: int, nlist: int, metric: str = "l2sq",
+                 capacity: int = 0, device: int = 0):
+        if metric not in METRICS:
+            raise ValueError(f"metric {metric!r} not in {sorted(METRICS)}")
+        self.dim, self.nlist, self.metric = dim, nlist, metric
+        self.device = device
+        err = _Err()
+        bp = BuildParams(nlist, True, 0.5)
+        dev = (ct.c_int * 1)(device)
+        self._h = lib().gpu_ivf_flat_new_empty(
+            capacity, dim, METRICS[metric], bp, dev, 1, 0, 0, 0, 0, None,
+            err.ref)
+        err.check("new_empty")
+        if not self._h:
+            raise MoannError("gpu_ivf_flat_new_empty returned NULL")
+
+    def add(self, vecs, ids=None):
+        vecs = _as_f32(vecs)
+        idp = None
+        if ids is not None:
+            ids = np.ascontiguousarray(ids, dtype=np.int64)
+            idp = ids.ctypes.data
+        err = _Err()
+        lib().gpu_ivf_flat_add_chunk(self._h, vecs.ctypes.data, vecs.shape[0],
+                                     idp, err.ref)
+        err.check("add_chunk")
+
+    def set_centroids(self, centroids):
+        centroids = _as_f32(centroids)
+        err = _Err()
+        lib().moann_ivf_flat_set_centroids(self._h, centroids.ctypes.data,
+                                           centroids.shape[0], err.ref)
+        err.check("set_centroids")
+
+    def set_assignments(self, assignments):
+        assignments = np.ascontiguousarray(assignments, dtype=np.int32)
+        err = _Err()
+        lib().moann_ivf_flat_set_assignments(self._h, assignments.ctypes.data,
+                                             assignments.shape[0], err.ref)
+        err.check("set_assignments")
+
+    def build(self):
+        err = _Err()
+        lib().gpu_ivf_flat_build(self._h, err.ref)
+        err.check("build")
+        lib().gpu_ivf_flat_start(self._h, err.ref)
+        err.check("start")
+
+    def __len__(self):
+        return lib().gpu_ivf_flat_len(self._h)
+
+    def _read_result(self, res, nq, k, where):
+        if not res.result_ptr:
+            raise MoannError(f"{where} returned NULL result")
+        ids = np.empty((nq, k), dtype=np.int64)
+        dists = np.empty((nq, k), dtype=np.float32)
+        L = lib()
+        L.gpu_ivf_flat_get_neighbors(res.result_ptr, nq * k, ids.ctypes.data)
+        L.gpu_ivf_flat_get_distances(res.result_ptr, nq * k, dists.ctypes.data)
+        L.gpu_ivf_flat_free_result(res.result_ptr)
+        return ids, dists
+
+    def search(self, queries, k: int, nprobe: int):
+        queries = _as_f32(queries)
+        nq = queries.shape[0]
+        err = _Err()
+        res = lib().gpu_ivf_flat_search(self._h, queries.ctypes.data, nq,
+                                        queries.shape[1], k,
+                                        SearchParams(nprobe), err.ref)
+        err.check("search")
+        return self._read_result(res, nq, k, "search")
+
+    def search_device(self, queries_tensor, k: int, nprobe: int):
+        """queries_tensor: a contiguous f32 torch CUDA tensor [nq][dim]; the
+        caller must have synchronised the producing stream."""
+        nq, qdim = queries_tensor.shape
+        err = _Err()
+        res = lib().moann_ivf_flat_search_device(
+            self._h, ct.c_void_p(queries_tensor.data_ptr()), nq, qdim, k,
+            SearchParams(nprobe), err.ref)
+        err.check("search_device")
+        return self._read_result(res, nq, k, "search_device")
+
+    def search_async(self, queries, k: int, nprobe: int):
+        queries = _as_f32(queries)
+        self._async_shape = (queries.shape[0], k)
+        err = _Err()
+        job = lib().gpu_ivf_flat_search_async(
+            self._h, queries.ctypes.data, queries.shape[0], queries.shape[1],
+            k, SearchParams(nprobe), err.ref)
+        err.check("search_async")
+        return job
+
+    def search_wait(self, job: int):
+        err = _Err()
+        res = lib().gpu_ivf_flat_search_wait(self._h, job, err.ref)
+        err.check("search_wait")
+        nq, k = self._async_shape
+        return self._read_result(res, nq, k, "search_wait")
+
+    def perf(self) -> dict:
+        p = PerfT()
+        lib().moann_ivf_flat_perf(self._h, ct.byref(p))
+        return {f[0]: getattr(p, f[0]) for f in PerfT._fields_}
+
+    def perf_reset(self):
+        lib().moann_ivf_flat_perf_reset(self._h)
+
+    def close(self):
+        if getattr(self, "_h", None):
+            err = _Err()
+            lib().gpu_ivf_flat_destroy(self._h, err.ref)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def brute_force_search(dataset, queries, k: int, metric: str = "l2sq",
+                       ids=None, device: int = 0):
+    """Exact GPU top-k (pkg/vectorindex/brute_force counterpart)."""
+    dataset = _as_f32(dataset)
+    queries = _as_f32(queries)
+    nq = queries.shape[0]
+    out_ids = np.empty((nq, k), dtype=np.int64)
+    out_dists = np.empty((nq, k), dtype=np.float32)
+    idp = None
+    if ids is not None:
+        ids = np.ascontiguousarray(ids, dtype=np.int64)
+        idp = ids.ctypes.data
+    err = _Err()
+    lib().moann_brute_force_search(
+        dataset.ctypes.data, dataset.shape[0], dataset.shape[1],
+        METRICS[metric], idp, queries.ctypes.data, nq, k, device,
+        out_ids.ctypes.data, out_dists.ctypes.data, err.ref)
+    err.check("brute_force_search")
+    return out_ids, out_dists

@@ -1,0 +1,93 @@
+"""Multi-process CPU tests (gloo, world_size=2) for the sharded-search
+exchange path: per-shard top-k -> all_gather -> merge must equal a whole-set
+search. Shard results are produced by the oracle (no GPU here); on the GPU
+the same merge runs after RCCL all-gather (matrixone_amd/distributed.py)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from oracle import oracle as orc
+from matrixone_amd.distributed import allgather_merge, merge_topk_torch
+
+
+def test_merge_topk_torch_matches_oracle_merge():
+    rng = np.random.Generator(np.random.PCG64(1))
+    nshard, nq, k = 4, 16, 12
+    ids = rng.integers(0, 10000, size=(nshard, nq, k)).astype(np.int64)
+    dists = rng.random((nshard, nq, k)).astype(np.float32)
+    # sentinel-pad some tails + sort each block ascending (dense contract)
+    dists.sort(axis=2)
+    ids[:, :, -3:] = -1
+    dists[:, :, -3:] = np.finfo(np.float32).max
+    ref_i, ref_d = orc.topk_merge(ids, dists, k)
+    out_i, out_d = merge_topk_torch(torch.from_numpy(ids),
+                                    torch.from_numpy(dists), k)
+    np.testing.assert_allclose(out_d.numpy(), ref_d, rtol=0, atol=0)
+    # ids may differ at exact FP ties only
+    mism = out_i.numpy() != ref_i
+    assert (out_d.numpy()[mism] == ref_d[mism]).all()
+
+
+def _worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        rng = np.random.Generator(np.random.PCG64(77))
+        n, d, nlist, nq, probe, k = 4000, 32, 16, 8, 4, 10
+        vecs = rng.standard_normal((n, d), dtype=np.float32)
+        cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+        assign = ((vecs[:, None, :].astype(np.float64) -
+                   cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+        queries = rng.standard_normal((nq, d), dtype=np.float32)
+
+        # shard by list parity (disjoint lists per rank — SHARDED mode shape,
+        # index_base.hpp:104-132); ids = global row numbers
+        mine = (assign % world) == rank
+        rows = np.nonzero(mine)[0]
+        sub_assign = assign[rows] // world  # relabel lists 0..nlist/world-1
+        sub_cents = cents[rank::world]
+        idx = orc.IvfIndex(sub_cents, vecs[rows], sub_assign,
+                           ids=rows.astype(np.int64))
+        # probe proportionally within the shard
+        local_ids, local_dists = idx.search(orc.METRIC_L2SQ, queries,
+                                            probe, k)
+        out_i, out_d = allgather_merge(local_ids, local_dists, k)
+
+        if rank == 0:
+            # reference: whole-set search probing the union of each shard's
+            # probed lists == merge of shard results only when every shard
+            # probes its own top-`probe`; so compare against the oracle merge
+            # of the same shard blocks (exchange-path equivalence).
+            gathered_i = np.stack([local_ids, local_ids], 0)
+            # rank 0 cannot see rank 1's block here; instead recompute both
+            # shards' results deterministically:
+            blocks_i, blocks_d = [], []
+            for r in range(world):
+                m = (assign % world) == r
+                rr = np.nonzero(m)[0]
+                sidx = orc.IvfIndex(cents[r::world], vecs[rr],
+                                    assign[rr] // world,
+                                    ids=rr.astype(np.int64))
+                bi, bd = sidx.search(orc.METRIC_L2SQ, queries, probe, k)
+                blocks_i.append(bi)
+                blocks_d.append(bd)
+            ref_i, ref_d = orc.topk_merge(np.stack(blocks_i),
+                                          np.stack(blocks_d), k)
+            np.testing.assert_allclose(out_d, ref_d, rtol=0, atol=0)
+            mism = out_i != ref_i
+            assert (out_d[mism] == ref_d[mism]).all()
+            np.save(os.path.join(tmpdir, "ok.npy"), np.array([1]))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_allgather_merge_gloo(tmp_path):
+    port = 29511
+    mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    assert os.path.exists(tmp_path / "ok.npy")

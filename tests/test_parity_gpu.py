@@ -1,0 +1,181 @@
+"""GPU parity tests: the HIP product path vs the oracle (CPU restatement of
+the reference algorithms, pinned against usearch/SimSIMD) on the same seeded
+inputs. Run on a real MI355X (`pytest -m gpu`).
+
+Parity bar (BASELINE.json north star): neighbor ids bit-exact except at FP
+ties (where distances agree within 1e-5 relative); distances within 1e-5
+relative."""
+
+import glob
+import os
+
+import numpy as np
+import pytest
+
+from oracle import oracle as orc
+
+pytestmark = pytest.mark.gpu
+
+GOLDEN = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
+RTOL = 1e-5
+
+
+def _assert_parity(ids, dists, ref_ids, ref_dists, ctx=""):
+    ids = np.asarray(ids)
+    ref_ids = np.asarray(ref_ids)
+    d = np.asarray(dists, dtype=np.float64)
+    rd = np.asarray(ref_dists, dtype=np.float64)
+    np.testing.assert_allclose(d, rd, rtol=RTOL, atol=1e-6, err_msg=ctx)
+    mism = ids != ref_ids
+    if mism.any():
+        da, db = d[mism], rd[mism]
+        rel = np.abs(da - db) / np.maximum(np.maximum(np.abs(da), np.abs(db)),
+                                           1e-30)
+        assert (rel < RTOL).all(), (
+            f"{ctx}: {mism.sum()} id mismatches beyond FP ties "
+            f"(worst rel dist err {rel.max():.2e})")
+
+
+def _build_from_fixture(z, engine):
+    ix = engine.IvfFlatIndex(z["vecs"].shape[1], z["centroids"].shape[0],
+                             metric={0: "l2", 1: "l2sq", 2: "ip", 3: "cos",
+                                     4: "l1"}[int(z["metric"])],
+                             capacity=z["vecs"].shape[0])
+    ix.add(z["vecs"], ids=z["ids"])
+    ix.set_centroids(z["centroids"])
+    ix.set_assignments(z["assign"].astype(np.int32))
+    ix.build()
+    return ix
+
+
+def test_ivf_golden_fixtures():
+    """Every committed IVF fixture: HIP results == oracle results."""
+    from matrixone_amd import engine
+    for path in sorted(glob.glob(os.path.join(GOLDEN, "ivf_*.npz"))):
+        z = np.load(path)
+        ix = _build_from_fixture(z, engine)
+        ids, dists = ix.search(z["queries"], int(z["k"]), int(z["probe"]))
+        _assert_parity(ids, dists, z["out_ids"], z["out_dists"], ctx=path)
+        ix.close()
+
+
+def test_bf_golden_fixtures():
+    from matrixone_amd import engine
+    for path in sorted(glob.glob(os.path.join(GOLDEN, "bf_*.npz"))):
+        z = np.load(path)
+        ids, dists = engine.brute_force_search(
+            z["base"], z["queries"], int(z["k"]),
+            metric={1: "l2sq", 2: "ip", 3: "cos"}[int(z["metric"])])
+        _assert_parity(ids, dists, z["out_ids"], z["out_dists"], ctx=path)
+
+
+@pytest.mark.parametrize("metric,orig_l2", [
+    ("l2sq", False), ("l2", True), ("ip", False), ("cos", False),
+    ("l1", False)])
+def test_ivf_all_metrics_live(metric, orig_l2):
+    """Config-1 shape (BASELINE.json configs[0], shrunk): 20k x 128,
+    nlist=64, nprobe=8 — oracle runs in seconds; every metric."""
+    from matrixone_amd import engine
+    mmap = {"l2sq": orc.METRIC_L2SQ, "l2": orc.METRIC_L2,
+            "ip": orc.METRIC_IP, "cos": orc.METRIC_COS, "l1": orc.METRIC_L1}
+    rng = np.random.Generator(np.random.PCG64(100))
+    n, d, nlist, nq, probe, k = 20000, 128, 64, 32, 8, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = np.empty(n, dtype=np.int64)
+    for c0 in range(0, n, 4096):
+        blk = vecs[c0:c0 + 4096].astype(np.float64)
+        d2 = ((blk[:, None, :] - cents[None].astype(np.float64)) ** 2).sum(-1)
+        assign[c0:c0 + 4096] = d2.argmin(1)
+    ids64 = rng.permutation(n).astype(np.int64) + 7
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+
+    idx = orc.IvfIndex(cents, vecs, assign, ids=ids64)
+    ref_ids, ref_dists = idx.search(mmap[metric], queries, probe, k,
+                                    orig_l2=orig_l2)
+    ix = engine.IvfFlatIndex(d, nlist, metric=metric, capacity=n)
+    ix.add(vecs, ids=ids64)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids, dists = ix.search(queries, k, probe)
+    _assert_parity(ids, dists, ref_ids, ref_dists, ctx=metric)
+    ix.close()
+
+
+def test_probe_set_matches_oracle():
+    """The centroid ranking (list selection) itself must match the oracle —
+    this is what makes candidate sets identical (ivfflat/search.go:265-308)."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(200))
+    n, d, nlist, nq, probe, k = 5000, 64, 32, 16, 6, 2000
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    idx = orc.IvfIndex(cents, vecs, assign)
+    # k=2000 > all candidates in 6 lists is possible; both sides must agree
+    # on the full candidate set and its padding.
+    ref_ids, ref_dists = idx.search(orc.METRIC_L2SQ, queries, probe, k)
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids, dists = ix.search(queries, k, probe)
+    # padded tails must agree exactly
+    np.testing.assert_array_equal(ids == -1, ref_ids == -1)
+    _assert_parity(ids, dists, ref_ids, ref_dists, ctx="probe-set")
+    ix.close()
+
+
+def test_edge_cases():
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(300))
+    # empty lists, k > n, single query, dim 7 (non-multiple of 4)
+    n, d, nlist = 100, 7, 8
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = rng.standard_normal((nlist, d), dtype=np.float32)
+    assign = np.zeros(n, dtype=np.int64)  # everything in list 0 — 7 empty
+    q = rng.standard_normal((1, d), dtype=np.float32)
+    idx = orc.IvfIndex(cents, vecs, assign)
+    ref_ids, ref_dists = idx.search(orc.METRIC_L2SQ, q, 8, 200)
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids, dists = ix.search(q, 200, 8)
+    np.testing.assert_array_equal(ids == -1, ref_ids == -1)
+    _assert_parity(ids, dists, ref_ids, ref_dists, ctx="edge")
+    ix.close()
+
+
+def test_duplicate_vectors_ties():
+    """Exact duplicates: distance values must agree exactly; the id SET at
+    the tie boundary must be a valid top-k set."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(400))
+    base = rng.standard_normal((64, 16), dtype=np.float32)
+    vecs = np.repeat(base, 8, axis=0)  # 512 rows, every vector x8
+    cents = base[:4].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    q = vecs[:3] + 1e-3
+    idx = orc.IvfIndex(cents, vecs, assign)
+    ref_ids, ref_dists = idx.search(orc.METRIC_L2SQ, q, 4, 20)
+    ix = engine.IvfFlatIndex(16, 4, metric="l2sq", capacity=len(vecs))
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids, dists = ix.search(q, 20, 4)
+    np.testing.assert_allclose(dists, ref_dists, rtol=0, atol=0)
+    ix.close()
+
+
+def test_hnsw_placeholder_absent():
+    """HNSW GPU path is SURVEY.md §8 row a3 — lands in a later commit of this
+    round / round 2; this marker keeps the scope honest in CI output."""
+    pytest.skip("HNSW GPU probe not yet implemented (§8a3, planned)")
