@@ -39,6 +39,14 @@ using namespace moann;
                                      hipGetErrorString(_e) + " at " #expr); \
     } while (0)
 
+#define KCHECK(tag)                                                          \
+    do {                                                                     \
+        hipError_t _e = hipGetLastError();                                   \
+        if (_e != hipSuccess)                                                \
+            throw std::runtime_error(std::string("HIP launch error [") +     \
+                                     tag + "]: " + hipGetErrorString(_e));   \
+    } while (0)
+
 /* errmsg convention: cgo/cuvs/helper.h:59 — errmsg is a char** as void*. */
 void set_errmsg(void* errmsg, const char* context, const char* message) {
     if (!errmsg) return;
@@ -199,6 +207,7 @@ void pack_lists(IvfIndex* ix, const float* d_rows, int64_t nrows,
     launch_pack(d_rows, (int)ix->dim, (int)ix->dpad, tmp_rb.as<int64_t>(),
                 tmp_gv.as<int32_t>(), tmp_sr.as<int64_t>(), ngroups, d_packed,
                 ix->stream);
+    KCHECK("pack");
     HIP_CHECK(hipStreamSynchronize(ix->stream));
     HIP_CHECK(hipGetLastError());
     *d_packed_out = d_packed;
@@ -244,6 +253,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     if (ix->kmetric == KM_COS) {
         ix->w_qnorms.ensure(nq * 4);
         launch_qnorms(d_q, (int)nq, dpad, ix->w_qnorms.as<float>(), s);
+        KCHECK("qnorms");
         d_qn = ix->w_qnorms.as<float>();
     }
 
@@ -285,7 +295,9 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
             }
         }
         const size_t b64 = sizeof(int64_t), b32 = sizeof(int32_t);
-        size_t need = njobs * (b64 + 4 * b32 + b32) + nqslots * (b32 + b64);
+        auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
+        size_t need = al8(njobs * b64) + 5 * al8(njobs * b32) +
+                      al8(nqslots * b32) + al8(nqslots * b64);
         ix->w_jobs.ensure(need);
         uint8_t* p = ix->w_jobs.as<uint8_t>();
         ScanJobs jb;
@@ -307,12 +319,14 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         HIP_CHECK(hipEventRecord(ix->perf.ev[0], s));
         launch_scan(ix->kmetric, qt, ix->d_cent_packed, d_q, d_qn, dpad, jb,
                     ix->w_cdists.as<float>(), s);
+        KCHECK("rank-scan");
         HIP_CHECK(hipEventRecord(ix->perf.ev[1], s));
         ix->w_csel_slots.ensure((size_t)nq * probe * 4);
         ix->w_csel_dists.ensure((size_t)nq * probe * 4);
         launch_topk(ix->w_cdists.as<float>(), nullptr, ix->nlist, (int)nq,
                     (int)probe, ix->w_csel_slots.as<int32_t>(),
                     ix->w_csel_dists.as<float>(), s);
+        KCHECK("rank-topk");
     }
 
     /* 3. D2H the probe selection; build list-scan jobs on the host
@@ -404,8 +418,9 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     ix->w_cand.ensure(std::max<int64_t>(1, total_cand) * 4);
     {
         const size_t b64 = sizeof(int64_t), b32 = sizeof(int32_t);
-        size_t need = (size_t)njobs * (2 * b64 + 5 * b32) +
-                      (size_t)nqslots * (b32 + b64) + 64 * 8;
+        auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
+        size_t need = al8((size_t)njobs * b64) + 5 * al8((size_t)njobs * b32) +
+                      al8((size_t)nqslots * b32) + al8((size_t)nqslots * b64) + 64;
         ix->w_qslots.ensure(need);
         uint8_t* p = ix->w_qslots.as<uint8_t>();
         auto up = [&](const void* src, size_t bytes) {
@@ -428,6 +443,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         HIP_CHECK(hipEventRecord(ix->perf.ev[2], s));
         launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
                     ix->w_cand.as<float>(), s);
+        KCHECK("list-scan");
         HIP_CHECK(hipEventRecord(ix->perf.ev[3], s));
     }
 
@@ -441,6 +457,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     launch_topk(ix->w_cand.as<float>(), ix->w_qoffs.as<int64_t>(), 0, (int)nq,
                 (int)limit, ix->w_sel_slots.as<int32_t>(),
                 ix->w_sel_dists.as<float>(), s);
+    KCHECK("cand-topk");
     HIP_CHECK(hipEventRecord(ix->perf.ev[5], s));
     ix->w_probe_lists.ensure((size_t)nq * probe * 4);
     ix->w_probe_offs.ensure(probe_offs.size() * 8);
@@ -456,6 +473,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                   ix->d_id_by_slot, (int)probe, (int)nq, (int)limit, do_sqrt,
                   inv_mul2, ix->w_out_ids.as<int64_t>(),
                   ix->w_out_dists.as<float>(), s);
+    KCHECK("gather");
     HIP_CHECK(hipMemcpyAsync(res->neighbors.data(), ix->w_out_ids.ptr,
                              (size_t)nq * limit * 8, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipMemcpyAsync(res->distances.data(), ix->w_out_dists.ptr,
