@@ -171,7 +171,12 @@ def test_duplicate_vectors_ties():
     ix.set_assignments(assign.astype(np.int32))
     ix.build()
     ids, dists = ix.search(q, 20, 4)
-    np.testing.assert_allclose(dists, ref_dists, rtol=0, atol=0)
+    # accumulation order differs (oracle 8-wide pairwise vs GPU float4
+    # sequential) so distances agree to 1e-5 relative, not bitwise; the tie
+    # GROUPS must still line up position by position.
+    np.testing.assert_allclose(dists, ref_dists, rtol=1e-5, atol=1e-6)
+    for row_g, row_o in zip(dists, ref_dists):
+        np.testing.assert_array_equal(np.diff(row_g) == 0, np.diff(row_o) == 0)
     ix.close()
 
 
