@@ -42,6 +42,35 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
+def make_mixture(rows, dim, world, rank, device):
+    """Synthetic embedding-like corpus: a seeded Gaussian MIXTURE (shared
+    component centers across ranks; per-rank point draws). Pure i.i.d.
+    Gaussian vectors have no neighbor structure in d=768 — no IVF index (the
+    reference's included) reaches recall 0.95 at a 0.78% probe fraction on
+    them — so the workload mirrors real embedding clustering (the blog's
+    wiki_all regime) with data = center[c] + 0.6*noise.
+    Queries are held-out draws from the same mixture, identical on all ranks.
+    """
+    ncomp = min(131072, max(256, rows * world // 100))
+    gc = torch.Generator(device="cpu").manual_seed(777)
+    centers = torch.randn(ncomp, dim, generator=gc).to(device)
+    gp = torch.Generator(device="cpu").manual_seed(1000 + rank)
+    comp = torch.randint(0, ncomp, (rows,), generator=gp).to(device)
+    data = centers[comp]
+    del comp
+    gn = torch.Generator(device=device)
+    gn.manual_seed(2000 + rank)
+    for c0 in range(0, rows, 1 << 20):
+        blk = data[c0:c0 + (1 << 20)]
+        blk += 0.6 * torch.randn(blk.shape, generator=gn, device=device)
+    gq = torch.Generator(device="cpu").manual_seed(4242)
+    qcomp = torch.randint(0, ncomp, (10000,), generator=gq)
+    qnoise = 0.6 * torch.randn(10000, dim, generator=gq)
+    queries_host = centers[qcomp.to(device)].cpu() + qnoise
+    del centers
+    return data, queries_host
+
+
 def kmeans_torch(data_gpu: torch.Tensor, nlist: int, iters: int, seed: int):
     """Harness-side k-means (index BUILD — out of the judged search path,
     SURVEY.md §2 kmeans row). f32 matmul Lloyd on the GPU."""
@@ -109,19 +138,25 @@ def run_cpu_baseline(host_vecs: np.ndarray, cent: np.ndarray,
     + top-k, OpenMP over queries) on a bounded sample of the same workload."""
     from oracle import oracle as orc
     idx = orc.IvfIndex(cent, host_vecs, assign.astype(np.int64))
-    qs = queries[:sample_q]
-    idx.search(orc.METRIC_L2SQ, qs[:2], nprobe, k)  # warm
-    t0 = time.perf_counter()
-    idx.search(orc.METRIC_L2SQ, qs, nprobe, k)
-    dt = time.perf_counter() - t0
+    idx.search(orc.METRIC_L2SQ, queries[:2], nprobe, k)  # warm
+    # grow the sample until >= ~8 s of CPU wall (bounded at 1024 queries)
+    nq, dt = sample_q, 0.0
+    while True:
+        t0 = time.perf_counter()
+        idx.search(orc.METRIC_L2SQ, queries[:nq], nprobe, k)
+        dt = time.perf_counter() - t0
+        if dt >= 8.0 or nq >= min(1024, queries.shape[0]):
+            break
+        nq = min(min(1024, queries.shape[0]),
+                 max(nq * 2, int(nq * 10.0 / max(dt, 0.1))))
     cores = orc.lib().moann_oracle_num_threads()
     return {
-        "value": round(sample_q / dt, 3),
+        "value": round(nq / dt, 3),
         "unit": "queries/s",
         "cores": int(cores),
         "kind": "port",
-        "sample": f"{sample_q} queries of the same 10M-shard probe "
-                  f"(nprobe={nprobe}, k={k}), {dt:.1f}s wall",
+        "sample": f"{nq} queries of the same per-shard probe "
+                  f"(nprobe={nprobe}, k={k}), {dt:.2f}s wall",
     }
 
 
@@ -157,17 +192,13 @@ def main():
 
     # ---- data + build (per-rank shard; seed differs per rank) -------------
     t_build0 = time.perf_counter()
-    torch.manual_seed(42 + rank)
-    log(f"[bench] generating {args.rows}x{args.dim} f32 shard on GPU ...")
-    data = torch.randn(args.rows, args.dim, device=dev, dtype=torch.float32)
+    log(f"[bench] generating {args.rows}x{args.dim} f32 mixture shard on GPU ...")
+    data, queries_host = make_mixture(args.rows, args.dim, world, rank, dev)
     log("[bench] k-means (harness-side build) ...")
     cent = kmeans_torch(data, args.nlist, iters=8, seed=123 + rank)
     assign = assign_torch(data, cent)
-    # queries: held-out, same on every rank
-    qg = torch.Generator(device="cpu").manual_seed(4242)
-    queries_host = torch.randn(10000, args.dim, generator=qg,
-                               dtype=torch.float32)
     queries = queries_host.to(dev)
+    torch.cuda.synchronize()
 
     log("[bench] ingest + pack into moann index ...")
     host_vecs = data.cpu().numpy()  # host copy (oracle baseline + ingest)

@@ -202,7 +202,7 @@ typedef struct {
     /* list-scan kernel (the HBM-roofline kernel) */
     uint64_t scan_launches;
     double scan_ms;              /* HIP-event time on the index stream        */
-    uint64_t scan_bytes;         /* ALGORITHMIC bytes: rows_scanned * dim * 4 */
+    uint64_t scan_bytes;         /* COMPULSORY bytes: distinct probed lists * dim * 4 (each byte once per launch; candidate multiplicity is scan_rows * dim * 4) */
     uint64_t scan_rows;          /* candidate rows scanned (excl. padding)    */
     /* centroid-rank kernel (the MFMA/GEMM-stage counter) */
     uint64_t rank_launches;

@@ -488,9 +488,12 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     ix->perf.c.rank_ms += ms;
     ix->perf.c.rank_flops += 2ull * nq * ix->nlist * ix->dim;
     HIP_CHECK(hipEventElapsedTime(&ms, ix->perf.ev[2], ix->perf.ev[3]));
+    uint64_t unique_rows = 0;
+    for (uint32_t l = 0; l < ix->nlist; ++l)
+        if (lcount[l]) unique_rows += (uint64_t)ix->list_rows[l];
     ix->perf.c.scan_launches++;
     ix->perf.c.scan_ms += ms;
-    ix->perf.c.scan_bytes += (uint64_t)total_cand * ix->dim * 4;
+    ix->perf.c.scan_bytes += unique_rows * ix->dim * 4;
     ix->perf.c.scan_rows += (uint64_t)total_cand;
     HIP_CHECK(hipEventElapsedTime(&ms, ix->perf.ev[4], ix->perf.ev[5]));
     ix->perf.c.select_ms += ms;
