@@ -216,6 +216,55 @@ typedef struct {
 void moann_ivf_flat_perf(gpu_ivf_flat_c index_c, moann_perf_t* out);
 void moann_ivf_flat_perf_reset(gpu_ivf_flat_c index_c);
 
+/* --- HNSW (pkg/vectorindex/hnsw; usearch graph ingested to HBM) ----------- */
+
+/* The GPU replacement for the usearch traversal behind HnswSearch
+ * (pkg/vectorindex/hnsw/model.go:881-901 SearchUnsafe -> usearch_search,
+ * tarball c/usearch.h:390). The graph is ingested from usearch's own
+ * structures (entry slot, per-node levels, per-level adjacency — exported
+ * host-side); distances are returned in USEARCH conventions
+ * (l2sq; ip = 1 - a.b; cos = 1 - cossim) and the host applies
+ * DistanceTransformHnsw (metric/types.go:234-243) exactly like the Go layer.
+ *
+ * usearch_metric: 0 = l2sq, 1 = ip, 2 = cos
+ * (usearch metric_kind subset per OpTypeToUsearchMetric, types.go:152-163) */
+
+typedef void* moann_hnsw_c;
+
+moann_hnsw_c moann_hnsw_new(uint32_t dimension, int usearch_metric,
+                            uint64_t n_nodes, int64_t entry_slot,
+                            int32_t max_level, const int* devices,
+                            int device_count, void* errmsg);
+/* vectors by SLOT order, f32 [n][dim] host pointer */
+void moann_hnsw_set_vectors(moann_hnsw_c h, const float* vecs, void* errmsg);
+/* slot -> key map (usearch keys = MO pks), int64 [n] */
+void moann_hnsw_set_keys(moann_hnsw_c h, const int64_t* keys, void* errmsg);
+/* base-level adjacency CSR: offs [n+1], nbrs (slot ids) */
+void moann_hnsw_set_level0(moann_hnsw_c h, const int64_t* offs,
+                           const uint32_t* nbrs, int64_t nnz, void* errmsg);
+/* upper levels: u_idx [n] (slot -> dense upper index, -1 below level 1);
+ * up_offs [max_level * (u_count+1)] absolute indices into up_nbrs;
+ * level l (1-based) occupies rows [(l-1)*(u_count+1), l*(u_count+1)) */
+void moann_hnsw_set_upper(moann_hnsw_c h, const int32_t* u_idx,
+                          int64_t u_count, const int32_t* up_offs,
+                          const uint32_t* up_nbrs, int64_t up_nnz,
+                          void* errmsg);
+void moann_hnsw_build(moann_hnsw_c h, void* errmsg);
+
+/* ef = usearch expansion_search (MO HnswParam.EfSearch,
+ * hnsw_search.go:113-254). Results: dense ascending [nq][k], (-1, FLT_MAX)
+ * padded. */
+void moann_hnsw_search(moann_hnsw_c h, const float* queries, uint64_t nq,
+                       uint32_t dim, uint32_t ef, uint32_t k,
+                       int64_t* out_keys, float* out_dists, void* errmsg);
+void moann_hnsw_search_device(moann_hnsw_c h, const void* queries_dev,
+                              uint64_t nq, uint32_t dim, uint32_t ef,
+                              uint32_t k, int64_t* out_keys, float* out_dists,
+                              void* errmsg);
+void moann_hnsw_perf(moann_hnsw_c h, moann_perf_t* out);
+void moann_hnsw_perf_reset(moann_hnsw_c h);
+void moann_hnsw_destroy(moann_hnsw_c h, void* errmsg);
+
 /* --- misc ----------------------------------------------------------------- */
 
 int moann_device_count(void);            /* hipGetDeviceCount, 0 on error */

@@ -58,7 +58,18 @@ __global__ __launch_bounds__(256) void scan_kernel(
     float* ldsq = lds;
     float* ldsn = lds + QT * dpad;
 
-    const int j = blockIdx.x;
+    /* XCD-aware bijective blockIdx remap (T1): the dispatcher places block
+     * b on XCD b%8; this permutation gives each XCD a CONTIGUOUS job range,
+     * so the query tiles of one list (consecutive jobs) stream through one
+     * XCD's L2 instead of re-fetching from HBM on every XCD. */
+    int j;
+    {
+        const int nwg = gridDim.x, bid = blockIdx.x;
+        const int q8 = nwg >> 3, r8 = nwg & 7;
+        const int xcd = bid & 7, idx = bid >> 3;
+        j = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+        if (j >= nwg) j = bid; /* safety: cannot happen for bijective map */
+    }
     const int nq = j_nq[j];
     const int qbase = j_qbase[j];
 
@@ -184,6 +195,10 @@ static void launch_scan_qt(int qt, const float* packed, const float* queries,
     jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase, \
     dists_out
     switch (qt) {
+    case 16:
+        hipLaunchKernelGGL((scan_kernel<METRIC, 16>), grid, block,
+                           (16 * dpad + 16) * sizeof(float), stream, ARGS);
+        break;
     case 8:
         hipLaunchKernelGGL((scan_kernel<METRIC, 8>), grid, block,
                            (8 * dpad + 8) * sizeof(float), stream, ARGS);

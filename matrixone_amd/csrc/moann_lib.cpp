@@ -156,9 +156,12 @@ struct IvfIndex {
 };
 
 int pick_qt(uint32_t dpad) {
-    /* QT query rows in LDS; keep the tile <= 64 KiB so >=2 blocks/CU fit */
-    for (int qt : {8, 4, 2, 1})
-        if ((size_t)qt * dpad * 4 + qt * 4 <= 64 * 1024) return qt;
+    /* QT query rows in LDS; <= 52 KiB keeps >=3 blocks/CU resident.
+     * QT=16 halves list re-reads when many queries probe one list (the
+     * hub-list multiplicity); VALU ceiling at QT=16/d768 is ~9.8 TB/s,
+     * still above the HBM bound. */
+    for (int qt : {16, 8, 4, 2, 1})
+        if ((size_t)qt * dpad * 4 + qt * 4 <= 52 * 1024) return qt;
     return 1;
 }
 

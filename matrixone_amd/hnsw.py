@@ -1,0 +1,224 @@
+"""HnswSearch — the MatrixOne-side host mirror of the HNSW search
+implementation (pkg/vectorindex/hnsw), backed by the MI355X kernel instead of
+cgo → usearch.
+
+Mirrors pkg/vectorindex/hnsw/search.go semantics: HnswParam
+{M, efConstruction, efSearch, op_type} (hnsw_search.go:113-254), per-model
+search + merge (search.go:89-183; multi-model handling is the embedding
+host's loop — this class serves one model), and the usearch→MO score
+transform DistanceTransformHnsw (metric/types.go:234-243: IP = usearch − 1,
+L2 = sqrt of l2sq when the query's function is l2_distance).
+
+The GPU index ingests a usearch-exported graph (entry slot, levels,
+adjacency, vectors by slot) — the shape oracle/hnsw.py's export produces.
+"""
+
+from __future__ import annotations
+
+import ctypes as ct
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+
+from . import engine
+
+# usearch metric kinds accepted by the C layer (include/moann.h HNSW section)
+USEARCH_METRICS = {"l2sq": 0, "ip": 1, "cos": 2}
+
+# op_type -> usearch metric (OpTypeToUsearchMetric, metric/types.go:152-163)
+OP_TYPE_USEARCH = {
+    "vector_l2_ops": "l2sq",
+    "vector_l2sq_ops": "l2sq",
+    "vector_ip_ops": "ip",
+    "vector_cosine_ops": "cos",
+}
+
+
+def _bind(L):
+    if getattr(L, "_hnsw_bound", False):
+        return L
+    L.moann_hnsw_new.restype = ct.c_void_p
+    L.moann_hnsw_new.argtypes = [ct.c_uint32, ct.c_int, ct.c_uint64,
+                                 ct.c_int64, ct.c_int32, ct.POINTER(ct.c_int),
+                                 ct.c_int, ct.c_void_p]
+    L.moann_hnsw_set_vectors.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_void_p]
+    L.moann_hnsw_set_keys.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_void_p]
+    L.moann_hnsw_set_level0.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_void_p,
+                                        ct.c_int64, ct.c_void_p]
+    L.moann_hnsw_set_upper.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_int64,
+                                       ct.c_void_p, ct.c_void_p, ct.c_int64,
+                                       ct.c_void_p]
+    L.moann_hnsw_build.argtypes = [ct.c_void_p, ct.c_void_p]
+    L.moann_hnsw_search.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_uint64,
+                                    ct.c_uint32, ct.c_uint32, ct.c_uint32,
+                                    ct.c_void_p, ct.c_void_p, ct.c_void_p]
+    L.moann_hnsw_search_device.argtypes = L.moann_hnsw_search.argtypes
+    L.moann_hnsw_perf.argtypes = [ct.c_void_p, ct.POINTER(engine.PerfT)]
+    L.moann_hnsw_perf_reset.argtypes = [ct.c_void_p]
+    L.moann_hnsw_destroy.argtypes = [ct.c_void_p, ct.c_void_p]
+    L._hnsw_bound = True
+    return L
+
+
+@dataclass
+class HnswGraphData:
+    """A usearch-exported graph (the arrays oracle/hnsw.py exports, or the
+    embedding host's equivalent ingestion of a model file)."""
+    dim: int
+    metric: str          # "l2sq" | "ip" | "cos" (usearch kind)
+    entry_slot: int
+    max_level: int
+    levels: np.ndarray   # [n] int32
+    keys: np.ndarray     # [n] uint64/int64 slot -> key (MO pk)
+    vecs: np.ndarray     # [n][dim] f32 by slot
+    adj: list            # per level: (offsets[n+1] int64, neighbors u32)
+
+
+class HnswIndex:
+    """One HNSW model on one MI355X (ingested usearch graph)."""
+
+    def __init__(self, g: HnswGraphData, device: int = 0):
+        L = _bind(engine.lib())
+        self.dim = g.dim
+        self.metric = g.metric
+        n = g.vecs.shape[0]
+        err = engine._Err()
+        dev = (ct.c_int * 1)(device)
+        self._h = L.moann_hnsw_new(g.dim, USEARCH_METRICS[g.metric], n,
+                                   g.entry_slot, g.max_level, dev, 1, err.ref)
+        err.check("hnsw_new")
+        if not self._h:
+            raise engine.MoannError("moann_hnsw_new returned NULL")
+        vecs = np.ascontiguousarray(g.vecs, dtype=np.float32)
+        L.moann_hnsw_set_vectors(self._h, vecs.ctypes.data, err.ref)
+        err.check("set_vectors")
+        keys = np.ascontiguousarray(g.keys.astype(np.int64))
+        L.moann_hnsw_set_keys(self._h, keys.ctypes.data, err.ref)
+        err.check("set_keys")
+        offs0, nbrs0 = g.adj[0]
+        offs0 = np.ascontiguousarray(offs0, dtype=np.int64)
+        nbrs0 = np.ascontiguousarray(nbrs0, dtype=np.uint32)
+        L.moann_hnsw_set_level0(self._h, offs0.ctypes.data, nbrs0.ctypes.data,
+                                nbrs0.size, err.ref)
+        err.check("set_level0")
+        if g.max_level > 0:
+            u_mask = g.levels >= 1
+            u_slots = np.nonzero(u_mask)[0]
+            u_count = len(u_slots)
+            u_idx = np.full(n, -1, dtype=np.int32)
+            u_idx[u_slots] = np.arange(u_count, dtype=np.int32)
+            up_offs = np.zeros((g.max_level, u_count + 1), dtype=np.int64)
+            nbr_chunks = []
+            base = 0
+            for lvl in range(1, g.max_level + 1):
+                offs, nbrs = g.adj[lvl]
+                # offsets restricted to upper nodes, rebased into the concat
+                cnt = (offs[u_slots + 1] - offs[u_slots]).astype(np.int64)
+                up_offs[lvl - 1, 1:] = np.cumsum(cnt)
+                up_offs[lvl - 1, :] += base
+                for s in u_slots:
+                    if offs[s + 1] > offs[s]:
+                        nbr_chunks.append(nbrs[offs[s]:offs[s + 1]])
+                base = up_offs[lvl - 1, -1]
+            up_nbrs = (np.concatenate(nbr_chunks).astype(np.uint32)
+                       if nbr_chunks else np.empty(0, dtype=np.uint32))
+            up_offs32 = np.ascontiguousarray(up_offs.reshape(-1),
+                                             dtype=np.int32)
+            u_idx = np.ascontiguousarray(u_idx)
+            up_nbrs = np.ascontiguousarray(up_nbrs)
+            L.moann_hnsw_set_upper(self._h, u_idx.ctypes.data, u_count,
+                                   up_offs32.ctypes.data, up_nbrs.ctypes.data,
+                                   up_nbrs.size, err.ref)
+            err.check("set_upper")
+        L.moann_hnsw_build(self._h, err.ref)
+        err.check("hnsw_build")
+
+    def search(self, queries, ef: int, k: int):
+        """Raw usearch-convention results (ids by key, dists)."""
+        L = _bind(engine.lib())
+        queries = np.ascontiguousarray(queries, dtype=np.float32)
+        nq = queries.shape[0]
+        out_ids = np.empty((nq, k), dtype=np.int64)
+        out_dists = np.empty((nq, k), dtype=np.float32)
+        err = engine._Err()
+        L.moann_hnsw_search(self._h, queries.ctypes.data, nq,
+                            queries.shape[1], ef, k, out_ids.ctypes.data,
+                            out_dists.ctypes.data, err.ref)
+        err.check("hnsw_search")
+        return out_ids, out_dists
+
+    def search_device(self, queries_tensor, ef: int, k: int):
+        L = _bind(engine.lib())
+        nq, qdim = queries_tensor.shape
+        out_ids = np.empty((nq, k), dtype=np.int64)
+        out_dists = np.empty((nq, k), dtype=np.float32)
+        err = engine._Err()
+        L.moann_hnsw_search_device(self._h,
+                                   ct.c_void_p(queries_tensor.data_ptr()),
+                                   nq, qdim, ef, k, out_ids.ctypes.data,
+                                   out_dists.ctypes.data, err.ref)
+        err.check("hnsw_search_device")
+        return out_ids, out_dists
+
+    def perf(self) -> dict:
+        p = engine.PerfT()
+        _bind(engine.lib()).moann_hnsw_perf(self._h, ct.byref(p))
+        return {f[0]: getattr(p, f[0]) for f in engine.PerfT._fields_}
+
+    def perf_reset(self):
+        _bind(engine.lib()).moann_hnsw_perf_reset(self._h)
+
+    def close(self):
+        if getattr(self, "_h", None):
+            err = engine._Err()
+            _bind(engine.lib()).moann_hnsw_destroy(self._h, err.ref)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class HnswSearch:
+    """VectorIndexSearchIf mirror for HNSW (cache.go:65-84), one model.
+    Applies DistanceTransformHnsw like hnsw/search.go:167-182."""
+
+    def __init__(self, graph: HnswGraphData, op_type: str = "vector_l2_ops",
+                 ef_search: int = 64, device: int = 0):
+        self.graph = graph
+        self.op_type = op_type
+        self.ef_search = ef_search
+        self.device = device
+        self.index: Optional[HnswIndex] = None
+
+    def Load(self, proc=None) -> None:
+        self.index = HnswIndex(self.graph, device=self.device)
+
+    def Search(self, proc, query, rt):
+        if self.index is None:
+            raise RuntimeError("index not loaded — call Load() first")
+        q = np.atleast_2d(np.asarray(query, dtype=np.float32))
+        ids, dists = self.index.search(q, self.ef_search, rt.limit)
+        d64 = dists.astype(np.float64)
+        valid = ids >= 0
+        # DistanceTransformHnsw (metric/types.go:234-243)
+        if self.graph.metric == "ip":
+            d64 = np.where(valid, d64 - 1.0, d64)
+        elif (self.graph.metric == "l2sq"
+              and rt.orig_func_name == "l2_distance"):
+            d64 = np.where(valid, np.sqrt(np.maximum(d64, 0.0)), d64)
+        return ids, d64
+
+    def SearchFloat32(self, proc, query, rt, out_keys, out_dists):
+        ids, d64 = self.Search(proc, query, rt)
+        n = ids.size
+        out_keys[:n] = ids.reshape(-1)
+        out_dists[:n] = d64.reshape(-1).astype(np.float32)
+
+    def Destroy(self):
+        if self.index is not None:
+            self.index.close()
+            self.index = None

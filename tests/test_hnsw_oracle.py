@@ -1,0 +1,71 @@
+"""CPU tests: the usearch shim, the graph export, and the numpy restatement
+of usearch's HNSW search must agree with usearch's own results on the same
+index (oracle/hnsw.py module docstring for algorithm cites)."""
+
+import numpy as np
+import pytest
+
+from oracle import hnsw as H
+
+
+def _mk(metric, n=4000, d=48, conn=16, seed=5):
+    rng = np.random.Generator(np.random.PCG64(seed))
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    ix = H.RefHnsw(d, metric=metric, connectivity=conn, expansion_add=64,
+                   expansion_search=32, capacity=n, threads=4)
+    ix.add(vecs)
+    qs = rng.standard_normal((16, d), dtype=np.float32)
+    return ix, qs
+
+
+@pytest.mark.parametrize("metric", [H.METRIC_L2SQ, H.METRIC_IP, H.METRIC_COS])
+def test_oracle_restatement_matches_usearch(metric):
+    ix, qs = _mk(metric)
+    ref_ids, ref_d = ix.search(qs, ef=32, k=10)
+    g = ix.export_graph()
+    assert g.adj[0][1].size > 0
+    pos = 0
+    for i in range(len(qs)):
+        oi, od = H.oracle_search(g, qs[i], ef=32, k=10)
+        pos += int((oi == ref_ids[i]).sum())
+        np.testing.assert_allclose(od, ref_d[i], rtol=2e-5, atol=5e-5)
+    # exact-position parity modulo rare FP tie flips
+    assert pos >= len(qs) * 10 - 2
+
+
+def test_export_graph_structure():
+    ix, _ = _mk(H.METRIC_L2SQ, n=2000)
+    g = ix.export_graph()
+    n = len(g.levels)
+    assert n == 2000
+    assert g.levels[g.entry_slot] == g.max_level
+    # every neighbor slot is valid; level-l adjacency only on nodes >= l
+    for lvl in range(g.max_level + 1):
+        offs, nbrs = g.adj[lvl]
+        assert offs[-1] == nbrs.size
+        if nbrs.size:
+            assert nbrs.max() < n
+        has = (offs[1:] - offs[:-1]) > 0
+        assert not has[g.levels < lvl].any()
+
+
+def test_ref_hnsw_search_recall():
+    """usearch itself reaches high recall on clustered data (sanity that the
+    shim drives it correctly)."""
+    rng = np.random.Generator(np.random.PCG64(9))
+    n, d = 8000, 32
+    centers = rng.standard_normal((80, d), dtype=np.float32)
+    comp = rng.integers(0, 80, n)
+    vecs = centers[comp] + 0.4 * rng.standard_normal((n, d), dtype=np.float32)
+    ix = H.RefHnsw(d, metric=H.METRIC_L2SQ, connectivity=16,
+                   expansion_add=128, expansion_search=64, capacity=n,
+                   threads=4)
+    ix.add(vecs)
+    qc = rng.integers(0, 80, 16)
+    qs = centers[qc] + 0.4 * rng.standard_normal((16, d), dtype=np.float32)
+    ids, _ = ix.search(qs, ef=64, k=10)
+    from oracle import oracle as orc
+    gt, _ = orc.bruteforce(orc.METRIC_L2SQ, vecs, qs, 10)
+    recall = np.mean([len(set(ids[i]).intersection(gt[i])) / 10
+                      for i in range(16)])
+    assert recall > 0.9, recall
