@@ -27,6 +27,8 @@ struct ScanJobs {
     const int32_t* qbase;      /* [njobs] offset into qslot arrays           */
     const int32_t* qslot_query;   /* [nqslots] query index                   */
     const int64_t* qslot_outbase; /* [nqslots] output base in dists buffer   */
+    const int64_t* slot_base;     /* [njobs] global slot of list row 0 (for
+                                     the membership filter; may be null)     */
     int njobs;
 };
 
@@ -38,7 +40,8 @@ struct ScanJobs {
  * dists_out: ragged candidate buffer, written at qslot_outbase + row. */
 void launch_scan(int metric, int qt, const float* packed, const float* queries,
                  const float* qnorms, int dpad, const ScanJobs& jobs,
-                 float* dists_out, hipStream_t stream);
+                 float* dists_out, hipStream_t stream,
+                 const uint32_t* filter_bitset = nullptr);
 
 /* Per-query ascending top-k select (radix select + in-LDS bitonic sort).
  * Candidates for query q live at dists[off(q) .. off(q)+count(q)) where

@@ -53,6 +53,10 @@ __global__ __launch_bounds__(256) void scan_kernel(
     const int32_t* __restrict__ j_qbase,
     const int32_t* __restrict__ qslot_query,
     const int64_t* __restrict__ qslot_outbase,
+    const int64_t* __restrict__ j_slot_base,
+    const uint32_t* __restrict__ filter_bitset, /* bit per GLOBAL slot; a
+        cleared bit excludes the entry (cuVS bitset_filter precedent,
+        cgo/cuvs/ivf_flat.hpp:908-924); null = unfiltered */
     float* __restrict__ dists_out) {
     extern __shared__ float lds[];  /* [QT][dpad] (+[QT] qnorm for cos) */
     float* ldsq = lds;
@@ -109,6 +113,9 @@ __global__ __launch_bounds__(256) void scan_kernel(
         for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0.f;
         float rn0 = 0.f, rn1 = 0.f;
 
+#pragma unroll 2 /* two iterations in flight: 4 dwordx4 issued together,
+                    counted vmcnt waits (hipcc pipelines the pair the same
+                    way it does for the QT=8 instantiation) */
         for (int q4 = 0; q4 < d4; ++q4) {
             const float4 x0 = d0[q4 * 64];
             const float4 x1 = d1[q4 * 64];
@@ -149,6 +156,18 @@ __global__ __launch_bounds__(256) void scan_kernel(
 
         const int row0 = g0 * 64 + lane;
         const int row1 = g1 * 64 + lane;
+        bool pass0 = true, pass1 = true;
+        if (filter_bitset) {
+            const int64_t sb = j_slot_base[j];
+            if (row0 < rows) {
+                const int64_t gs = sb + row0;
+                pass0 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+            if (has1 && row1 < rows) {
+                const int64_t gs = sb + row1;
+                pass1 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+        }
 #pragma unroll
         for (int t = 0; t < QT; ++t) {
             if (t >= nq) break;
@@ -166,7 +185,7 @@ __global__ __launch_bounds__(256) void scan_kernel(
                         dist = (float)(1.0 - sim);
                     }
                 } else dist = acc0[t];
-                dists_out[ob + row0] = dist;
+                dists_out[ob + row0] = pass0 ? dist : FLT_MAX;
             }
             if (has1 && row1 < rows) {
                 float dist;
@@ -180,7 +199,7 @@ __global__ __launch_bounds__(256) void scan_kernel(
                         dist = (float)(1.0 - sim);
                     }
                 } else dist = acc1[t];
-                dists_out[ob + row1] = dist;
+                dists_out[ob + row1] = pass1 ? dist : FLT_MAX;
             }
         }
     }
@@ -189,11 +208,12 @@ __global__ __launch_bounds__(256) void scan_kernel(
 template <int METRIC>
 static void launch_scan_qt(int qt, const float* packed, const float* queries,
                            const float* qnorms, int dpad, const ScanJobs& jb,
-                           float* dists_out, hipStream_t stream) {
+                           float* dists_out, hipStream_t stream,
+                           const uint32_t* filter_bitset) {
     const dim3 grid(jb.njobs), block(256);
 #define ARGS packed, queries, qnorms, dpad, jb.databaseg, jb.gstart, \
     jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase, \
-    dists_out
+    jb.slot_base, filter_bitset, dists_out
     switch (qt) {
     case 16:
         hipLaunchKernelGGL((scan_kernel<METRIC, 16>), grid, block,
@@ -221,13 +241,14 @@ static void launch_scan_qt(int qt, const float* packed, const float* queries,
 
 void launch_scan(int metric, int qt, const float* packed, const float* queries,
                  const float* qnorms, int dpad, const ScanJobs& jb,
-                 float* dists_out, hipStream_t stream) {
+                 float* dists_out, hipStream_t stream,
+                 const uint32_t* filter_bitset) {
     if (jb.njobs == 0) return;
     switch (metric) {
-    case KM_L2SQ: launch_scan_qt<KM_L2SQ>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
-    case KM_IP:   launch_scan_qt<KM_IP>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
-    case KM_COS:  launch_scan_qt<KM_COS>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
-    default:      launch_scan_qt<KM_L1>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream); break;
+    case KM_L2SQ: launch_scan_qt<KM_L2SQ>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_IP:   launch_scan_qt<KM_IP>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_COS:  launch_scan_qt<KM_COS>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    default:      launch_scan_qt<KM_L1>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
     }
 }
 
@@ -379,7 +400,10 @@ __global__ void gather_kernel(
     if (idx >= (int64_t)nq * k) return;
     const int q = (int)(idx / k);
     const int32_t slot = sel_slots[idx];
-    if (slot < 0) { out_ids[idx] = -1; out_dists[idx] = FLT_MAX; return; }
+    /* FLT_MAX marks padding AND membership-filtered rows */
+    if (slot < 0 || sel_dists[idx] == FLT_MAX) {
+        out_ids[idx] = -1; out_dists[idx] = FLT_MAX; return;
+    }
     /* binary search rank r: probe_offs[q][r] <= slot < probe_offs[q][r+1] */
     const int64_t* offs = probe_offs + (int64_t)q * (nprobe + 1);
     int lo = 0, hi = nprobe - 1;

@@ -15,8 +15,10 @@
 #include <algorithm>
 #include <cfloat>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <map>
+#include <unordered_map>
 #include <memory>
 #include <mutex>
 #include <numeric>
@@ -138,6 +140,9 @@ struct IvfIndex {
     DevBuf w_queries, w_qnorms, w_cdists, w_csel_slots, w_csel_dists;
     DevBuf w_jobs, w_qslots, w_cand, w_sel_slots, w_sel_dists;
     DevBuf w_probe_lists, w_probe_offs, w_qoffs, w_out_ids, w_out_dists;
+    DevBuf w_filter;
+    std::vector<int64_t> h_id_by_slot;  /* kept for the membership filter */
+    std::unordered_map<int64_t, int64_t> id_to_slot; /* lazy */
     std::vector<uint8_t> h_jobstage; /* pinned-ish host staging */
 
     std::mutex mu;
@@ -156,6 +161,11 @@ struct IvfIndex {
 };
 
 int pick_qt(uint32_t dpad) {
+    /* MOANN_QT overrides for A/B tuning */
+    if (const char* e = getenv("MOANN_QT")) {
+        const int v = atoi(e);
+        if (v == 1 || v == 2 || v == 4 || v == 8 || v == 16) return v;
+    }
     /* QT query rows in LDS; <= 52 KiB keeps >=3 blocks/CU resident.
      * QT=16 halves list re-reads when many queries probe one list (the
      * hub-list multiplicity); VALU ceiling at QT=16/d768 is ~9.8 TB/s,
@@ -222,7 +232,8 @@ void pack_lists(IvfIndex* ix, const float* d_rows, int64_t nrows,
 
 std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                                    bool queries_on_device, uint64_t nq,
-                                   uint32_t limit, uint32_t n_probes) {
+                                   uint32_t limit, uint32_t n_probes,
+                                   const uint32_t* filter_words = nullptr) {
     std::lock_guard<std::mutex> lk(ix->mu);
     HIP_CHECK(hipSetDevice(ix->device));
     auto res = std::make_unique<Result>();
@@ -268,9 +279,10 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         /* tile queries, chunk centroid groups to fill the chip */
         const int ntiles = (int)((nq + qt - 1) / qt);
         const int64_t cgroups = ix->cent_groups;
-        /* aim for ~4096 workgroups total to fill 256 CUs */
+        /* target ~1024 jobs: enough to fill 256 CUs without paying the
+         * 48 KiB query-tile LDS fill per tiny single-group job */
         int64_t want_chunks = std::min<int64_t>(
-            cgroups, std::max<int64_t>(1, (4096 + ntiles - 1) / ntiles));
+            cgroups, std::max<int64_t>(1, (1024 + ntiles - 1) / ntiles));
         const int chunk = (int)((cgroups + want_chunks - 1) / want_chunks);
         const int nchunk = (int)((cgroups + chunk - 1) / chunk);
         const int njobs = ntiles * nchunk;
@@ -318,6 +330,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         jb.qbase = (int32_t*)up(qbase.data(), njobs * b32);
         jb.qslot_query = (int32_t*)up(qslot_query.data(), nqslots * b32);
         jb.qslot_outbase = (int64_t*)up(qslot_outbase.data(), nqslots * b64);
+        jb.slot_base = nullptr;
         jb.njobs = njobs;
         HIP_CHECK(hipEventRecord(ix->perf.ev[0], s));
         launch_scan(ix->kmetric, qt, ix->d_cent_packed, d_q, d_qn, dpad, jb,
@@ -390,7 +403,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     if (tiles_total > 0 && tiles_total < 2048)
         split = (int)std::min<int64_t>(64, (2048 + tiles_total - 1) / tiles_total);
 
-    std::vector<int64_t> databaseg, qslot_outbase;
+    std::vector<int64_t> databaseg, qslot_outbase, jslot_base;
     std::vector<int32_t> gstart, gcount, rows, jnq, qbase, qslot_query;
     for (uint32_t l = 0; l < ix->nlist; ++l) {
         if (!lcount[l]) continue;
@@ -405,6 +418,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                 qslot_outbase.push_back(pair_ob[pi]);
             }
             for (int64_t g0 = 0; g0 < lg; g0 += chunk) {
+                jslot_base.push_back(ix->list_slot_base[l]);
                 databaseg.push_back(ix->list_gbase[l]);
                 gstart.push_back((int32_t)g0);
                 gcount.push_back((int32_t)std::min<int64_t>(chunk, lg - g0));
@@ -418,11 +432,19 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const int nqslots = (int)qslot_query.size();
 
     /* 4. list scan into the ragged candidate buffer */
+    const uint32_t* d_filter = nullptr;
+    if (filter_words) {
+        const size_t fw = (size_t)((ix->count + 31) / 32);
+        ix->w_filter.ensure(fw * 4);
+        HIP_CHECK(hipMemcpyAsync(ix->w_filter.ptr, filter_words, fw * 4,
+                                 hipMemcpyHostToDevice, s));
+        d_filter = ix->w_filter.as<uint32_t>();
+    }
     ix->w_cand.ensure(std::max<int64_t>(1, total_cand) * 4);
     {
         const size_t b64 = sizeof(int64_t), b32 = sizeof(int32_t);
         auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
-        size_t need = al8((size_t)njobs * b64) + 5 * al8((size_t)njobs * b32) +
+        size_t need = 2 * al8((size_t)njobs * b64) + 5 * al8((size_t)njobs * b32) +
                       al8((size_t)nqslots * b32) + al8((size_t)nqslots * b64) + 64;
         ix->w_qslots.ensure(need);
         uint8_t* p = ix->w_qslots.as<uint8_t>();
@@ -442,10 +464,11 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         jb.qbase = (int32_t*)up(qbase.data(), njobs * b32);
         jb.qslot_query = (int32_t*)up(qslot_query.data(), nqslots * b32);
         jb.qslot_outbase = (int64_t*)up(qslot_outbase.data(), nqslots * b64);
+        jb.slot_base = (int64_t*)up(jslot_base.data(), njobs * b64);
         jb.njobs = njobs;
         HIP_CHECK(hipEventRecord(ix->perf.ev[2], s));
         launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
-                    ix->w_cand.as<float>(), s);
+                    ix->w_cand.as<float>(), s, d_filter);
         KCHECK("list-scan");
         HIP_CHECK(hipEventRecord(ix->perf.ev[3], s));
     }
@@ -639,6 +662,7 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
         HIP_CHECK(hipMalloc(&ix->d_id_by_slot, id_by_slot.size() * 8));
         HIP_CHECK(hipMemcpy(ix->d_id_by_slot, id_by_slot.data(),
                             id_by_slot.size() * 8, hipMemcpyHostToDevice));
+        ix->h_id_by_slot = std::move(id_by_slot);
         HIP_CHECK(hipMalloc(&ix->d_list_slot_base, ix->nlist * 8));
         HIP_CHECK(hipMemcpy(ix->d_list_slot_base, ix->list_slot_base.data(),
                             ix->nlist * 8, hipMemcpyHostToDevice));
@@ -832,6 +856,55 @@ void moann_brute_force_search(const float* dataset, uint64_t count,
         set_errmsg(errmsg, "moann_brute_force_search", e.what());
     }
     if (inner) free(inner);
+}
+
+/* Membership-filtered search: slot_bitset has one bit per
+ * stored entry in SLOT order (moann_ivf_flat_slots_of_ids maps ids); a
+ * cleared bit excludes the entry before top-k, mirroring the cuVS device
+ * bitset filter (cgo/cuvs/ivf_flat.hpp:908-924) that backs MO's docfilter
+ * membership kinds on the GPU path. */
+gpu_ivf_flat_search_res_t moann_ivf_flat_search_filtered(
+    gpu_ivf_flat_c h, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit, ivf_flat_search_params_t sp,
+    const uint32_t* slot_bitset, uint64_t nbits, void* errmsg) {
+    gpu_ivf_flat_search_res_t out {nullptr};
+    try {
+        auto ix = IX(h);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        if (nbits < ix->count)
+            throw std::runtime_error("filter bitset smaller than index");
+        out.result_ptr = run_search(ix, (const float*)queries_data, false,
+                                    num_queries, limit, sp.n_probes,
+                                    slot_bitset)
+                             .release();
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_search_filtered", e.what());
+    }
+    return out;
+}
+
+/* id -> global slot lookup for building filter bitsets host-side.
+ * out_slots[i] = -1 for unknown ids. */
+void moann_ivf_flat_slots_of_ids(gpu_ivf_flat_c h, const int64_t* ids,
+                                 uint64_t n, int64_t* out_slots,
+                                 void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (!ix->built) throw std::runtime_error("index not built");
+        if (ix->id_to_slot.empty() && ix->count) {
+            ix->id_to_slot.reserve(ix->count * 2);
+            for (uint64_t sidx = 0; sidx < ix->count; ++sidx)
+                ix->id_to_slot.emplace(ix->h_id_by_slot[sidx], (int64_t)sidx);
+        }
+        for (uint64_t i = 0; i < n; ++i) {
+            auto it = ix->id_to_slot.find(ids[i]);
+            out_slots[i] = it == ix->id_to_slot.end() ? -1 : it->second;
+        }
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_slots_of_ids", e.what());
+    }
 }
 
 void moann_ivf_flat_perf(gpu_ivf_flat_c h, moann_perf_t* out) {
