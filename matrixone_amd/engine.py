@@ -95,6 +95,12 @@ def lib() -> ct.CDLL:
         f.argtypes = [ct.c_void_p]
     L.gpu_ivf_flat_get_n_list.restype = ct.c_uint32
     L.gpu_ivf_flat_get_n_list.argtypes = [ct.c_void_p]
+    L.moann_ivf_flat_search_filtered.restype = SearchRes
+    L.moann_ivf_flat_search_filtered.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
+        SearchParams, ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    L.moann_ivf_flat_slots_of_ids.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
     L.moann_brute_force_search.argtypes = [
         ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, ct.c_void_p,
         ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, ct.c_void_p,
@@ -215,6 +221,39 @@ class IvfFlatIndex:
                                         SearchParams(nprobe), err.ref)
         err.check("search")
         return self._read_result(res, nq, k, "search")
+
+    def slots_of_ids(self, ids):
+        ids = np.ascontiguousarray(ids, dtype=np.int64)
+        out = np.empty(ids.size, dtype=np.int64)
+        err = _Err()
+        lib().moann_ivf_flat_slots_of_ids(self._h, ids.ctypes.data, ids.size,
+                                          out.ctypes.data, err.ref)
+        err.check("slots_of_ids")
+        return out
+
+    def filter_bitset(self, allowed_ids):
+        """Bitset over stored slots with bits set for allowed_ids — the
+        host-side filter build (cuVS eval_filter_bitmap_cpu shape)."""
+        n = len(self)
+        words = np.zeros((n + 31) // 32, dtype=np.uint32)
+        slots = self.slots_of_ids(allowed_ids)
+        slots = slots[slots >= 0]
+        np.bitwise_or.at(words, slots // 32,
+                         (np.uint32(1) << (slots % 32).astype(np.uint32)))
+        return words
+
+    def search_filtered(self, queries, k: int, nprobe: int, bitset_words):
+        """Membership-filtered search."""
+        queries = _as_f32(queries)
+        bitset_words = np.ascontiguousarray(bitset_words, dtype=np.uint32)
+        nq = queries.shape[0]
+        err = _Err()
+        res = lib().moann_ivf_flat_search_filtered(
+            self._h, queries.ctypes.data, nq, queries.shape[1], k,
+            SearchParams(nprobe), bitset_words.ctypes.data,
+            bitset_words.size * 32, err.ref)
+        err.check("search_filtered")
+        return self._read_result(res, nq, k, "search_filtered")
 
     def search_device(self, queries_tensor, k: int, nprobe: int):
         """queries_tensor: a contiguous f32 torch CUDA tensor [nq][dim]; the

@@ -184,3 +184,45 @@ def test_hnsw_placeholder_absent():
     """HNSW GPU path is SURVEY.md §8 row a3 — lands in a later commit of this
     round / round 2; this marker keeps the scope honest in CI output."""
     pytest.skip("HNSW GPU probe not yet implemented (§8a3, planned)")
+
+
+def test_filtered_search():
+    """Membership-filtered search (§8a4): results equal the oracle's full
+    candidate ranking post-filtered to the allowed id set."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(500))
+    n, d, nlist, nq, probe, k = 8000, 48, 16, 12, 4, 15
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    ids64 = rng.permutation(n).astype(np.int64) * 3 + 11
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    allowed = ids64[rng.random(n) < 0.2]
+
+    # oracle expectation: rank ALL candidates (k = n), keep allowed, truncate
+    idx = orc.IvfIndex(cents, vecs, assign, ids=ids64)
+    all_ids, all_d = idx.search(orc.METRIC_L2SQ, queries, probe, 4000)
+    aset = set(allowed.tolist())
+    exp_ids = np.full((nq, k), -1, dtype=np.int64)
+    exp_d = np.full((nq, k), np.finfo(np.float32).max, dtype=np.float32)
+    for i in range(nq):
+        keep = [(all_d[i, j], all_ids[i, j]) for j in range(all_ids.shape[1])
+                if all_ids[i, j] >= 0 and all_ids[i, j] in aset]
+        for j, (dd, ii) in enumerate(keep[:k]):
+            exp_ids[i, j] = ii
+            exp_d[i, j] = dd
+
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs, ids=ids64)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    words = ix.filter_bitset(allowed)
+    gids, gdists = ix.search_filtered(queries, k, probe, words)
+    np.testing.assert_array_equal(gids == -1, exp_ids == -1)
+    _assert_parity(gids, gdists, exp_ids, exp_d, ctx="filtered")
+    # unfiltered still works after a filtered call
+    u_ids, _ = ix.search(queries, k, probe)
+    assert (u_ids >= 0).all()
+    ix.close()
