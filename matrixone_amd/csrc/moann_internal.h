@@ -78,6 +78,12 @@ void launch_pack(const float* vecs, int dim, int dpad,
                  const int64_t* slot_rows, int64_t ngroups, float* packed,
                  hipStream_t stream);
 
+/* MFMA f32 centroid-rank GEMM (mfma_rank.hip). Returns false when the
+ * metric has no MFMA path (L1). */
+bool launch_rank_gemm(int metric, const float* queries, const float* cents,
+                      const float* qnorms, const float* cnorms, int nq,
+                      int nlist, int dpad, float* out, hipStream_t stream);
+
 /* Sum-of-squares per query row (cosine). queries: [nq][dpad]. */
 void launch_qnorms(const float* queries, int nq, int dpad, float* qnorms,
                    hipStream_t stream);

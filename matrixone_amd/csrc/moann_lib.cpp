@@ -128,6 +128,9 @@ struct IvfIndex {
     /* built state */
     float* d_packed = nullptr;           /* interleaved entry data            */
     float* d_cent_packed = nullptr;      /* interleaved centroid matrix       */
+    float* d_cent_rows = nullptr;        /* [nlist][dpad] row-major (MFMA)    */
+    float* d_cnorms = nullptr;           /* [nlist] |c|^2 (MFMA rank)         */
+    bool rank_mfma = false;              /* MOANN_RANK=mfma opt-in            */
     int64_t* d_id_by_slot = nullptr;     /* [total_slots]                     */
     int64_t* d_list_slot_base = nullptr; /* [nlist]                           */
     std::vector<int64_t> list_gbase;     /* group base per list               */
@@ -154,6 +157,7 @@ struct IvfIndex {
     ~IvfIndex() {
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_staging, (void*)d_packed, (void*)d_cent_packed,
+                       (void*)d_cent_rows, (void*)d_cnorms,
                        (void*)d_id_by_slot, (void*)d_list_slot_base})
             if (p) (void)hipFree(p);
         if (stream) (void)hipStreamDestroy(stream);
@@ -170,7 +174,12 @@ int pick_qt(uint32_t dpad) {
      * QT=16 halves list re-reads when many queries probe one list (the
      * hub-list multiplicity); VALU ceiling at QT=16/d768 is ~9.8 TB/s,
      * still above the HBM bound. */
-    for (int qt : {16, 8, 4, 2, 1})
+    /* QT=8 measured best end-to-end: hipcc pipelines its inner loop with
+     * counted vmcnt (4 loads in flight) where the QT=16 instantiation
+     * degrades to a full vmcnt(0) stall per iteration; the extra tile
+     * re-reads at QT=8 mostly hit the XCD-affine L2 (physical fetch ~1.4x
+     * compulsory vs 1.14x at QT=16, but 4.6 vs 2.6 TB/s sustained). */
+    for (int qt : {8, 4, 2, 1})
         if ((size_t)qt * dpad * 4 + qt * 4 <= 52 * 1024) return qt;
     return 1;
 }
@@ -264,7 +273,8 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                                                  : hipMemcpyHostToDevice, s));
     const float* d_q = ix->w_queries.as<float>();
     const float* d_qn = nullptr;
-    if (ix->kmetric == KM_COS) {
+    const bool use_mfma_rank = ix->rank_mfma && ix->kmetric != KM_L1;
+    if (ix->kmetric == KM_COS || use_mfma_rank) {
         ix->w_qnorms.ensure(nq * 4);
         launch_qnorms(d_q, (int)nq, dpad, ix->w_qnorms.as<float>(), s);
         KCHECK("qnorms");
@@ -274,7 +284,21 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     /* 2. centroid rank: scan the centroid matrix (one "list" of nlist rows),
      * then top-probe per query (ivfflat/search.go:265-308 rankCentroids). */
     const int qt = pick_qt(ix->dpad);
-    {
+    if (use_mfma_rank) {
+        ix->w_cdists.ensure((size_t)nq * ix->nlist * 4);
+        HIP_CHECK(hipEventRecord(ix->perf.ev[0], s));
+        launch_rank_gemm(ix->kmetric, d_q, ix->d_cent_rows, d_qn,
+                         ix->d_cnorms, (int)nq, (int)ix->nlist, dpad,
+                         ix->w_cdists.as<float>(), s);
+        KCHECK("rank-gemm");
+        HIP_CHECK(hipEventRecord(ix->perf.ev[1], s));
+        ix->w_csel_slots.ensure((size_t)nq * probe * 4);
+        ix->w_csel_dists.ensure((size_t)nq * probe * 4);
+        launch_topk(ix->w_cdists.as<float>(), nullptr, ix->nlist, (int)nq,
+                    (int)probe, ix->w_csel_slots.as<int32_t>(),
+                    ix->w_csel_dists.as<float>(), s);
+        KCHECK("rank-topk");
+    } else {
         ix->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         /* tile queries, chunk centroid groups to fill the chip */
         const int ntiles = (int)((nq + qt - 1) / qt);
@@ -681,6 +705,32 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
             pack_lists(ix, d_cent.as<float>(), ix->nlist, csr, cl,
                        &ix->d_cent_packed, &cgb, &ix->cent_groups);
         }
+
+        /* MFMA rank operands: row-major zero-padded centroid matrix +
+         * host-computed |c|^2 */
+        {
+            HIP_CHECK(hipMalloc(&ix->d_cent_rows,
+                                (size_t)ix->nlist * ix->dpad * 4));
+            if (ix->dim != ix->dpad)
+                HIP_CHECK(hipMemset(ix->d_cent_rows, 0,
+                                    (size_t)ix->nlist * ix->dpad * 4));
+            HIP_CHECK(hipMemcpy2D(ix->d_cent_rows, (size_t)ix->dpad * 4,
+                                  ix->h_centroids.data(), (size_t)ix->dim * 4,
+                                  (size_t)ix->dim * 4, ix->nlist,
+                                  hipMemcpyHostToDevice));
+            std::vector<float> cn(ix->nlist);
+            for (uint32_t c = 0; c < ix->nlist; ++c) {
+                float ssum = 0.f;
+                const float* row = ix->h_centroids.data() + (size_t)c * ix->dim;
+                for (uint32_t e = 0; e < ix->dim; ++e) ssum += row[e] * row[e];
+                cn[c] = ssum;
+            }
+            HIP_CHECK(hipMalloc(&ix->d_cnorms, ix->nlist * 4));
+            HIP_CHECK(hipMemcpy(ix->d_cnorms, cn.data(), ix->nlist * 4,
+                                hipMemcpyHostToDevice));
+        }
+        if (const char* e = getenv("MOANN_RANK"))
+            ix->rank_mfma = std::string(e) == "mfma";
 
         /* drop staging */
         HIP_CHECK(hipFree(ix->d_staging));
