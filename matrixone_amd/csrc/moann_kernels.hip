@@ -43,7 +43,7 @@ __device__ __forceinline__ float u2f(uint32_t u) {
  * 16 B data quads (halves LDS traffic per byte). Per j4 step and query:
  * 8 fma-class VALU ops per 32 B/lane -> VALU ceiling ~19 TB/s equivalent,
  * comfortably above the ~6.3 TB/s HBM bound this kernel targets. */
-template <int METRIC, int QT>
+template <int METRIC, int QT, int GP>
 __global__ __launch_bounds__(256) void scan_kernel(
     const float* __restrict__ packed, const float* __restrict__ queries,
     const float* __restrict__ qnorms, int dpad,
@@ -98,10 +98,13 @@ __global__ __launch_bounds__(256) void scan_kernel(
     const int64_t baseg = j_databaseg[j];
     const int d4 = dpad >> 2;
 
-    /* each wave takes group pairs (2*wave, 2*wave+1) strided by 8 */
-    for (int gp = gstart + 2 * wave; gp < gstart + gcount; gp += 8) {
+    /* GP=2: each wave walks ADJACENT group pairs (one LDS broadcast feeds
+     * 32 B of data — best for QT<=8). GP=1: one group per wave — half the
+     * live registers, which is what lets hipcc software-pipeline the loads
+     * at QT=16. */
+    for (int gp = gstart + GP * wave; gp < gstart + gcount; gp += 4 * GP) {
         const int g0 = gp;
-        const bool has1 = (gp + 1) < (gstart + gcount);
+        const bool has1 = GP == 2 && (gp + 1) < (gstart + gcount);
         const int g1 = has1 ? gp + 1 : gp;
         const float4* __restrict__ d0 =
             (const float4*)(packed + (baseg + g0) * (int64_t)64 * dpad) + lane;
@@ -113,15 +116,83 @@ __global__ __launch_bounds__(256) void scan_kernel(
         for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0.f;
         float rn0 = 0.f, rn1 = 0.f;
 
+        /* GP==1: hand-rolled 3-deep load ring in inline asm. hipcc will not
+         * software-pipeline this loop at QT=16 register pressure (it emits a
+         * full vmcnt(0) drain per iteration); the asm ring keeps 2 loads in
+         * flight across iterations with counted waits (guide rules: asm
+         * loads are invisible to hipcc's s_waitcnt bookkeeping, so the
+         * counted wait + sched_barrier(0) before the first consumer is
+         * ours). */
+        if (GP == 1) {
+            /* named ring slots (no arrays: a runtime-indexed ext_vector
+             * array goes to scratch) */
+            float4 r0, r1, r2;
+#define MOANN_ISSUE(slot, idx)                                              \
+            {                                                               \
+                const float4* ap = d0 + ((idx) < d4 ? (idx) : d4 - 1) * 64; \
+                asm volatile("global_load_dwordx4 %0, %1, off"              \
+                             : "=v"(slot) : "v"(ap));                       \
+            }
+#define MOANN_CONSUME(slot, idx)                                            \
+            {                                                               \
+                asm volatile("s_waitcnt vmcnt(2)" ::: "memory");            \
+                __builtin_amdgcn_sched_barrier(0);                          \
+                const float4 x0 = slot;                                     \
+                const int q4 = (idx);                                       \
+                if (METRIC == KM_COS)                                       \
+                    rn0 += x0.x * x0.x + x0.y * x0.y + x0.z * x0.z +        \
+                           x0.w * x0.w;                                     \
+                _Pragma("unroll")                                           \
+                for (int t = 0; t < QT; ++t) {                              \
+                    const float4 qv =                                       \
+                        ((const float4*)(ldsq + t * dpad))[q4];             \
+                    if (METRIC == KM_L2SQ) {                                \
+                        float e;                                            \
+                        e = x0.x - qv.x; acc0[t] = fmaf(e, e, acc0[t]);     \
+                        e = x0.y - qv.y; acc0[t] = fmaf(e, e, acc0[t]);     \
+                        e = x0.z - qv.z; acc0[t] = fmaf(e, e, acc0[t]);     \
+                        e = x0.w - qv.w; acc0[t] = fmaf(e, e, acc0[t]);     \
+                    } else if (METRIC == KM_IP || METRIC == KM_COS) {       \
+                        acc0[t] = fmaf(x0.x, qv.x, acc0[t]);                \
+                        acc0[t] = fmaf(x0.y, qv.y, acc0[t]);                \
+                        acc0[t] = fmaf(x0.z, qv.z, acc0[t]);                \
+                        acc0[t] = fmaf(x0.w, qv.w, acc0[t]);                \
+                    } else {                                                \
+                        acc0[t] += fabsf(x0.x - qv.x) + fabsf(x0.y - qv.y) +\
+                                   fabsf(x0.z - qv.z) + fabsf(x0.w - qv.w); \
+                    }                                                       \
+                }                                                           \
+            }
+            MOANN_ISSUE(r0, 0)
+            MOANN_ISSUE(r1, 1)
+            MOANN_ISSUE(r2, 2)
+            int q4 = 0;
+            for (; q4 + 3 <= d4; q4 += 3) {
+                MOANN_CONSUME(r0, q4)
+                MOANN_ISSUE(r0, q4 + 3)
+                MOANN_CONSUME(r1, q4 + 1)
+                MOANN_ISSUE(r1, q4 + 4)
+                MOANN_CONSUME(r2, q4 + 2)
+                MOANN_ISSUE(r2, q4 + 5)
+            }
+            /* tail (<=2 iterations): drain, then consume without reissue */
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+            if (q4 < d4) { MOANN_CONSUME(r0, q4) }
+            if (q4 + 1 < d4) { MOANN_CONSUME(r1, q4 + 1) }
+#undef MOANN_ISSUE
+#undef MOANN_CONSUME
+        } else
 #pragma unroll 2 /* two iterations in flight: 4 dwordx4 issued together,
                     counted vmcnt waits (hipcc pipelines the pair the same
                     way it does for the QT=8 instantiation) */
         for (int q4 = 0; q4 < d4; ++q4) {
             const float4 x0 = d0[q4 * 64];
-            const float4 x1 = d1[q4 * 64];
+            const float4 x1 = GP == 2 ? d1[q4 * 64] : float4 {0, 0, 0, 0};
             if (METRIC == KM_COS) {
                 rn0 += x0.x * x0.x + x0.y * x0.y + x0.z * x0.z + x0.w * x0.w;
-                rn1 += x1.x * x1.x + x1.y * x1.y + x1.z * x1.z + x1.w * x1.w;
+                if (GP == 2)
+                    rn1 += x1.x * x1.x + x1.y * x1.y + x1.z * x1.z + x1.w * x1.w;
             }
 #pragma unroll
             for (int t = 0; t < QT; ++t) {
@@ -132,24 +203,29 @@ __global__ __launch_bounds__(256) void scan_kernel(
                     e = x0.y - qv.y; acc0[t] = fmaf(e, e, acc0[t]);
                     e = x0.z - qv.z; acc0[t] = fmaf(e, e, acc0[t]);
                     e = x0.w - qv.w; acc0[t] = fmaf(e, e, acc0[t]);
-                    e = x1.x - qv.x; acc1[t] = fmaf(e, e, acc1[t]);
-                    e = x1.y - qv.y; acc1[t] = fmaf(e, e, acc1[t]);
-                    e = x1.z - qv.z; acc1[t] = fmaf(e, e, acc1[t]);
-                    e = x1.w - qv.w; acc1[t] = fmaf(e, e, acc1[t]);
+                    if (GP == 2) {
+                        e = x1.x - qv.x; acc1[t] = fmaf(e, e, acc1[t]);
+                        e = x1.y - qv.y; acc1[t] = fmaf(e, e, acc1[t]);
+                        e = x1.z - qv.z; acc1[t] = fmaf(e, e, acc1[t]);
+                        e = x1.w - qv.w; acc1[t] = fmaf(e, e, acc1[t]);
+                    }
                 } else if (METRIC == KM_IP || METRIC == KM_COS) {
                     acc0[t] = fmaf(x0.x, qv.x, acc0[t]);
                     acc0[t] = fmaf(x0.y, qv.y, acc0[t]);
                     acc0[t] = fmaf(x0.z, qv.z, acc0[t]);
                     acc0[t] = fmaf(x0.w, qv.w, acc0[t]);
-                    acc1[t] = fmaf(x1.x, qv.x, acc1[t]);
-                    acc1[t] = fmaf(x1.y, qv.y, acc1[t]);
-                    acc1[t] = fmaf(x1.z, qv.z, acc1[t]);
-                    acc1[t] = fmaf(x1.w, qv.w, acc1[t]);
+                    if (GP == 2) {
+                        acc1[t] = fmaf(x1.x, qv.x, acc1[t]);
+                        acc1[t] = fmaf(x1.y, qv.y, acc1[t]);
+                        acc1[t] = fmaf(x1.z, qv.z, acc1[t]);
+                        acc1[t] = fmaf(x1.w, qv.w, acc1[t]);
+                    }
                 } else { /* KM_L1 */
                     acc0[t] += fabsf(x0.x - qv.x) + fabsf(x0.y - qv.y) +
                                fabsf(x0.z - qv.z) + fabsf(x0.w - qv.w);
-                    acc1[t] += fabsf(x1.x - qv.x) + fabsf(x1.y - qv.y) +
-                               fabsf(x1.z - qv.z) + fabsf(x1.w - qv.w);
+                    if (GP == 2)
+                        acc1[t] += fabsf(x1.x - qv.x) + fabsf(x1.y - qv.y) +
+                                   fabsf(x1.z - qv.z) + fabsf(x1.w - qv.w);
                 }
             }
         }
@@ -216,23 +292,23 @@ static void launch_scan_qt(int qt, const float* packed, const float* queries,
     jb.slot_base, filter_bitset, dists_out
     switch (qt) {
     case 16:
-        hipLaunchKernelGGL((scan_kernel<METRIC, 16>), grid, block,
+        hipLaunchKernelGGL((scan_kernel<METRIC, 16, 1>), grid, block,
                            (16 * dpad + 16) * sizeof(float), stream, ARGS);
         break;
     case 8:
-        hipLaunchKernelGGL((scan_kernel<METRIC, 8>), grid, block,
+        hipLaunchKernelGGL((scan_kernel<METRIC, 8, 2>), grid, block,
                            (8 * dpad + 8) * sizeof(float), stream, ARGS);
         break;
     case 4:
-        hipLaunchKernelGGL((scan_kernel<METRIC, 4>), grid, block,
+        hipLaunchKernelGGL((scan_kernel<METRIC, 4, 2>), grid, block,
                            (4 * dpad + 4) * sizeof(float), stream, ARGS);
         break;
     case 2:
-        hipLaunchKernelGGL((scan_kernel<METRIC, 2>), grid, block,
+        hipLaunchKernelGGL((scan_kernel<METRIC, 2, 2>), grid, block,
                            (2 * dpad + 2) * sizeof(float), stream, ARGS);
         break;
     default:
-        hipLaunchKernelGGL((scan_kernel<METRIC, 1>), grid, block,
+        hipLaunchKernelGGL((scan_kernel<METRIC, 1, 2>), grid, block,
                            (1 * dpad + 1) * sizeof(float), stream, ARGS);
         break;
     }
