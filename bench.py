@@ -206,14 +206,21 @@ def main():
     torch.cuda.synchronize()
 
     log("[bench] ingest + pack into moann index ...")
-    host_vecs = data.cpu().numpy()  # host copy (oracle baseline + ingest)
     cent_np = cent.cpu().numpy()
     assign_np = assign.cpu().numpy()
+    # keep a full host copy only where the CPU baseline will need it
+    # (rank 0, N=1); otherwise stream 2M-row chunks through a small host
+    # buffer so an 8-rank run never stages 8 x 30 GB in host RAM at once.
+    need_host = (rank == 0 and world == 1 and not args.skip_cpu_baseline)
+    host_vecs = data.cpu().numpy() if need_host else None
     ix = engine.IvfFlatIndex(args.dim, args.nlist, metric="l2sq",
                              capacity=args.rows, device=local_rank)
     CH = 1 << 21
     for c0 in range(0, args.rows, CH):
-        ix.add(host_vecs[c0:c0 + CH])
+        if host_vecs is not None:
+            ix.add(host_vecs[c0:c0 + CH])
+        else:
+            ix.add(data[c0:c0 + CH].cpu().numpy())
     ix.set_centroids(cent_np)
     ix.set_assignments(assign_np)
     ix.build()
@@ -298,7 +305,7 @@ def main():
 
     # ---- CPU baseline (rank 0, N=1 only) -----------------------------------
     cpu_baseline = None
-    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
+    if host_vecs is not None:
         log("[bench] CPU baseline (oracle port, OpenMP) ...")
         try:
             cpu_baseline = run_cpu_baseline(
