@@ -83,6 +83,15 @@ def lib():
         L.ref_hnsw_save.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_void_p]
         L.ref_hnsw_load.restype = ct.c_void_p
         L.ref_hnsw_load.argtypes = [ct.c_char_p, ct.c_void_p]
+        L.ref_hnsw_export_bulk.restype = ct.c_int64
+        L.ref_hnsw_export_bulk.argtypes = [ct.c_void_p, ct.c_size_t,
+                                           ct.c_void_p, ct.c_void_p,
+                                           ct.c_void_p, ct.c_void_p,
+                                           ct.c_void_p, ct.c_int64]
+        L.ref_hnsw_export_level.restype = ct.c_int64
+        L.ref_hnsw_export_level.argtypes = [ct.c_void_p, ct.c_int,
+                                            ct.c_void_p, ct.c_void_p,
+                                            ct.c_int64]
         L.ref_hnsw_free.argtypes = [ct.c_void_p]
         _lib = L
     return _lib
@@ -143,25 +152,25 @@ class RefHnsw:
         levels = np.empty(n, dtype=np.int32)
         keys = np.empty(n, dtype=np.uint64)
         vecs = np.empty((n, self.dim), dtype=np.float32)
-        for s in range(n):
-            levels[s] = L.ref_hnsw_node_level(self._h, s)
-            keys[s] = L.ref_hnsw_key_of_slot(self._h, s)
-            L.ref_hnsw_get_vector(self._h, s, vecs[s].ctypes.data, self.dim)
-        buf = np.empty(4096, dtype=np.uint32)
-        adj = []  # per level: (offsets int64[n+1], neighbors uint32[...])
-        for lvl in range(max_level + 1):
-            offs = np.zeros(n + 1, dtype=np.int64)
-            chunks = []
-            for s in range(n):
-                if levels[s] >= lvl:
-                    cnt = L.ref_hnsw_neighbors(self._h, s, lvl,
-                                               buf.ctypes.data, buf.size)
-                    chunks.append(buf[:cnt].copy())
-                    offs[s + 1] = offs[s] + cnt
-                else:
-                    offs[s + 1] = offs[s]
-            adj.append((offs, np.concatenate(chunks) if chunks
-                        else np.empty(0, dtype=np.uint32)))
+        # bulk C-side export (per-node ctypes calls do not scale past ~1e5)
+        cap = np.int64(n) * 128 + 64
+        l0_offs = np.empty(n + 1, dtype=np.int64)
+        l0_nbrs = np.empty(cap, dtype=np.uint32)
+        nnz = L.ref_hnsw_export_bulk(self._h, self.dim, levels.ctypes.data,
+                                     keys.ctypes.data, vecs.ctypes.data,
+                                     l0_offs.ctypes.data, l0_nbrs.ctypes.data,
+                                     cap)
+        if nnz < 0:
+            raise RuntimeError("l0 adjacency overflow")
+        adj = [(l0_offs, l0_nbrs[:nnz].copy())]
+        for lvl in range(1, max_level + 1):
+            offs = np.empty(n + 1, dtype=np.int64)
+            nbrs = np.empty(cap, dtype=np.uint32)
+            nz = L.ref_hnsw_export_level(self._h, lvl, offs.ctypes.data,
+                                         nbrs.ctypes.data, cap)
+            if nz < 0:
+                raise RuntimeError("upper adjacency overflow")
+            adj.append((offs, nbrs[:nz].copy()))
         return HnswGraph(self.dim, self.metric, int(entry), int(max_level),
                          levels, keys, vecs, adj)
 

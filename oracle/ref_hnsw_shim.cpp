@@ -165,6 +165,53 @@ int ref_hnsw_get_vector(void* hp, uint32_t slot, float* out, size_t dim) {
     return 0;
 }
 
+/* Bulk export: fills levels[n], keys[n], vecs[n*dim] and the level-0 CSR in
+ * one call; upper levels via ref_hnsw_export_level. Returns l0 nnz. */
+int64_t ref_hnsw_export_bulk(void* hp, size_t dim, int32_t* levels,
+                             uint64_t* keys, float* vecs, int64_t* l0_offs,
+                             uint32_t* l0_nbrs, int64_t l0_cap) {
+    auto h = (ref_hnsw_t*)hp;
+    auto& typed = *h->index.typed_;
+    const size_t n = h->index.size();
+    int64_t off = 0;
+    for (size_t s = 0; s < n; ++s) {
+        auto node = typed.node_at_((uint32_t)s);
+        levels[s] = (int32_t)node.level();
+        keys[s] = (uint64_t)node.ckey();
+        const byte_t* v = h->index.vectors_lookup_[s];
+        memcpy(vecs + s * dim, v, dim * sizeof(float));
+        l0_offs[s] = off;
+        auto nbrs = typed.neighbors_base_(node);
+        const size_t cnt = nbrs.size();
+        if (off + (int64_t)cnt > l0_cap) return -1;
+        for (size_t i = 0; i < cnt; ++i) l0_nbrs[off++] = (uint32_t)nbrs[i];
+    }
+    l0_offs[n] = off;
+    return off;
+}
+
+/* CSR for one upper level (1-based). Returns nnz or -1 on overflow. */
+int64_t ref_hnsw_export_level(void* hp, int level, int64_t* offs,
+                              uint32_t* nbrs, int64_t cap) {
+    auto h = (ref_hnsw_t*)hp;
+    auto& typed = *h->index.typed_;
+    const size_t n = h->index.size();
+    int64_t off = 0;
+    using lvl_t = std::remove_reference_t<decltype(typed)>::level_t;
+    for (size_t s = 0; s < n; ++s) {
+        auto node = typed.node_at_((uint32_t)s);
+        offs[s] = off;
+        if ((int)node.level() >= level) {
+            auto nb = typed.neighbors_non_base_(node, (lvl_t)level);
+            const size_t cnt = nb.size();
+            if (off + (int64_t)cnt > cap) return -1;
+            for (size_t i = 0; i < cnt; ++i) nbrs[off++] = (uint32_t)nb[i];
+        }
+    }
+    offs[n] = off;
+    return off;
+}
+
 int ref_hnsw_save(void* hp, const char* path, char** err) {
     auto h = (ref_hnsw_t*)hp;
     auto r = h->index.save(path);
