@@ -126,6 +126,14 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c index_c, void* errmsg);
 /* ivf_flat_c.h:57 */
 void gpu_ivf_flat_start(gpu_ivf_flat_c index_c, void* errmsg);
 
+/* ivf_flat_c.h:98-106 persistence: a directory with manifest.json +
+ * per-component blobs (index_base.hpp:239-249 contract). load_dir requires
+ * a freshly created handle with matching dimension/metric/n_lists. */
+void gpu_ivf_flat_save_dir(gpu_ivf_flat_c index_c, const char* dir,
+                           void* errmsg);
+void gpu_ivf_flat_load_dir(gpu_ivf_flat_c index_c, const char* dir,
+                           distribution_mode_t target_mode, void* errmsg);
+
 /* ivf_flat_c.h:53 */
 void gpu_ivf_flat_destroy(gpu_ivf_flat_c index_c, void* errmsg);
 

@@ -78,6 +78,12 @@ void launch_pack(const float* vecs, int dim, int dpad,
                  const int64_t* slot_rows, int64_t ngroups, float* packed,
                  hipStream_t stream);
 
+/* Inverse of pack (save_dir): packed -> row-major rows in slot order. */
+void launch_unpack(const float* packed, int dim, int dpad,
+                   const int64_t* group_slotbase, const int32_t* group_valid,
+                   int64_t ngroups, int64_t slot_lo, int64_t slot_hi,
+                   float* out, hipStream_t stream);
+
 /* MFMA f32 centroid-rank GEMM (mfma_rank.hip). Returns false when the
  * metric has no MFMA path (L1). */
 bool launch_rank_gemm(int metric, const float* queries, const float* cents,

@@ -575,6 +575,47 @@ void launch_pack(const float* vecs, int dim, int dpad,
                        slot_rows, ngroups, packed);
 }
 
+/* ------------------------------ unpack -----------------------------------
+ * Inverse of pack_kernel for save_dir: packed groups -> row-major f32 rows
+ * in SLOT order (valid slots only), one chunk of slots at a time. */
+__global__ void unpack_kernel(const float* __restrict__ packed, int dim,
+                              int dpad,
+                              const int64_t* __restrict__ group_slotbase,
+                              const int32_t* __restrict__ group_valid,
+                              int64_t ngroups, int64_t slot_lo,
+                              int64_t slot_hi, float* __restrict__ out) {
+    const int64_t nquads = ngroups * (dpad >> 2) * 64;
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= nquads) return;
+    const int lane = (int)(idx & 63);
+    const int64_t rest = idx >> 6;
+    const int j4 = (int)(rest % (dpad >> 2));
+    const int64_t g = rest / (dpad >> 2);
+    if (lane >= group_valid[g]) return;
+    const int64_t slot = group_slotbase[g] + lane;
+    if (slot < slot_lo || slot >= slot_hi) return;
+    const float4 v = ((const float4*)packed)[idx];
+    float* dst = out + (slot - slot_lo) * (int64_t)dim;
+    const int base = 4 * j4;
+    if (base + 0 < dim) dst[base + 0] = v.x;
+    if (base + 1 < dim) dst[base + 1] = v.y;
+    if (base + 2 < dim) dst[base + 2] = v.z;
+    if (base + 3 < dim) dst[base + 3] = v.w;
+}
+
+void launch_unpack(const float* packed, int dim, int dpad,
+                   const int64_t* group_slotbase, const int32_t* group_valid,
+                   int64_t ngroups, int64_t slot_lo, int64_t slot_hi,
+                   float* out, hipStream_t stream) {
+    const int64_t nquads = ngroups * (int64_t)(dpad >> 2) * 64;
+    if (!nquads) return;
+    const int block = 256;
+    const int64_t grid = (nquads + block - 1) / block;
+    hipLaunchKernelGGL(unpack_kernel, dim3((uint32_t)grid), dim3(block), 0,
+                       stream, packed, dim, dpad, group_slotbase, group_valid,
+                       ngroups, slot_lo, slot_hi, out);
+}
+
 /* --------------------------- query norms ---------------------------------
  * one wave per query; f32 accumulate (matches distance_func.go cosine's f32
  * norm accumulation within the 1e-5 parity tolerance). */

@@ -134,6 +134,8 @@ struct IvfIndex {
     int64_t* d_id_by_slot = nullptr;     /* [total_slots]                     */
     int64_t* d_list_slot_base = nullptr; /* [nlist]                           */
     std::vector<int64_t> list_gbase;     /* group base per list               */
+    std::vector<int64_t> group_slotbase; /* per group: global slot of lane 0  */
+    std::vector<int32_t> group_valid;    /* per group: valid rows             */
     std::vector<int32_t> list_rows;      /* rows per list                     */
     std::vector<int64_t> list_slot_base; /* slot base per list                */
     int64_t total_groups = 0;
@@ -678,6 +680,16 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
         /* pack entries */
         pack_lists(ix, ix->d_staging, ix->count, slot_rows, loffsets,
                    &ix->d_packed, &ix->list_gbase, &ix->total_groups);
+        /* group -> slot metadata kept for save_dir's unpack */
+        ix->group_slotbase.resize(ix->total_groups);
+        ix->group_valid.resize(ix->total_groups);
+        for (uint32_t l = 0; l < ix->nlist; ++l)
+            for (int64_t g = ix->list_gbase[l]; g < ix->list_gbase[l + 1]; ++g) {
+                const int64_t gi = g - ix->list_gbase[l];
+                ix->group_slotbase[g] = loffsets[l] + gi * 64;
+                ix->group_valid[g] = (int32_t)std::min<int64_t>(
+                    64, (loffsets[l + 1] - loffsets[l]) - gi * 64);
+            }
 
         /* id map: slot -> entry id */
         std::vector<int64_t> id_by_slot(std::max<uint64_t>(1, ix->count));
@@ -954,6 +966,158 @@ void moann_ivf_flat_slots_of_ids(gpu_ivf_flat_c h, const int64_t* ids,
         }
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "moann_ivf_flat_slots_of_ids", e.what());
+    }
+}
+
+/* save_dir / load_dir — the persistence contract of the reference GPU
+ * index (cgo/cuvs/index_base.hpp:239-249, gpu_ivf_flat_save_dir/load_dir in
+ * ivf_flat_c.h:98-106): a directory with manifest.json + per-component
+ * blobs; load recreates the index into a handle that was created (with
+ * matching dim/metric/n_lists) and started. Vectors/ids are stored in SLOT
+ * order (equivalent list contents and in-list order). */
+void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (!ix->built) throw std::runtime_error("index not built");
+        HIP_CHECK(hipSetDevice(ix->device));
+        std::string d(dir);
+        auto wfile = [&](const std::string& name, const void* p, size_t n) {
+            FILE* f = fopen((d + "/" + name).c_str(), "wb");
+            if (!f) throw std::runtime_error("cannot write " + name);
+            if (n && fwrite(p, 1, n, f) != n) {
+                fclose(f);
+                throw std::runtime_error("short write " + name);
+            }
+            fclose(f);
+        };
+        {
+            char buf[512];
+            snprintf(buf, sizeof buf,
+                     "{\n  \"format\": \"moann-ivf-flat-v1\",\n"
+                     "  \"dimension\": %u,\n  \"n_lists\": %u,\n"
+                     "  \"metric\": %d,\n  \"count\": %llu,\n"
+                     "  \"quant_mul\": %.17g\n}\n",
+                     ix->dim, ix->nlist, (int)ix->metric,
+                     (unsigned long long)ix->count, ix->quant_mul);
+            wfile("manifest.json", buf, strlen(buf));
+        }
+        wfile("centroids.f32", ix->h_centroids.data(),
+              ix->h_centroids.size() * 4);
+        wfile("ids.i64", ix->h_id_by_slot.data(), ix->count * 8);
+        /* per-slot list id (assignments in slot order) */
+        {
+            std::vector<int32_t> a(ix->count);
+            for (uint32_t l = 0; l < ix->nlist; ++l)
+                for (int64_t sidx = ix->list_slot_base[l];
+                     sidx < ix->list_slot_base[l] + ix->list_rows[l]; ++sidx)
+                    a[sidx] = (int32_t)l;
+            wfile("assign.i32", a.data(), ix->count * 4);
+        }
+        /* vectors.f32: unpack in chunks of groups */
+        {
+            FILE* f = fopen((d + "/vectors.f32").c_str(), "wb");
+            if (!f) throw std::runtime_error("cannot write vectors.f32");
+            DevBuf d_gsb, d_gv, d_tmp;
+            d_gsb.ensure(ix->total_groups * 8);
+            d_gv.ensure(ix->total_groups * 4);
+            HIP_CHECK(hipMemcpy(d_gsb.ptr, ix->group_slotbase.data(),
+                                ix->total_groups * 8, hipMemcpyHostToDevice));
+            HIP_CHECK(hipMemcpy(d_gv.ptr, ix->group_valid.data(),
+                                ix->total_groups * 4, hipMemcpyHostToDevice));
+            const int64_t CH = 1 << 20; /* slots per chunk */
+            d_tmp.ensure((size_t)CH * ix->dim * 4);
+            std::vector<float> hbuf((size_t)CH * ix->dim);
+            for (int64_t lo = 0; lo < (int64_t)ix->count; lo += CH) {
+                const int64_t hi = std::min<int64_t>(lo + CH, ix->count);
+                launch_unpack(ix->d_packed, (int)ix->dim, (int)ix->dpad,
+                              d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
+                              ix->total_groups, lo, hi, d_tmp.as<float>(),
+                              ix->stream);
+                HIP_CHECK(hipMemcpyAsync(hbuf.data(), d_tmp.ptr,
+                                         (size_t)(hi - lo) * ix->dim * 4,
+                                         hipMemcpyDeviceToHost, ix->stream));
+                HIP_CHECK(hipStreamSynchronize(ix->stream));
+                HIP_CHECK(hipGetLastError());
+                const size_t n = (size_t)(hi - lo) * ix->dim * 4;
+                if (fwrite(hbuf.data(), 1, n, f) != n) {
+                    fclose(f);
+                    throw std::runtime_error("short write vectors.f32");
+                }
+            }
+            fclose(f);
+        }
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_save_dir", e.what());
+    }
+}
+
+void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
+                           distribution_mode_t /*target_mode*/, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (ix->built) throw std::runtime_error("index already built");
+        std::string d(dir);
+        auto rfile = [&](const std::string& name, void* p, size_t n) {
+            FILE* f = fopen((d + "/" + name).c_str(), "rb");
+            if (!f) throw std::runtime_error("cannot read " + name);
+            const size_t got = fread(p, 1, n, f);
+            fclose(f);
+            if (got != n) throw std::runtime_error("short read " + name);
+        };
+        /* manifest: minimal field validation */
+        {
+            FILE* f = fopen((d + "/manifest.json").c_str(), "rb");
+            if (!f) throw std::runtime_error("cannot read manifest.json");
+            char buf[1024] = {0};
+            size_t got = fread(buf, 1, sizeof buf - 1, f);
+            fclose(f);
+            (void)got;
+            unsigned dim = 0, nlist = 0;
+            unsigned long long count = 0;
+            int metric = 0;
+            if (!strstr(buf, "moann-ivf-flat-v1"))
+                throw std::runtime_error("unknown manifest format");
+            sscanf(strstr(buf, "\"dimension\":"), "\"dimension\": %u", &dim);
+            sscanf(strstr(buf, "\"n_lists\":"), "\"n_lists\": %u", &nlist);
+            sscanf(strstr(buf, "\"metric\":"), "\"metric\": %d", &metric);
+            sscanf(strstr(buf, "\"count\":"), "\"count\": %llu", &count);
+            if (dim != ix->dim || nlist != ix->nlist ||
+                metric != (int)ix->metric)
+                throw std::runtime_error(
+                    "manifest does not match the created index");
+            if (count > ix->cap)
+                throw std::runtime_error("manifest count exceeds capacity");
+            std::vector<float> cents((size_t)nlist * dim);
+            rfile("centroids.f32", cents.data(), cents.size() * 4);
+            moann_ivf_flat_set_centroids(h, cents.data(), nlist, errmsg);
+            std::vector<int64_t> ids(count);
+            rfile("ids.i64", ids.data(), count * 8);
+            std::vector<int32_t> assign(count);
+            rfile("assign.i32", assign.data(), count * 4);
+            /* vectors streamed into add_chunk */
+            FILE* vf = fopen((d + "/vectors.f32").c_str(), "rb");
+            if (!vf) throw std::runtime_error("cannot read vectors.f32");
+            const size_t CH = 1 << 20;
+            std::vector<float> vbuf(CH * dim);
+            unsigned long long done = 0;
+            while (done < count) {
+                const size_t n = (size_t)std::min<unsigned long long>(
+                    CH, count - done);
+                if (fread(vbuf.data(), 4, n * dim, vf) != n * dim) {
+                    fclose(vf);
+                    throw std::runtime_error("short read vectors.f32");
+                }
+                gpu_ivf_flat_add_chunk(h, vbuf.data(), n, ids.data() + done,
+                                       errmsg);
+                done += n;
+            }
+            fclose(vf);
+            moann_ivf_flat_set_assignments(h, assign.data(), count, errmsg);
+            gpu_ivf_flat_build(h, errmsg);
+        }
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_load_dir", e.what());
     }
 }
 

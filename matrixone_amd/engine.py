@@ -95,6 +95,9 @@ def lib() -> ct.CDLL:
         f.argtypes = [ct.c_void_p]
     L.gpu_ivf_flat_get_n_list.restype = ct.c_uint32
     L.gpu_ivf_flat_get_n_list.argtypes = [ct.c_void_p]
+    L.gpu_ivf_flat_save_dir.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_void_p]
+    L.gpu_ivf_flat_load_dir.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_int,
+                                        ct.c_void_p]
     L.moann_ivf_flat_search_filtered.restype = SearchRes
     L.moann_ivf_flat_search_filtered.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
@@ -221,6 +224,16 @@ class IvfFlatIndex:
                                         SearchParams(nprobe), err.ref)
         err.check("search")
         return self._read_result(res, nq, k, "search")
+
+    def save_dir(self, path: str):
+        err = _Err()
+        lib().gpu_ivf_flat_save_dir(self._h, path.encode(), err.ref)
+        err.check("save_dir")
+
+    def load_dir(self, path: str):
+        err = _Err()
+        lib().gpu_ivf_flat_load_dir(self._h, path.encode(), 0, err.ref)
+        err.check("load_dir")
 
     def slots_of_ids(self, ids):
         ids = np.ascontiguousarray(ids, dtype=np.int64)

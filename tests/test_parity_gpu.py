@@ -226,3 +226,38 @@ def test_filtered_search():
     u_ids, _ = ix.search(queries, k, probe)
     assert (u_ids >= 0).all()
     ix.close()
+
+
+def test_save_load_dir(tmp_path):
+    """save_dir -> fresh handle + load_dir reproduces identical results
+    (the gpu_ivf_flat_save_dir/load_dir persistence contract)."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(600))
+    n, d, nlist, nq, probe, k = 5000, 40, 16, 8, 4, 12
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    ids64 = rng.permutation(n).astype(np.int64) + 5
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs, ids=ids64)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids1, d1 = ix.search(queries, k, probe)
+    ix.save_dir(str(tmp_path))
+    ix.close()
+
+    ix2 = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix2.load_dir(str(tmp_path))
+    ids2, d2 = ix2.search(queries, k, probe)
+    np.testing.assert_array_equal(ids1, ids2)
+    np.testing.assert_array_equal(d1, d2)
+    # mismatched geometry must be rejected
+    ix3 = engine.IvfFlatIndex(d, nlist * 2, metric="l2sq", capacity=n)
+    with pytest.raises(engine.MoannError):
+        ix3.load_dir(str(tmp_path))
+    ix2.close()
+    ix3.close()
