@@ -136,7 +136,10 @@ __global__ __launch_bounds__(256) void scan_kernel(
 #define MOANN_CONSUME(slot, idx)                                            \
             {                                                               \
                 asm volatile("s_waitcnt vmcnt(2)" ::: "memory");            \
-                __builtin_amdgcn_sched_barrier(0);                          \
+                /* fence only ALU past the wait: the qv ds_reads do not     \
+                 * depend on the ring and may pipeline across iterations    \
+                 * (mask: DS_READ 0x100 | VMEM_READ 0x20 may cross) */      \
+                __builtin_amdgcn_sched_barrier(0x120);                      \
                 const float4 x0 = slot;                                     \
                 const int q4 = (idx);                                       \
                 if (METRIC == KM_COS)                                       \
