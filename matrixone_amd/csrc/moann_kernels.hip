@@ -136,10 +136,13 @@ __global__ __launch_bounds__(256) void scan_kernel(
 #define MOANN_CONSUME(slot, idx)                                            \
             {                                                               \
                 asm volatile("s_waitcnt vmcnt(2)" ::: "memory");            \
-                /* fence only ALU past the wait: the qv ds_reads do not     \
-                 * depend on the ring and may pipeline across iterations    \
-                 * (mask: DS_READ 0x100 | VMEM_READ 0x20 may cross) */      \
-                __builtin_amdgcn_sched_barrier(0x120);                      \
+                /* full fence (mask 0) is REQUIRED: any relaxation that     \
+                 * lets VMEM ops cross lets a ring REISSUE hoist above this \
+                 * wait, the outstanding count then under-retires and the   \
+                 * consumer reads an un-landed slot (measured: recall 0.0   \
+                 * at mask DS_READ|VMEM_READ while 1.6x "faster").          \
+                 * Counted-wait discipline = nothing crosses the wait.      */\
+                __builtin_amdgcn_sched_barrier(0);                          \
                 const float4 x0 = slot;                                     \
                 const int q4 = (idx);                                       \
                 if (METRIC == KM_COS)                                       \
