@@ -224,6 +224,41 @@ typedef struct {
 void moann_ivf_flat_perf(gpu_ivf_flat_c index_c, moann_perf_t* out);
 void moann_ivf_flat_perf_reset(gpu_ivf_flat_c index_c);
 
+/* --- IVF-PQ (cgo/cuvs/ivf_pq_c.h precedent) -------------- */
+
+/* ADC search over per-subspace codebooks with coarse-residual encoding (the
+ * published IVF-PQ algorithm the reference drives through cuVS; CPU
+ * restatement in oracle/pq.py). Harness-trained ingestion like the IVF-Flat
+ * moann path: centroids + codebooks [n_sub][256][dim/n_sub] f32 + uint8
+ * codes [n][n_sub] + assignments. L2 only this round. Results are read with
+ * the shared gpu_ivf_flat_get_neighbors/get_distances/free_result. */
+
+typedef void* moann_ivf_pq_c;
+
+moann_ivf_pq_c moann_ivf_pq_new(uint32_t dimension, uint32_t n_sub,
+                                uint32_t n_lists, uint64_t total_count,
+                                const int* devices, int device_count,
+                                void* errmsg);
+void moann_ivf_pq_set_centroids(moann_ivf_pq_c h, const float* centroids,
+                                uint32_t n_lists, void* errmsg);
+void moann_ivf_pq_set_codebooks(moann_ivf_pq_c h, const float* codebooks,
+                                void* errmsg);
+void moann_ivf_pq_add_codes(moann_ivf_pq_c h, const uint8_t* codes,
+                            uint64_t n, const int64_t* ids, void* errmsg);
+void moann_ivf_pq_set_assignments(moann_ivf_pq_c h,
+                                  const int32_t* assignments, uint64_t count,
+                                  void* errmsg);
+void moann_ivf_pq_build(moann_ivf_pq_c h, void* errmsg);
+gpu_ivf_flat_search_res_t moann_ivf_pq_search(moann_ivf_pq_c h,
+                                              const void* queries_data,
+                                              uint64_t num_queries,
+                                              uint32_t query_dimension,
+                                              uint32_t limit,
+                                              ivf_flat_search_params_t search_params,
+                                              void* errmsg);
+void moann_ivf_pq_perf(moann_ivf_pq_c h, moann_perf_t* out);
+void moann_ivf_pq_destroy(moann_ivf_pq_c h, void* errmsg);
+
 /* --- HNSW (pkg/vectorindex/hnsw; usearch graph ingested to HBM) ----------- */
 
 /* The GPU replacement for the usearch traversal behind HnswSearch

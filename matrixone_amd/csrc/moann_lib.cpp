@@ -27,6 +27,7 @@
 #include <vector>
 
 #include "../../include/moann.h"
+#include "moann_host_common.h"
 #include "moann_internal.h"
 
 namespace {
@@ -87,10 +88,8 @@ struct DevBuf {
     T* as() { return (T*)ptr; }
 };
 
-struct Result {
-    std::vector<int64_t> neighbors;
-    std::vector<float> distances;
-};
+using Result = MoannResult;  /* shared with the PQ/HNSW libs so the
+    gpu_ivf_flat_get_* readout works on every family's results */
 
 struct PerfState {
     moann_perf_t c {};

@@ -108,6 +108,25 @@ def lib() -> ct.CDLL:
         ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, ct.c_void_p,
         ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, ct.c_void_p,
         ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_pq_new.restype = ct.c_void_p
+    L.moann_ivf_pq_new.argtypes = [ct.c_uint32, ct.c_uint32, ct.c_uint32,
+                                   ct.c_uint64, ct.POINTER(ct.c_int),
+                                   ct.c_int, ct.c_void_p]
+    L.moann_ivf_pq_set_centroids.argtypes = [ct.c_void_p, ct.c_void_p,
+                                             ct.c_uint32, ct.c_void_p]
+    L.moann_ivf_pq_set_codebooks.argtypes = [ct.c_void_p, ct.c_void_p,
+                                             ct.c_void_p]
+    L.moann_ivf_pq_add_codes.argtypes = [ct.c_void_p, ct.c_void_p,
+                                         ct.c_uint64, ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_pq_set_assignments.argtypes = [ct.c_void_p, ct.c_void_p,
+                                               ct.c_uint64, ct.c_void_p]
+    L.moann_ivf_pq_build.argtypes = [ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_pq_search.restype = SearchRes
+    L.moann_ivf_pq_search.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_uint64,
+                                      ct.c_uint32, ct.c_uint32, SearchParams,
+                                      ct.c_void_p]
+    L.moann_ivf_pq_perf.argtypes = [ct.c_void_p, ct.POINTER(PerfT)]
+    L.moann_ivf_pq_destroy.argtypes = [ct.c_void_p, ct.c_void_p]
     L.moann_ivf_flat_perf.argtypes = [ct.c_void_p, ct.POINTER(PerfT)]
     L.moann_ivf_flat_perf_reset.argtypes = [ct.c_void_p]
     L.moann_device_count.restype = ct.c_int
@@ -336,3 +355,90 @@ def brute_force_search(dataset, queries, k: int, metric: str = "l2sq",
         out_ids.ctypes.data, out_dists.ctypes.data, err.ref)
     err.check("brute_force_search")
     return out_ids, out_dists
+
+
+class IvfPqIndex:
+    """IVF-PQ index: harness-trained codebooks + codes,
+    LUT-in-LDS ADC search on the GPU. L2 only this round."""
+
+    def __init__(self, dim: int, nsub: int, nlist: int, capacity: int = 0,
+                 device: int = 0):
+        err = _Err()
+        dev = (ct.c_int * 1)(device)
+        self.dim, self.nsub, self.nlist = dim, nsub, nlist
+        self._h = lib().moann_ivf_pq_new(dim, nsub, nlist, capacity, dev, 1,
+                                         err.ref)
+        err.check("pq_new")
+        if not self._h:
+            raise MoannError("moann_ivf_pq_new returned NULL")
+
+    def set_centroids(self, centroids):
+        centroids = _as_f32(centroids)
+        err = _Err()
+        lib().moann_ivf_pq_set_centroids(self._h, centroids.ctypes.data,
+                                         centroids.shape[0], err.ref)
+        err.check("pq_set_centroids")
+
+    def set_codebooks(self, cb):
+        cb = _as_f32(cb)
+        err = _Err()
+        lib().moann_ivf_pq_set_codebooks(self._h, cb.ctypes.data, err.ref)
+        err.check("pq_set_codebooks")
+
+    def add_codes(self, codes, ids=None):
+        codes = np.ascontiguousarray(codes, dtype=np.uint8)
+        idp = None
+        if ids is not None:
+            ids = np.ascontiguousarray(ids, dtype=np.int64)
+            idp = ids.ctypes.data
+        err = _Err()
+        lib().moann_ivf_pq_add_codes(self._h, codes.ctypes.data,
+                                     codes.shape[0], idp, err.ref)
+        err.check("pq_add_codes")
+
+    def set_assignments(self, assignments):
+        assignments = np.ascontiguousarray(assignments, dtype=np.int32)
+        err = _Err()
+        lib().moann_ivf_pq_set_assignments(self._h, assignments.ctypes.data,
+                                           assignments.shape[0], err.ref)
+        err.check("pq_set_assignments")
+
+    def build(self):
+        err = _Err()
+        lib().moann_ivf_pq_build(self._h, err.ref)
+        err.check("pq_build")
+
+    def search(self, queries, k: int, nprobe: int):
+        queries = _as_f32(queries)
+        nq = queries.shape[0]
+        err = _Err()
+        res = lib().moann_ivf_pq_search(self._h, queries.ctypes.data, nq,
+                                        queries.shape[1], k,
+                                        SearchParams(nprobe), err.ref)
+        err.check("pq_search")
+        if not res.result_ptr:
+            raise MoannError("pq_search returned NULL result")
+        ids = np.empty((nq, k), dtype=np.int64)
+        dists = np.empty((nq, k), dtype=np.float32)
+        L = lib()
+        L.gpu_ivf_flat_get_neighbors(res.result_ptr, nq * k, ids.ctypes.data)
+        L.gpu_ivf_flat_get_distances(res.result_ptr, nq * k, dists.ctypes.data)
+        L.gpu_ivf_flat_free_result(res.result_ptr)
+        return ids, dists
+
+    def perf(self) -> dict:
+        p = PerfT()
+        lib().moann_ivf_pq_perf(self._h, ct.byref(p))
+        return {f[0]: getattr(p, f[0]) for f in PerfT._fields_}
+
+    def close(self):
+        if getattr(self, "_h", None):
+            err = _Err()
+            lib().moann_ivf_pq_destroy(self._h, err.ref)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
