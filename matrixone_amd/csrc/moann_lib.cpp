@@ -20,7 +20,11 @@
 #include <map>
 #include <unordered_map>
 #include <memory>
+#include <atomic>
+#include <condition_variable>
 #include <mutex>
+#include <thread>
+#include <chrono>
 #include <numeric>
 #include <stdexcept>
 #include <string>
@@ -155,7 +159,33 @@ struct IvfIndex {
     std::map<uint64_t, std::unique_ptr<Result>> jobs;
     uint64_t next_job = 1;
 
+    /* dynamic request batching (cgo/cuvs/dynamic_batching.hpp:19-60 shape:
+     * coalesce concurrent small search_async calls inside a µs window) */
+    std::atomic<int64_t> batch_window_us {0};
+    struct PendingReq {
+        std::vector<float> queries;
+        uint64_t nq;
+        uint32_t limit, nprobe;
+        uint64_t job_id;
+    };
+    std::mutex bmu;
+    std::condition_variable bcv;       /* worker wakeup */
+    std::condition_variable bdone_cv;  /* waiters */
+    std::vector<PendingReq> pending;
+    std::map<uint64_t, std::unique_ptr<Result>> done_jobs;
+    std::thread batcher;
+    bool batcher_stop = false;
+    bool batcher_started = false;
+
     ~IvfIndex() {
+        if (batcher_started) {
+            {
+                std::lock_guard<std::mutex> lk(bmu);
+                batcher_stop = true;
+            }
+            bcv.notify_all();
+            batcher.join();
+        }
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_staging, (void*)d_packed, (void*)d_cent_packed,
                        (void*)d_cent_rows, (void*)d_cnorms,
@@ -551,6 +581,63 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     return res;
 }
 
+void batcher_loop(IvfIndex* ix) {
+    std::unique_lock<std::mutex> lk(ix->bmu);
+    for (;;) {
+        ix->bcv.wait(lk, [&] { return ix->batcher_stop || !ix->pending.empty(); });
+        if (ix->batcher_stop) return;
+        /* window: let more requests coalesce */
+        const auto window =
+            std::chrono::microseconds(ix->batch_window_us.load());
+        ix->bcv.wait_for(lk, window, [&] { return ix->batcher_stop; });
+        if (ix->batcher_stop) return;
+        auto batch = std::move(ix->pending);
+        ix->pending.clear();
+        lk.unlock();
+        /* group by (limit, nprobe); one fused search per group */
+        using PReq = IvfIndex::PendingReq;
+    std::map<std::pair<uint32_t, uint32_t>, std::vector<PReq*>> gs;
+        for (auto& r : batch) gs[{r.limit, r.nprobe}].push_back(&r);
+        for (auto& [key, reqs] : gs) {
+            const auto [limit, nprobe] = key;
+            uint64_t total = 0;
+            for (auto* r : reqs) total += r->nq;
+            std::vector<float> qcat((size_t)total * ix->dim);
+            uint64_t off = 0;
+            for (auto* r : reqs) {
+                memcpy(qcat.data() + off * ix->dim, r->queries.data(),
+                       r->queries.size() * 4);
+                off += r->nq;
+            }
+            std::unique_ptr<Result> merged;
+            std::string err;
+            try {
+                merged = run_search(ix, qcat.data(), false, total, limit,
+                                    nprobe);
+            } catch (const std::exception& e) {
+                err = e.what();
+            }
+            std::lock_guard<std::mutex> g(ix->bmu);
+            off = 0;
+            for (auto* r : reqs) {
+                auto part = std::make_unique<Result>();
+                if (merged) {
+                    part->neighbors.assign(
+                        merged->neighbors.begin() + off * limit,
+                        merged->neighbors.begin() + (off + r->nq) * limit);
+                    part->distances.assign(
+                        merged->distances.begin() + off * limit,
+                        merged->distances.begin() + (off + r->nq) * limit);
+                } /* on error: empty result — search_wait reports it */
+                ix->done_jobs[r->job_id] = std::move(part);
+                off += r->nq;
+            }
+        }
+        lk.lock();
+        ix->bdone_cv.notify_all();
+    }
+}
+
 IvfIndex* IX(gpu_ivf_flat_c h) { return (IvfIndex*)h; }
 
 }  // namespace
@@ -820,6 +907,28 @@ uint64_t gpu_ivf_flat_search_async(gpu_ivf_flat_c h, const void* queries_data,
         auto ix = IX(h);
         if (query_dimension != ix->dim)
             throw std::runtime_error("query dimension mismatch");
+        if (ix->batch_window_us.load() > 0) {
+            /* dynamic batching: enqueue and let the batcher fuse requests
+             * arriving within the window into one device search */
+            std::lock_guard<std::mutex> lk(ix->bmu);
+            if (!ix->batcher_started) {
+                ix->batcher = std::thread(batcher_loop, ix);
+                ix->batcher_started = true;
+            }
+            IvfIndex::PendingReq r;
+            r.queries.assign((const float*)queries_data,
+                             (const float*)queries_data +
+                                 num_queries * ix->dim);
+            r.nq = num_queries;
+            r.limit = limit;
+            r.nprobe = sp.n_probes;
+            std::lock_guard<std::mutex> lk2(ix->mu);
+            r.job_id = ix->next_job++;
+            const uint64_t id = r.job_id;
+            ix->pending.push_back(std::move(r));
+            ix->bcv.notify_one();
+            return id;
+        }
         auto res = run_search(ix, (const float*)queries_data, false,
                               num_queries, limit, sp.n_probes);
         std::lock_guard<std::mutex> lk(ix->mu);
@@ -838,11 +947,27 @@ gpu_ivf_flat_search_res_t gpu_ivf_flat_search_wait(gpu_ivf_flat_c h,
     gpu_ivf_flat_search_res_t out {nullptr};
     try {
         auto ix = IX(h);
-        std::lock_guard<std::mutex> lk(ix->mu);
-        auto it = ix->jobs.find(job_id);
-        if (it == ix->jobs.end()) throw std::runtime_error("unknown job id");
-        out.result_ptr = it->second.release();
-        ix->jobs.erase(it);
+        {
+            std::lock_guard<std::mutex> lk(ix->mu);
+            auto it = ix->jobs.find(job_id);
+            if (it != ix->jobs.end()) {
+                out.result_ptr = it->second.release();
+                ix->jobs.erase(it);
+                return out;
+            }
+        }
+        /* batched job: block until the batcher publishes it */
+        std::unique_lock<std::mutex> lk(ix->bmu);
+        if (!ix->bdone_cv.wait_for(lk, std::chrono::seconds(120), [&] {
+                return ix->done_jobs.count(job_id) != 0;
+            }))
+            throw std::runtime_error("timed out waiting for batched job");
+        auto it = ix->done_jobs.find(job_id);
+        auto res = std::move(it->second);
+        ix->done_jobs.erase(it);
+        if (res->neighbors.empty() && res->distances.empty())
+            throw std::runtime_error("batched search failed");
+        out.result_ptr = res.release();
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_search_wait", e.what());
     }
@@ -1117,6 +1242,18 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
         }
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_load_dir", e.what());
+    }
+}
+
+/* ivf_flat_c.h:91 gpu_ivf_flat_set_batch_window: window (µs) within which
+ * concurrent search_async calls are fused into one device batch
+ * (dynamic_batching.hpp:19-60 precedent). 0 disables. */
+void gpu_ivf_flat_set_batch_window(gpu_ivf_flat_c h, int64_t window_us,
+                                   void* errmsg) {
+    try {
+        IX(h)->batch_window_us.store(window_us);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_set_batch_window", e.what());
     }
 }
 

@@ -116,3 +116,12 @@ def test_cache_load_once_and_remove():
     c.search("k1", FakeImpl, None, None, None)
     assert FakeImpl.loads == 2  # reload after Remove
     c.destroy()
+
+
+def test_batch_window_abi_present():
+    """gpu_ivf_flat_set_batch_window (ivf_flat_c.h:91) is exported and
+    settable without a GPU."""
+    import ctypes as ct
+    from matrixone_amd import engine
+    L = engine.lib()
+    assert hasattr(L, "gpu_ivf_flat_set_batch_window")
