@@ -233,7 +233,7 @@ std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
     {
         auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
         const size_t need = 2 * al8((size_t)njobs * 8) +
-                            3 * al8((size_t)njobs * 4) + 64;
+                            4 * al8((size_t)njobs * 4) + 64;
         MoannDevBuf& wj = ix->w_jobs;
         wj.ensure(std::max(wj.bytes, need)); /* may still hold rank jobs —
             rank stage completed before the sync above */
