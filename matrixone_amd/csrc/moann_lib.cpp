@@ -151,7 +151,6 @@ struct IvfIndex {
     DevBuf w_filter;
     std::vector<int64_t> h_id_by_slot;  /* kept for the membership filter */
     std::unordered_map<int64_t, int64_t> id_to_slot; /* lazy */
-    std::vector<uint8_t> h_jobstage; /* pinned-ish host staging */
 
     std::mutex mu;
     PerfState perf;
@@ -447,13 +446,9 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     }
 
     /* jobs: per list, query tiles of qt; split groups if too few jobs */
-    int64_t tiles_total = 0, groups_total = 0;
-    for (uint32_t l = 0; l < ix->nlist; ++l) {
-        if (!lcount[l]) continue;
-        tiles_total += (lcount[l] + qt - 1) / qt;
-        groups_total += ((lcount[l] + qt - 1) / qt) *
-                        ((ix->list_rows[l] + 63) / 64);
-    }
+    int64_t tiles_total = 0;
+    for (uint32_t l = 0; l < ix->nlist; ++l)
+        if (lcount[l]) tiles_total += (lcount[l] + qt - 1) / qt;
     int split = 1;
     if (tiles_total > 0 && tiles_total < 2048)
         split = (int)std::min<int64_t>(64, (2048 + tiles_total - 1) / tiles_total);

@@ -95,6 +95,8 @@ def lib() -> ct.CDLL:
         f.argtypes = [ct.c_void_p]
     L.gpu_ivf_flat_get_n_list.restype = ct.c_uint32
     L.gpu_ivf_flat_get_n_list.argtypes = [ct.c_void_p]
+    L.gpu_ivf_flat_set_batch_window.argtypes = [ct.c_void_p, ct.c_int64,
+                                                ct.c_void_p]
     L.gpu_ivf_flat_save_dir.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_void_p]
     L.gpu_ivf_flat_load_dir.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_int,
                                         ct.c_void_p]
@@ -297,6 +299,13 @@ class IvfFlatIndex:
             SearchParams(nprobe), err.ref)
         err.check("search_device")
         return self._read_result(res, nq, k, "search_device")
+
+    def set_batch_window(self, window_us: int):
+        """ivf_flat_c.h:91 — fuse concurrent search_async calls arriving
+        within window_us into one device batch (0 disables)."""
+        err = _Err()
+        lib().gpu_ivf_flat_set_batch_window(self._h, window_us, err.ref)
+        err.check("set_batch_window")
 
     def search_async(self, queries, k: int, nprobe: int):
         queries = _as_f32(queries)

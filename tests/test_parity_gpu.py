@@ -261,3 +261,33 @@ def test_save_load_dir(tmp_path):
         ix3.load_dir(str(tmp_path))
     ix2.close()
     ix3.close()
+
+
+def test_dynamic_batching():
+    """search_async calls inside the batch window fuse into one device
+    search and return per-request results identical to the unbatched path
+    (dynamic_batching.hpp:19-60 contract)."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(700))
+    n, d, nlist = 20000, 64, 32
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[:nlist].copy()
+    assign = ((vecs[:, None, :] - cents[None]) ** 2).sum(-1).argmin(1)
+    ix = engine.IvfFlatIndex(d, nlist, capacity=n)
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    qs = rng.standard_normal((16, d), dtype=np.float32)
+    ref, ref_d = ix.search(qs, 5, 4)
+    ix.set_batch_window(2000)
+    jobs = [ix.search_async(qs[i:i + 1], 5, 4) for i in range(16)]
+    for i, j in enumerate(jobs):
+        ix._async_shape = (1, 5)
+        ids, dists = ix.search_wait(j)
+        np.testing.assert_array_equal(ids[0], ref[i])
+        np.testing.assert_array_equal(dists[0], ref_d[i])
+    ix.set_batch_window(0)
+    ids2, _ = ix.search(qs, 5, 4)
+    np.testing.assert_array_equal(ids2, ref)
+    ix.close()
