@@ -304,6 +304,19 @@ void moann_hnsw_search_device(moann_hnsw_c h, const void* queries_dev,
                               uint64_t nq, uint32_t dim, uint32_t ef,
                               uint32_t k, int64_t* out_keys, float* out_dists,
                               void* errmsg);
+/* Membership-filtered HNSW search — the GPU counterpart of
+ * usearchex_filtered_search_with_membership (cgo/usearchex.h:38-42,
+ * cgo/usearchex.c:51-70): the membership test gates the REPORTABLE set
+ * exactly like usearch's search predicate (traversal continues through
+ * filtered nodes, index.hpp:4712-4752). slot_bitset: one bit per slot; the
+ * host evaluates its docfilter handle (bloom/croaring/cbitmap/sorted64)
+ * over the slot->key map once per filter, the eval_filter_bitmap_cpu shape
+ * (cgo/cuvs/filter.hpp:855-960). */
+void moann_hnsw_search_filtered(moann_hnsw_c h, const float* queries,
+                                uint64_t nq, uint32_t dim, uint32_t ef,
+                                uint32_t k, const uint32_t* slot_bitset,
+                                int64_t* out_keys, float* out_dists,
+                                void* errmsg);
 void moann_hnsw_perf(moann_hnsw_c h, moann_perf_t* out);
 void moann_hnsw_perf_reset(moann_hnsw_c h);
 void moann_hnsw_destroy(moann_hnsw_c h, void* errmsg);

@@ -42,6 +42,10 @@ __device__ __forceinline__ float hu2f(uint32_t u) {
 
 #define EXPANDED_BIT 0x80000000ull
 #define SLOT_MASK 0x7FFFFFFFull
+/* filtered variant: bit30 of the slot word = "passes the membership
+ * filter"; slots then fit 30 bits (host rejects n >= 2^30) */
+#define PASS_BIT 0x40000000ull
+#define FSLOT_MASK 0x3FFFFFFFull
 
 /* one wave computes the metric distance of one row vs the LDS query */
 template <int METRIC>
@@ -83,7 +87,7 @@ __device__ float wave_dist(const float* __restrict__ vecs, int64_t slot,
     return (float)(1.0 - sim);
 }
 
-template <int METRIC>
+template <int METRIC, bool FILTERED>
 __global__ __launch_bounds__(256) void hnsw_search_kernel(
     const float* __restrict__ vecs, int dpad, int64_t n,
     const int64_t* __restrict__ l0_offs, const uint32_t* __restrict__ l0_nbrs,
@@ -91,7 +95,9 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
     const int32_t* __restrict__ up_offs, const uint32_t* __restrict__ up_nbrs,
     int max_level, int64_t entry_slot,
     const float* __restrict__ queries, const float* __restrict__ qnorms,
-    int ef, int k, int maxdeg_p2,
+    int ef, int efc /* list capacity: ef, or 2*ef when FILTERED */,
+    int k, int maxdeg_p2,
+    const uint32_t* __restrict__ slot_filter, /* FILTERED only: bit/slot */
     uint32_t* __restrict__ visited, int64_t vwords,
     const int64_t* __restrict__ key_of_slot,
     int64_t* __restrict__ out_ids, float* __restrict__ out_dists,
@@ -102,14 +108,16 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
     const int wave = tid >> 6, lane = tid & 63;
 
     float* ldsq = (float*)smem;                               /* [dpad]    */
-    uint64_t* list = (uint64_t*)(smem + ((dpad * 4 + 7) & ~7));/* [ef]     */
-    uint64_t* merged = list + ef;                             /* [ef]      */
-    uint64_t* newk = merged + ef;                             /* [maxdegP2]*/
+    uint64_t* list = (uint64_t*)(smem + ((dpad * 4 + 7) & ~7));/* [efc]    */
+    uint64_t* merged = list + efc;                            /* [efc]     */
+    uint64_t* newk = merged + efc;                            /* [maxdegP2]*/
     uint32_t* stage = (uint32_t*)(newk + maxdeg_p2);          /* [maxdegP2]*/
     float* wdist = (float*)(stage + maxdeg_p2);               /* [8]       */
     /* scalars */
-    int* sc = (int*)(wdist + 8); /* 0:list_sz 1:nbr_idx 2:changed 3:n_new
-                                    4:best_idx 5:evals 6:best_d(float) */
+    int* sc = (int*)(wdist + 8); /* 0:list_sz 1:nbr_idx 2:kept 3:n_new
+                                    4:best_idx 5:evals 6:best_d/radius
+                                    7:pass_cnt */
+    int* cnt_scratch = sc + 8;                            /* [256] */
 
     const float* qsrc = queries + (int64_t)q * dpad;
     for (int e = tid; e < dpad; e += blockDim.x) ldsq[e] = qsrc[e];
@@ -164,35 +172,89 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
     }
     const float cur_d = *best_d;
 
-    /* ---- base-level beam (search_to_find_in_base_) ---- */
-    /* init: visited(entry), list = {entry} */
+    /* ---- base-level beam (search_to_find_in_base_) ----
+     * FILTERED follows usearch's predicate semantics exactly
+     * (index.hpp:4712-4752): the membership test gates entry into `top`
+     * (= the reportable set and the radius) while traversal continues
+     * through filtered nodes; the list holds BOTH kinds, flagged, with
+     * capacity efc = 2*ef (entries past efc are dropped — they are farther
+     * than the 2ef-th best candidate; usearch's unbounded `next` would keep
+     * them, a bounded-memory deviation that matters only when more than ef
+     * filtered candidates interleave the passing front). */
+    const auto passes = [&](int64_t slot) -> bool {
+        if (!FILTERED) return true;
+        return (slot_filter[slot >> 5] >> (slot & 31)) & 1u;
+    };
     if (tid == 0) {
         const int64_t w = cur >> 5;
         vis[w] |= (1u << (cur & 31));
-        list[0] = ((uint64_t)hf2u(cur_d) << 32) | (uint64_t)cur;
+        uint64_t kv = ((uint64_t)hf2u(cur_d) << 32) | (uint64_t)cur;
+        if (FILTERED && passes(cur)) kv |= PASS_BIT;
+        list[0] = kv;
         sc[0] = 1;
     }
     __syncthreads();
+    const uint64_t slot_mask = FILTERED ? FSLOT_MASK : SLOT_MASK;
 
-    const int max_expansions = 16 * ef + 64; /* bounded spin (guide rule) */
+    const int max_expansions = 16 * efc + 64; /* bounded spin (guide rule) */
     for (int it = 0; it < max_expansions; ++it) {
+        const int sz = sc[0];
+        /* FILTERED: passing count + the ef-th passing dist (= radius).
+         * contiguous per-thread blocks so thread 0 can localise the ef-th. */
+        if (FILTERED) {
+            const int per = (sz + blockDim.x - 1) / blockDim.x;
+            int cnt = 0;
+            for (int i = tid * per; i < min((tid + 1) * per, sz); ++i)
+                if (list[i] & PASS_BIT) cnt++;
+            cnt_scratch[tid] = cnt;
+            __syncthreads();
+            if (tid == 0) {
+                int acc = 0, radius_bits = (int)0xFFFFFFFFu; /* +inf key */
+                int total = 0;
+                for (int t = 0; t < (int)blockDim.x; ++t)
+                    total += cnt_scratch[t];
+                for (int t = 0; t < (int)blockDim.x && acc + 1 <= ef; ++t) {
+                    const int c = cnt_scratch[t];
+                    if (acc + c >= ef) {
+                        int need = ef - acc;
+                        for (int i = t * per; i < min((t + 1) * per, sz);
+                             ++i)
+                            if (list[i] & PASS_BIT && --need == 0) {
+                                radius_bits = (int)(list[i] >> 32);
+                                break;
+                            }
+                        acc = ef;
+                        break;
+                    }
+                    acc += c;
+                }
+                sc[6] = radius_bits; /* u32 key of the radius */
+                sc[7] = total;       /* passing count */
+            }
+            __syncthreads();
+        }
+        const uint32_t radius_key = FILTERED ? (uint32_t)sc[6] : 0;
+        const int pass_cnt = FILTERED ? sc[7] : 0;
+
         /* find best unexpanded entry (list ascending => first unexpanded) */
         if (tid == 0) sc[4] = INT_MAX;
         __syncthreads();
-        const int sz = sc[0];
         for (int i = tid; i < sz; i += blockDim.x)
             if (!(list[i] & EXPANDED_BIT)) atomicMin(&sc[4], i);
         __syncthreads();
         const int bi = sc[4];
         if (bi == INT_MAX) break;
-        const int64_t c_slot = (int64_t)(list[bi] & SLOT_MASK);
+        if (FILTERED && pass_cnt >= ef &&
+            (uint32_t)(list[bi] >> 32) > radius_key)
+            break; /* usearch: candidate.dist > radius && top full */
+        const int64_t c_slot = (int64_t)(list[bi] & slot_mask);
         if (tid == 0) list[bi] |= EXPANDED_BIT;
         __syncthreads();
 
         /* stage unvisited neighbors */
         const int64_t a0 = l0_offs[c_slot], a1 = l0_offs[c_slot + 1];
         const int m = (int)(a1 - a0);
-        if (tid == 0) sc[3] = 0;
+        if (tid == 0) { sc[3] = 0; sc[2] = 0; }
         __syncthreads();
         for (int i = tid; i < m; i += blockDim.x) {
             const uint32_t nb = l0_nbrs[a0 + i];
@@ -206,21 +268,33 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
         const int mp = sc[3];
         if (tid == 0) atomicAdd(&sc[5], mp);
 
-        /* distances: 4 waves round-robin over the staged slots */
+        /* distances: 4 waves round-robin; FILTERED drops candidates that
+         * fail usearch's push condition (top full && d >= radius) and tags
+         * the rest with the membership bit */
         for (int r = 0; r < mp; r += 4) {
             const int i = r + wave;
             float d = 0.f;
             if (i < mp)
                 d = wave_dist<METRIC>(vecs, (int64_t)stage[i], dpad, ldsq, qn);
-            if (i < mp && lane == 0)
-                newk[i] = ((uint64_t)hf2u(d) << 32) | (uint64_t)stage[i];
+            if (i < mp && lane == 0) {
+                const uint32_t dk = hf2u(d);
+                if (FILTERED && pass_cnt >= ef && dk >= radius_key) {
+                    /* not pushed (neither next nor top) */
+                } else {
+                    const int pos = atomicAdd(&sc[2], 1);
+                    uint64_t kv = ((uint64_t)dk << 32) | (uint64_t)stage[i];
+                    if (FILTERED && passes((int64_t)stage[i])) kv |= PASS_BIT;
+                    newk[pos] = kv;
+                }
+            }
             __syncthreads();
         }
-        for (int i = mp + tid; i < maxdeg_p2; i += blockDim.x)
+        const int kept = sc[2]; /* == mp when unfiltered */
+        for (int i = kept + tid; i < maxdeg_p2; i += blockDim.x)
             newk[i] = ~0ull;
         __syncthreads();
 
-        /* bitonic sort of the <=maxdeg_p2 new keys */
+        /* bitonic sort of the <=maxdeg_p2 new keys (mask out flag bits) */
         for (int ks = 2; ks <= maxdeg_p2; ks <<= 1) {
             for (int jj = ks >> 1; jj > 0; jj >>= 1) {
                 for (int i = tid; i < maxdeg_p2; i += blockDim.x) {
@@ -235,10 +309,10 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
             }
         }
 
-        /* merge-path: C = first min(sz+mp, ef) of merge(list[0..sz), newk[0..mp)) */
-        const int outn = min(sz + mp, ef);
+        /* merge-path: C = first min(sz+kept, efc) of merge(list, newk) */
+        const int outn = min(sz + kept, efc);
         for (int i = tid; i < outn; i += blockDim.x) {
-            int lo = max(0, i - mp), hi = min(i, sz);
+            int lo = max(0, i - kept), hi = min(i, sz);
             while (lo < hi) {
                 const int mid = (lo + hi) >> 1;
                 if ((list[mid] & ~EXPANDED_BIT) <=
@@ -249,7 +323,7 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
             }
             const int a = lo, b = i - lo;
             const uint64_t va = a < sz ? list[a] : ~0ull;
-            const uint64_t vb = b < mp ? newk[b] : ~0ull;
+            const uint64_t vb = b < kept ? newk[b] : ~0ull;
             merged[i] = (va & ~EXPANDED_BIT) <= (vb & ~EXPANDED_BIT) ? va : vb;
         }
         __syncthreads();
@@ -258,8 +332,30 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
         __syncthreads();
     }
 
-    /* output first k (ascending), padded */
+    /* output first k (ascending, PASSING only when FILTERED), padded */
     const int sz = sc[0];
+    if (FILTERED) {
+        /* order-preserving compaction of passing entries (thread 0: <=efc
+         * scan once per query) */
+        if (tid == 0) {
+            int o = 0;
+            for (int i = 0; i < sz && o < k; ++i)
+                if (list[i] & PASS_BIT) merged[o++] = list[i];
+            sc[1] = o;
+        }
+        __syncthreads();
+        const int o = sc[1];
+        for (int i = tid; i < k; i += blockDim.x) {
+            if (i < o) {
+                const uint64_t kv = merged[i];
+                out_ids[(int64_t)q * k + i] = key_of_slot[kv & slot_mask];
+                out_dists[(int64_t)q * k + i] = hu2f((uint32_t)(kv >> 32));
+            } else {
+                out_ids[(int64_t)q * k + i] = -1;
+                out_dists[(int64_t)q * k + i] = FLT_MAX;
+            }
+        }
+    } else
     for (int i = tid; i < k; i += blockDim.x) {
         if (i < sz) {
             const uint64_t kv = list[i];
@@ -279,33 +375,39 @@ void launch_hnsw_search(int metric, const float* vecs, int dpad, int64_t n,
                         const int32_t* up_offs, const uint32_t* up_nbrs,
                         int max_level, int64_t entry_slot,
                         const float* queries, const float* qnorms, int nq,
-                        int ef, int k, int maxdeg_p2, uint32_t* visited,
+                        int ef, int k, int maxdeg_p2,
+                        const uint32_t* slot_filter, uint32_t* visited,
                         int64_t vwords, const int64_t* key_of_slot,
                         int64_t* out_ids, float* out_dists,
                         unsigned long long* eval_count, size_t* shmem_out,
                         hipStream_t stream) {
-    const size_t shmem = ((dpad * 4 + 7) & ~7) + (size_t)ef * 8 * 2 +
+    const bool filtered = slot_filter != nullptr;
+    const int efc = filtered ? 2 * ef : ef;
+    const size_t shmem = ((dpad * 4 + 7) & ~7) + (size_t)efc * 8 * 2 +
                          (size_t)maxdeg_p2 * 8 + (size_t)maxdeg_p2 * 4 +
-                         8 * 4 + 8 * 4 + 16;
+                         8 * 4 + 8 * 4 + 256 * 4 + 16;
     if (shmem_out) *shmem_out = shmem;
     if (!nq) return;
 #define HARGS vecs, dpad, n, l0_offs, l0_nbrs, u_idx, u_count, up_offs,     \
-    up_nbrs, max_level, entry_slot, queries, qnorms, ef, k, maxdeg_p2,      \
-    visited, vwords, key_of_slot, out_ids, out_dists, eval_count
-    switch (metric) {
-    case U_L2SQ:
-        hipLaunchKernelGGL((hnsw_search_kernel<U_L2SQ>), dim3(nq), dim3(256),
-                           shmem, stream, HARGS);
-        break;
-    case U_IP:
-        hipLaunchKernelGGL((hnsw_search_kernel<U_IP>), dim3(nq), dim3(256),
-                           shmem, stream, HARGS);
-        break;
-    default:
-        hipLaunchKernelGGL((hnsw_search_kernel<U_COS>), dim3(nq), dim3(256),
-                           shmem, stream, HARGS);
-        break;
+    up_nbrs, max_level, entry_slot, queries, qnorms, ef, efc, k, maxdeg_p2, \
+    slot_filter, visited, vwords, key_of_slot, out_ids, out_dists, eval_count
+#define HLAUNCH(M, F)                                                       \
+    hipLaunchKernelGGL((hnsw_search_kernel<M, F>), dim3(nq), dim3(256),     \
+                       shmem, stream, HARGS)
+    if (filtered) {
+        switch (metric) {
+        case U_L2SQ: HLAUNCH(U_L2SQ, true); break;
+        case U_IP: HLAUNCH(U_IP, true); break;
+        default: HLAUNCH(U_COS, true); break;
+        }
+    } else {
+        switch (metric) {
+        case U_L2SQ: HLAUNCH(U_L2SQ, false); break;
+        case U_IP: HLAUNCH(U_IP, false); break;
+        default: HLAUNCH(U_COS, false); break;
+        }
     }
+#undef HLAUNCH
 #undef HARGS
 }
 

@@ -23,7 +23,8 @@ void launch_hnsw_search(int metric, const float* vecs, int dpad, int64_t n,
                         const int32_t* up_offs, const uint32_t* up_nbrs,
                         int max_level, int64_t entry_slot,
                         const float* queries, const float* qnorms, int nq,
-                        int ef, int k, int maxdeg_p2, uint32_t* visited,
+                        int ef, int k, int maxdeg_p2,
+                        const uint32_t* slot_filter, uint32_t* visited,
                         int64_t vwords, const int64_t* key_of_slot,
                         int64_t* out_ids, float* out_dists,
                         unsigned long long* eval_count, size_t* shmem_out,
@@ -87,7 +88,8 @@ struct HnswIndex {
     uint32_t* d_up_nbrs = nullptr;
     int64_t* d_keys = nullptr;
 
-    HnswDevBuf w_queries, w_qnorms, w_visited, w_out_ids, w_out_dists, w_evals;
+    HnswDevBuf w_queries, w_qnorms, w_visited, w_out_ids, w_out_dists,
+        w_evals, w_filter;
 
     std::mutex mu;
     /* perf: whole-kernel time under scan_*, evals -> bytes */
@@ -111,7 +113,8 @@ HnswIndex* HX(moann_hnsw_c h) { return (HnswIndex*)h; }
 
 void run_hnsw_search(HnswIndex* ix, const float* queries, bool on_device,
                      uint64_t nq, uint32_t ef, uint32_t k, int64_t* out_keys,
-                     float* out_dists) {
+                     float* out_dists,
+                     const uint32_t* slot_filter_host = nullptr) {
     std::lock_guard<std::mutex> lk(ix->mu);
     if (!ix->built) throw std::runtime_error("index not built");
     if (nq == 0) return;
@@ -149,6 +152,16 @@ void run_hnsw_search(HnswIndex* ix, const float* queries, bool on_device,
     ix->w_out_dists.ensure((size_t)nq * k * 4);
     ix->w_evals.ensure(8);
     HIP_CHECK(hipMemsetAsync(ix->w_evals.ptr, 0, 8, s));
+    const uint32_t* d_filter = nullptr;
+    if (slot_filter_host) {
+        if (ix->n >= (1ull << 30))
+            throw std::runtime_error("filtered search needs n < 2^30");
+        const size_t fw = (size_t)((ix->n + 31) / 32);
+        ix->w_filter.ensure(fw * 4);
+        HIP_CHECK(hipMemcpyAsync(ix->w_filter.ptr, slot_filter_host, fw * 4,
+                                 hipMemcpyHostToDevice, s));
+        d_filter = ix->w_filter.as<uint32_t>();
+    }
 
     int maxdeg_p2 = 64;
     while (maxdeg_p2 < ix->maxdeg) maxdeg_p2 <<= 1;
@@ -159,7 +172,7 @@ void run_hnsw_search(HnswIndex* ix, const float* queries, bool on_device,
                        ix->d_l0_offs, ix->d_l0_nbrs, ix->d_u_idx, ix->u_count,
                        ix->d_up_offs, ix->d_up_nbrs, ix->max_level,
                        ix->entry_slot, d_q, d_qn, (int)nq, (int)ef, (int)k,
-                       maxdeg_p2, ix->w_visited.as<uint32_t>(), vwords,
+                       maxdeg_p2, d_filter, ix->w_visited.as<uint32_t>(), vwords,
                        ix->d_keys, ix->w_out_ids.as<int64_t>(),
                        ix->w_out_dists.as<float>(),
                        ix->w_evals.as<unsigned long long>(), &shmem, s);
@@ -337,6 +350,28 @@ void moann_hnsw_search_device(moann_hnsw_c h, const void* queries_dev,
                         out_keys, out_dists);
     } catch (const std::exception& e) {
         hn_set_errmsg(errmsg, "moann_hnsw_search_device", e.what());
+    }
+}
+
+/* Membership-filtered HNSW search — the GPU counterpart of
+ * usearchex_filtered_search_with_membership (cgo/usearchex.c:51-70): the
+ * membership test gates the reportable set exactly like usearch's predicate
+ * (index.hpp:4712-4752). slot_bitset: one bit per SLOT; the host evaluates
+ * its docfilter (bloom/croaring/cbitmap/sorted64 handle) over key_of_slot
+ * once per filter, the cuVS eval_filter_bitmap_cpu shape. */
+void moann_hnsw_search_filtered(moann_hnsw_c h, const float* queries,
+                                uint64_t nq, uint32_t dim, uint32_t ef,
+                                uint32_t k, const uint32_t* slot_bitset,
+                                int64_t* out_keys, float* out_dists,
+                                void* errmsg) {
+    try {
+        auto ix = HX(h);
+        if (dim != ix->dim) throw std::runtime_error("query dim mismatch");
+        if (!slot_bitset) throw std::runtime_error("null filter bitset");
+        run_hnsw_search(ix, queries, false, nq, ef, k, out_keys, out_dists,
+                        slot_bitset);
+    } catch (const std::exception& e) {
+        hn_set_errmsg(errmsg, "moann_hnsw_search_filtered", e.what());
     }
 }
 

@@ -54,6 +54,9 @@ def _bind(L):
                                     ct.c_uint32, ct.c_uint32, ct.c_uint32,
                                     ct.c_void_p, ct.c_void_p, ct.c_void_p]
     L.moann_hnsw_search_device.argtypes = L.moann_hnsw_search.argtypes
+    L.moann_hnsw_search_filtered.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
+        ct.c_uint32, ct.c_void_p, ct.c_void_p, ct.c_void_p]
     L.moann_hnsw_perf.argtypes = [ct.c_void_p, ct.POINTER(engine.PerfT)]
     L.moann_hnsw_perf_reset.argtypes = [ct.c_void_p]
     L.moann_hnsw_destroy.argtypes = [ct.c_void_p, ct.c_void_p]
@@ -148,6 +151,24 @@ class HnswIndex:
         err.check("hnsw_search")
         return out_ids, out_dists
 
+    def search_filtered(self, queries, ef: int, k: int,
+                        slot_bitset: np.ndarray):
+        """Membership-filtered search (usearchex semantics; §8a4)."""
+        L = _bind(engine.lib())
+        queries = np.ascontiguousarray(queries, dtype=np.float32)
+        slot_bitset = np.ascontiguousarray(slot_bitset, dtype=np.uint32)
+        nq = queries.shape[0]
+        out_ids = np.empty((nq, k), dtype=np.int64)
+        out_dists = np.empty((nq, k), dtype=np.float32)
+        err = engine._Err()
+        L.moann_hnsw_search_filtered(self._h, queries.ctypes.data, nq,
+                                     queries.shape[1], ef, k,
+                                     slot_bitset.ctypes.data,
+                                     out_ids.ctypes.data,
+                                     out_dists.ctypes.data, err.ref)
+        err.check("hnsw_search_filtered")
+        return out_ids, out_dists
+
     def search_device(self, queries_tensor, ef: int, k: int):
         L = _bind(engine.lib())
         nq, qdim = queries_tensor.shape
@@ -222,3 +243,45 @@ class HnswSearch:
         if self.index is not None:
             self.index.close()
             self.index = None
+
+
+class MultiModelHnswSearch:
+    """Multi-model fan-out — the reference splits an HNSW index into
+    mini-models of <= hnsw_max_index_capacity vectors (sync.go:98-106) and
+    searches them concurrently, merging in a bounded SearchResultSafeHeap
+    (hnsw/search.go:89-183, per-shard k clamp :189-201). Here: one GPU index
+    per model, sequential device searches (one stream), dense merge with the
+    cpu_topk_merge_sharded semantics."""
+
+    def __init__(self, graphs, op_type: str = "vector_l2_ops",
+                 ef_search: int = 64, device: int = 0):
+        self.searchers = [HnswSearch(g, op_type, ef_search, device)
+                          for g in graphs]
+
+    def Load(self, proc=None):
+        for sch in self.searchers:
+            sch.Load()
+
+    def Search(self, proc, query, rt):
+        blocks_i, blocks_d = [], []
+        for sch in self.searchers:
+            ids, d64 = sch.Search(proc, query, rt)
+            blocks_i.append(ids)
+            blocks_d.append(d64)
+        ids = np.stack(blocks_i)   # [models][nq][k]
+        d = np.stack(blocks_d)
+        nm, nq, k = ids.shape
+        flat_i = ids.transpose(1, 0, 2).reshape(nq, nm * k)
+        flat_d = d.transpose(1, 0, 2).reshape(nq, nm * k)
+        fmax = np.finfo(np.float64).max
+        flat_d = np.where(flat_i < 0, fmax, flat_d)
+        order = np.lexsort((flat_i, flat_d), axis=1)[:, :k]
+        out_d = np.take_along_axis(flat_d, order, 1)
+        out_i = np.take_along_axis(flat_i, order, 1)
+        out_i[out_d >= fmax] = -1
+        out_d = np.where(out_i < 0, np.float64(np.finfo(np.float32).max), out_d)
+        return out_i, out_d
+
+    def Destroy(self):
+        for sch in self.searchers:
+            sch.Destroy()

@@ -83,6 +83,12 @@ def lib():
         L.ref_hnsw_save.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_void_p]
         L.ref_hnsw_load.restype = ct.c_void_p
         L.ref_hnsw_load.argtypes = [ct.c_char_p, ct.c_void_p]
+        L.ref_hnsw_search_filtered.restype = ct.c_size_t
+        L.ref_hnsw_search_filtered.argtypes = [ct.c_void_p, ct.c_void_p,
+                                               ct.c_size_t, ct.c_size_t,
+                                               ct.c_void_p, ct.c_uint64,
+                                               ct.c_void_p, ct.c_void_p,
+                                               ct.c_void_p]
         L.ref_hnsw_export_bulk.restype = ct.c_int64
         L.ref_hnsw_export_bulk.argtypes = [ct.c_void_p, ct.c_size_t,
                                            ct.c_void_p, ct.c_void_p,
@@ -142,6 +148,27 @@ class RefHnsw:
         ids = out_k.astype(np.int64)
         for i in range(nq):
             ids[i, counts[i]:] = -1
+        return ids, out_d
+
+    def search_filtered(self, queries, ef, k, key_bits: np.ndarray):
+        """usearch filtered_search with a key-bitset predicate (the
+        usearchex membership-callback shape, cgo/usearchex.c:27-70)."""
+        queries = np.ascontiguousarray(queries, dtype=np.float32)
+        key_bits = np.ascontiguousarray(key_bits, dtype=np.uint32)
+        nbits = key_bits.size * 32
+        nq = queries.shape[0]
+        out_k = np.zeros((nq, k), dtype=np.uint64)
+        out_d = np.full((nq, k), np.float32(np.finfo(np.float32).max))
+        err = ct.c_char_p(None)
+        ids = np.full((nq, k), -1, dtype=np.int64)
+        for i in range(nq):
+            cnt = lib().ref_hnsw_search_filtered(
+                self._h, queries[i].ctypes.data, ef, k, key_bits.ctypes.data,
+                nbits, out_k[i].ctypes.data, out_d[i].ctypes.data,
+                ct.byref(err))
+            if err.value:
+                raise RuntimeError(err.value)
+            ids[i, :cnt] = out_k[i, :cnt].astype(np.int64)
         return ids, out_d
 
     def export_graph(self) -> "HnswGraph":
@@ -221,7 +248,8 @@ def _usearch_dist(metric, a, b):
     return np.float32(1.0 - ab / (np.sqrt(a2) * np.sqrt(b2)))
 
 
-def oracle_search(g: HnswGraph, query: np.ndarray, ef: int, k: int):
+def oracle_search(g: HnswGraph, query: np.ndarray, ef: int, k: int,
+                  allowed=None):
     """The numpy restatement (see module docstring for cites)."""
     q = query.astype(np.float32)
 
@@ -242,11 +270,19 @@ def oracle_search(g: HnswGraph, query: np.ndarray, ef: int, k: int):
                     closest = int(nb)
                     changed = True
 
-    # base-level beam (search_to_find_in_base_, index.hpp:4688)
+    # base-level beam (search_to_find_in_base_, index.hpp:4688). `allowed`
+    # (a key predicate) gates TOP insertion only — traversal pushes to next
+    # under the same (top.size < ef or d < radius) condition either way
+    # (index.hpp:4712-4752).
+    def passes(slot):
+        return allowed is None or allowed(int(g.keys[slot]))
+
     visited = {closest}
     nxt = [(closest_d, closest)]          # min-heap
-    top = [(-closest_d, closest)]         # bounded max-heap (size <= ef)
+    top = []                              # bounded max-heap (size <= ef)
     radius = closest_d
+    if passes(closest):
+        top = [(-closest_d, closest)]
     while nxt:
         cd, c = nxt[0]
         if cd > radius and len(top) == ef:
@@ -260,11 +296,12 @@ def oracle_search(g: HnswGraph, query: np.ndarray, ef: int, k: int):
             d = dist(nb)
             if len(top) < ef or d < radius:
                 heapq.heappush(nxt, (d, nb))
-                if len(top) == ef:
-                    heapq.heappushpop(top, (-d, nb))
-                else:
-                    heapq.heappush(top, (-d, nb))
-                radius = -top[0][0]
+                if passes(nb):
+                    if len(top) == ef:
+                        heapq.heappushpop(top, (-d, nb))
+                    else:
+                        heapq.heappush(top, (-d, nb))
+                    radius = -top[0][0]
     out = sorted((-d, s) for d, s in top)[:k]
     ids = np.full(k, -1, dtype=np.int64)
     dists = np.full(k, np.finfo(np.float32).max, dtype=np.float32)

@@ -121,6 +121,25 @@ size_t ref_hnsw_search(void* hp, const float* query, size_t ef, size_t k,
     return r.dump_to(out_keys, out_dists);
 }
 
+/* filtered search: keep a candidate only when its KEY's bit is set —
+ * index_dense passes the KEY to the predicate (index_dense.hpp:2245-2247),
+ * exactly the shape of usearchex_filtered_search_with_membership's
+ * per-key docfilter callback (cgo/usearchex.c:27-70). */
+size_t ref_hnsw_search_filtered(void* hp, const float* query, size_t ef,
+                                size_t k, const uint32_t* key_bits,
+                                uint64_t nbits, uint64_t* out_keys,
+                                float* out_dists, char** err) {
+    auto h = (ref_hnsw_t*)hp;
+    h->index.change_expansion_search(ef);
+    auto pred = [key_bits, nbits](uint64_t key) {
+        if (key >= nbits) return false;
+        return ((key_bits[key >> 5] >> (key & 31)) & 1u) != 0;
+    };
+    auto r = h->index.filtered_search(query, k, pred);
+    if (!r) { set_err(err, r.error.release()); return 0; }
+    return r.dump_to(out_keys, out_dists);
+}
+
 size_t ref_hnsw_size(void* hp) { return ((ref_hnsw_t*)hp)->index.size(); }
 
 int ref_hnsw_max_level(void* hp) {

@@ -104,3 +104,73 @@ def test_hnsw_keys_mapping():
     assert ((gids[valid] - 1000) % 7 == 0).all()
     assert (gids == ref_ids).mean() > 0.9
     gpu.close()
+
+
+def test_hnsw_filtered_gpu():
+    """Filtered GPU beam vs usearch filtered_search and the oracle
+    restatement, same graph and key-bitset filter (§8a4)."""
+    from matrixone_amd.hnsw import HnswIndex
+    ix, qs = _mk(H.METRIC_L2SQ, n=5000, d=64)
+    g = ix.export_graph()
+    rng = np.random.Generator(np.random.PCG64(77))
+    n = len(g.levels)
+    # keys here are identity (slots == keys)
+    key_bits = np.zeros((n + 31) // 32, dtype=np.uint32)
+    idxs = np.nonzero(rng.random(n) < 0.3)[0]
+    np.bitwise_or.at(key_bits, idxs // 32,
+                     (np.uint32(1) << (idxs % 32).astype(np.uint32)))
+    # slot bitset for the GPU: bit per slot, set when key passes
+    slot_bits = np.zeros((n + 31) // 32, dtype=np.uint32)
+    pass_slots = np.nonzero(np.isin(g.keys.astype(np.int64), idxs))[0]
+    np.bitwise_or.at(slot_bits, pass_slots // 32,
+                     (np.uint32(1) << (pass_slots % 32).astype(np.uint32)))
+    gpu = HnswIndex(_graph_data(g))
+    ef, k = 48, 10
+    gids, gdists = gpu.search_filtered(qs, ef, k, slot_bits)
+    ref_ids, ref_d = ix.search_filtered(qs, ef, k, key_bits)
+    aset = set(int(i) for i in idxs)
+    valid = gids[gids >= 0]
+    assert all(int(x) in aset for x in valid), "filtered-out id returned"
+    agree = (gids == ref_ids).mean()
+    assert agree > 0.9, agree
+    # unfiltered still intact
+    u_ids, _ = gpu.search(qs, ef, k)
+    ru_ids, _ = ix.search(qs, ef=ef, k=k)
+    assert (u_ids == ru_ids).mean() > 0.95
+    gpu.close()
+
+
+def test_multi_model_fanout():
+    """Multi-model search (hnsw/search.go:89-183): merged results over two
+    disjoint mini-models equal usearch over each model merged the same way."""
+    from matrixone_amd.hnsw import MultiModelHnswSearch
+    from matrixone_amd.ivfflat import RuntimeConfig
+    rng = np.random.Generator(np.random.PCG64(21))
+    d, nq, k = 48, 8, 10
+    graphs, refs = [], []
+    for m in range(2):
+        vecs = rng.standard_normal((2500, d), dtype=np.float32)
+        keys = np.arange(2500, dtype=np.uint64) + m * 1_000_000
+        ix = H.RefHnsw(d, metric=H.METRIC_L2SQ, connectivity=16,
+                       expansion_add=64, expansion_search=48, capacity=2500,
+                       threads=2)
+        ix.add(vecs, keys=keys)
+        graphs.append(_graph_data(ix.export_graph()))
+        refs.append(ix)
+    qs = rng.standard_normal((nq, d), dtype=np.float32)
+    mm = MultiModelHnswSearch(graphs, op_type="vector_l2sq_ops", ef_search=48)
+    mm.Load()
+    ids, d64 = mm.Search(None, qs, RuntimeConfig(limit=k,
+                                                 orig_func_name="l2_distance_sq"))
+    # reference merge of per-model usearch results
+    bi, bd = [], []
+    for ix in refs:
+        ri, rd = ix.search(qs, ef=48, k=k)
+        bi.append(ri)
+        bd.append(rd)
+    from oracle import oracle as orc
+    mi, md = orc.topk_merge(np.stack(bi), np.stack(bd).astype(np.float32), k)
+    agree = (ids == mi).mean()
+    assert agree > 0.95, agree
+    np.testing.assert_allclose(d64, md.astype(np.float64), rtol=3e-5, atol=3e-5)
+    mm.Destroy()

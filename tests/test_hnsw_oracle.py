@@ -69,3 +69,29 @@ def test_ref_hnsw_search_recall():
     recall = np.mean([len(set(ids[i]).intersection(gt[i])) / 10
                       for i in range(16)])
     assert recall > 0.9, recall
+
+
+def test_filtered_oracle_matches_usearch():
+    """Filtered search: the numpy restatement (predicate gates `top` only)
+    matches usearch's own filtered_search on the same graph."""
+    rng = np.random.Generator(np.random.PCG64(5))
+    n, d = 4000, 48
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    ix = H.RefHnsw(d, metric=H.METRIC_L2SQ, connectivity=16, expansion_add=64,
+                   expansion_search=32, capacity=n, threads=4)
+    ix.add(vecs)
+    qs = rng.standard_normal((12, d), dtype=np.float32)
+    key_bits = np.zeros((n + 31) // 32, dtype=np.uint32)
+    idxs = np.nonzero(rng.random(n) < 0.25)[0]
+    np.bitwise_or.at(key_bits, idxs // 32,
+                     (np.uint32(1) << (idxs % 32).astype(np.uint32)))
+    ref_ids, _ = ix.search_filtered(qs, 32, 10, key_bits)
+    g = ix.export_graph()
+    aset = set(int(i) for i in idxs)
+    pos = 0
+    for i in range(12):
+        oi, _ = H.oracle_search(g, qs[i], 32, 10, allowed=lambda kk: kk in aset)
+        pos += int((oi == ref_ids[i]).sum())
+    assert pos >= 118, pos
+    valid = ref_ids[ref_ids >= 0]
+    assert all(int(x) in aset for x in valid)
