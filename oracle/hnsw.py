@@ -309,3 +309,64 @@ def oracle_search(g: HnswGraph, query: np.ndarray, ef: int, k: int,
         ids[i] = g.keys[s]
         dists[i] = d
     return ids, dists
+
+
+def oracle_search_filtered_batch(g: HnswGraph, query: np.ndarray, ef: int,
+                                 k: int, allowed):
+    """The GPU filtered kernel's SPEC: usearch's predicate semantics with two
+    bounded-memory deviations — the radius is recomputed once per EXPANSION
+    (usearch tightens it incrementally inside the neighbor loop), and the
+    candidate list is capped at 2*ef. Returned ids always satisfy the filter;
+    id agreement vs usearch's filtered_search measured ~0.90-0.93 at ef=48
+    and 30% selectivity with equivalent recall (see tests)."""
+    q = query.astype(np.float32)
+
+    def dist(slot):
+        return float(_usearch_dist(g.metric, g.vecs[slot], q))
+
+    cur = g.entry_slot
+    cur_d = dist(cur)
+    for level in range(g.max_level, 0, -1):
+        changed = True
+        while changed:
+            changed = False
+            best, bi = cur_d, -1
+            for nb in g.neighbors(cur, level):
+                d = dist(int(nb))
+                if d < best:
+                    best, bi = d, int(nb)
+            if bi >= 0:
+                cur, cur_d, changed = bi, best, True
+
+    efc = 2 * ef
+    lst = [(dist(cur), cur, bool(allowed(int(g.keys[cur]))), False)]
+    visited = {cur}
+    while True:
+        pd = [d for d, s_, p, e in lst if p]
+        pass_cnt = len(pd)
+        radius = pd[ef - 1] if pass_cnt >= ef else float("inf")
+        bi = next((i for i, t in enumerate(lst) if not t[3]), None)
+        if bi is None:
+            break
+        if pass_cnt >= ef and lst[bi][0] > radius:
+            break
+        d0, s0, p0, _ = lst[bi]
+        lst[bi] = (d0, s0, p0, True)
+        new = []
+        for nb in g.neighbors(s0, 0):
+            nb = int(nb)
+            if nb in visited:
+                continue
+            visited.add(nb)
+            d = dist(nb)
+            if pass_cnt >= ef and d >= radius:
+                continue
+            new.append((d, nb, bool(allowed(int(g.keys[nb]))), False))
+        lst = sorted(lst + new, key=lambda t: t[0])[:efc]
+    out = [(d, s_) for d, s_, p, e in lst if p][:k]
+    ids = np.full(k, -1, dtype=np.int64)
+    dists = np.full(k, np.finfo(np.float32).max, dtype=np.float32)
+    for i, (d, s_) in enumerate(out):
+        ids[i] = g.keys[s_]
+        dists[i] = np.float32(d)
+    return ids, dists
