@@ -204,6 +204,13 @@ void moann_brute_force_search(const float* dataset, uint64_t count,
                               int64_t* out_neighbors, float* out_distances,
                               void* errmsg);
 
+/* 1xN pairwise distances — the SQL distance-builtin const-query batch
+ * (metric.PairwiseDistanceLaunch, pkg/sql/plan/function/func_binary.go:9127;
+ * subsumes cgo/cuda/mocl.cu). MO conventions; synchronous. */
+void moann_pairwise_distance(distance_type_t metric, const float* query,
+                             const float* rows, uint64_t n, uint32_t dim,
+                             int device, float* out_dists, void* errmsg);
+
 /* --- perf counters (roofline evidence; DESIGN.md measurement section) ----- */
 
 typedef struct {

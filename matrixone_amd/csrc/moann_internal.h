@@ -90,6 +90,11 @@ bool launch_rank_gemm(int metric, const float* queries, const float* cents,
                       const float* qnorms, const float* cnorms, int nq,
                       int nlist, int dpad, float* out, hipStream_t stream);
 
+/* 1xN pairwise distances (SQL builtin batch / mocl.cu counterpart). */
+void launch_pairwise(int metric, const float* rows, const float* query,
+                     float qnorm, int64_t n, int dim, float* out,
+                     hipStream_t stream);
+
 /* Sum-of-squares per query row (cosine). queries: [nq][dpad]. */
 void launch_qnorms(const float* queries, int nq, int dpad, float* qnorms,
                    hipStream_t stream);

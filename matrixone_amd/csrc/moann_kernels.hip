@@ -622,6 +622,84 @@ void launch_unpack(const float* packed, int dim, int dpad,
                        ngroups, slot_lo, slot_hi, out);
 }
 
+/* --------------------------- pairwise 1xN --------------------------------
+ * The SQL distance-builtin batch (metric.PairwiseDistanceLaunch,
+ * pkg/sql/plan/function/func_binary.go:9127 const-query 1xN) and the legacy
+ * cgo/cuda/mocl.cu l2distance kernels, subsumed: one wave per row over
+ * row-major data, MO distance conventions. */
+template <int METRIC>
+__global__ __launch_bounds__(256) void pairwise_kernel(
+    const float* __restrict__ rows, const float* __restrict__ query,
+    float qnorm /* |q|^2, cos only */, int64_t n, int dim,
+    float* __restrict__ out) {
+    const int64_t r = (int64_t)blockIdx.x * (blockDim.x >> 6) +
+                      (threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (r >= n) return;
+    const float* row = rows + r * (int64_t)dim;
+    float acc = 0.f, nrm = 0.f;
+    for (int e = lane; e < dim; e += 64) {
+        const float x = row[e], qv = query[e];
+        if (METRIC == KM_L2SQ) {
+            const float d = x - qv;
+            acc = fmaf(d, d, acc);
+        } else if (METRIC == KM_L1) {
+            acc += fabsf(x - qv);
+        } else {
+            acc = fmaf(x, qv, acc);
+            if (METRIC == KM_COS) nrm += x * x;
+        }
+    }
+#pragma unroll
+    for (int w = 32; w; w >>= 1) {
+        acc += __shfl_down(acc, w, 64);
+        if (METRIC == KM_COS) nrm += __shfl_down(nrm, w, 64);
+    }
+    if (lane) return;
+    float dist;
+    if (METRIC == KM_IP) dist = -acc;
+    else if (METRIC == KM_COS) {
+        const double denom = sqrt((double)nrm) * sqrt((double)qnorm);
+        if (denom == 0.0) dist = 1.0f;
+        else {
+            double sim = (double)acc / denom;
+            sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+            dist = (float)(1.0 - sim);
+        }
+    } else dist = acc;
+    out[r] = dist;
+}
+
+void launch_pairwise(int metric, const float* rows, const float* query,
+                     float qnorm, int64_t n, int dim, float* out,
+                     hipStream_t stream) {
+    if (!n) return;
+    const int wpb = 4;
+    const int64_t grid = (n + wpb - 1) / wpb;
+    switch (metric) {
+    case KM_L2SQ:
+        hipLaunchKernelGGL((pairwise_kernel<KM_L2SQ>), dim3((uint32_t)grid),
+                           dim3(wpb * 64), 0, stream, rows, query, qnorm, n,
+                           dim, out);
+        break;
+    case KM_IP:
+        hipLaunchKernelGGL((pairwise_kernel<KM_IP>), dim3((uint32_t)grid),
+                           dim3(wpb * 64), 0, stream, rows, query, qnorm, n,
+                           dim, out);
+        break;
+    case KM_COS:
+        hipLaunchKernelGGL((pairwise_kernel<KM_COS>), dim3((uint32_t)grid),
+                           dim3(wpb * 64), 0, stream, rows, query, qnorm, n,
+                           dim, out);
+        break;
+    default:
+        hipLaunchKernelGGL((pairwise_kernel<KM_L1>), dim3((uint32_t)grid),
+                           dim3(wpb * 64), 0, stream, rows, query, qnorm, n,
+                           dim, out);
+        break;
+    }
+}
+
 /* --------------------------- query norms ---------------------------------
  * one wave per query; f32 accumulate (matches distance_func.go cosine's f32
  * norm accumulation within the 1e-5 parity tolerance). */

@@ -1260,6 +1260,41 @@ void moann_ivf_flat_perf_reset(gpu_ivf_flat_c h) {
     IX(h)->perf.c = moann_perf_t {};
 }
 
+/* 1xN pairwise distances — the SQL l2_distance/l2_distance_sq/
+ * inner_product/cosine_distance/l1_distance const-query batch
+ * (metric.PairwiseDistanceLaunch, func_binary.go:9127; subsumes the legacy
+ * cgo/cuda/mocl.cu l2distance kernels). Synchronous; MO conventions;
+ * the l2 metric applies sqrt on output. */
+void moann_pairwise_distance(distance_type_t metric, const float* query,
+                             const float* rows, uint64_t n, uint32_t dim,
+                             int device, float* out_dists, void* errmsg) {
+    try {
+        const int km = metric_kind(metric);
+        const int do_sqrt = metric == DistanceType_L2SqrtExpanded;
+        HIP_CHECK(hipSetDevice(device));
+        DevBuf d_rows, d_q, d_out;
+        d_rows.ensure((size_t)n * dim * 4);
+        d_q.ensure((size_t)dim * 4);
+        d_out.ensure((size_t)n * 4);
+        HIP_CHECK(hipMemcpy(d_rows.ptr, rows, (size_t)n * dim * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(d_q.ptr, query, (size_t)dim * 4,
+                            hipMemcpyHostToDevice));
+        float qn = 0.f;
+        for (uint32_t e = 0; e < dim; ++e) qn += query[e] * query[e];
+        launch_pairwise(km, d_rows.as<float>(), d_q.as<float>(), qn,
+                        (int64_t)n, (int)dim, d_out.as<float>(), nullptr);
+        KCHECK("pairwise");
+        HIP_CHECK(hipMemcpy(out_dists, d_out.ptr, (size_t)n * 4,
+                            hipMemcpyDeviceToHost));
+        if (do_sqrt)
+            for (uint64_t i = 0; i < n; ++i)
+                out_dists[i] = (float)sqrt((double)out_dists[i]);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_pairwise_distance", e.what());
+    }
+}
+
 int moann_device_count(void) {
     int n = 0;
     if (hipGetDeviceCount(&n) != hipSuccess) return 0;

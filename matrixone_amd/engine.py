@@ -131,6 +131,9 @@ def lib() -> ct.CDLL:
     L.moann_ivf_pq_destroy.argtypes = [ct.c_void_p, ct.c_void_p]
     L.moann_ivf_flat_perf.argtypes = [ct.c_void_p, ct.POINTER(PerfT)]
     L.moann_ivf_flat_perf_reset.argtypes = [ct.c_void_p]
+    L.moann_pairwise_distance.argtypes = [
+        ct.c_int, ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32,
+        ct.c_int, ct.c_void_p, ct.c_void_p]
     L.moann_device_count.restype = ct.c_int
     L.moann_version.restype = ct.c_char_p
     _lib = L
@@ -343,6 +346,21 @@ class IvfFlatIndex:
             self.close()
         except Exception:
             pass
+
+
+def pairwise_distance(metric: str, query, rows, device: int = 0):
+    """1xN pairwise distances (the SQL distance-builtin batch —
+    metric.PairwiseDistanceLaunch counterpart, MO conventions)."""
+    query = _as_f32(query).reshape(-1)
+    rows = _as_f32(rows)
+    out = np.empty(rows.shape[0], dtype=np.float32)
+    err = _Err()
+    lib().moann_pairwise_distance(METRICS[metric], query.ctypes.data,
+                                  rows.ctypes.data, rows.shape[0],
+                                  rows.shape[1], device, out.ctypes.data,
+                                  err.ref)
+    err.check("pairwise_distance")
+    return out
 
 
 def brute_force_search(dataset, queries, k: int, metric: str = "l2sq",

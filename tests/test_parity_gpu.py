@@ -291,3 +291,19 @@ def test_dynamic_batching():
     ids2, _ = ix.search(qs, 5, 4)
     np.testing.assert_array_equal(ids2, ref)
     ix.close()
+
+
+def test_pairwise_distance():
+    """1xN pairwise batch (the SQL builtin shape, func_binary.go:9127) vs
+    the oracle distance kernels, all metrics."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(800))
+    for metric, om in [("l2sq", orc.METRIC_L2SQ), ("l2", orc.METRIC_L2),
+                       ("ip", orc.METRIC_IP), ("cos", orc.METRIC_COS),
+                       ("l1", orc.METRIC_L1)]:
+        rows = rng.standard_normal((500, 131), dtype=np.float32)
+        q = rng.standard_normal(131, dtype=np.float32)
+        got = engine.pairwise_distance(metric, q, rows)
+        want = np.array([orc.distance(om, rows[i], q) for i in range(500)])
+        np.testing.assert_allclose(got, want, rtol=2e-5, atol=2e-5,
+                                   err_msg=metric)
