@@ -137,6 +137,14 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c index_c, const char* dir,
 /* ivf_flat_c.h:53 */
 void gpu_ivf_flat_destroy(gpu_ivf_flat_c index_c, void* errmsg);
 
+/* ivf_flat_c.h:93-94: affine scalar quantizer for INT8/UINT8 storage
+ * (quantizer.go Int8Params/Uint8Params formulas; search rescales raw scores
+ * by 1/mul^2 per ivfflat/search.go:1062-1077). Call set before build. */
+void gpu_ivf_flat_set_quantizer(gpu_ivf_flat_c index_c, float min, float max,
+                                void* errmsg);
+void gpu_ivf_flat_get_quantizer(gpu_ivf_flat_c index_c, float* min, float* max,
+                                void* errmsg);
+
 /* --- search (ivf_flat_c.h:110-141) --------------------------------------- */
 
 /* ivf_flat_c.h:114-116: queries_data is HOST memory, f32 row-major

@@ -90,6 +90,16 @@ bool launch_rank_gemm(int metric, const float* queries, const float* cents,
                       const float* qnorms, const float* cnorms, int nq,
                       int nlist, int dpad, float* out, hipStream_t stream);
 
+/* Quantized (int8/uint8) list scan + helpers. */
+void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
+                    const uint8_t* queries_q, const int32_t* qnorms, int dpad,
+                    const ScanJobs& jb, float* dists_out, hipStream_t stream,
+                    const uint32_t* filter_bitset = nullptr);
+void launch_quantize(bool uns, const float* in, int64_t n_elems, float fmul,
+                     float fadd, uint8_t* out, hipStream_t stream);
+void launch_qnorms_i8(bool uns, const uint8_t* q, int nq, int dpad,
+                      int32_t* out, hipStream_t stream);
+
 /* 1xN pairwise distances (SQL builtin batch / mocl.cu counterpart). */
 void launch_pairwise(int metric, const float* rows, const float* query,
                      float qnorm, int64_t n, int dim, float* out,

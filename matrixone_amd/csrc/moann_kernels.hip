@@ -622,6 +622,266 @@ void launch_unpack(const float* packed, int dim, int dpad,
                        ngroups, slot_lo, slot_hi, out);
 }
 
+/* -------------------------- quantized (int8/uint8) scan ------------------
+ * Narrow-storage variant of the list scan (reference: int8/uint8 entries
+ * quantized by q(x)=round(x*mul+add), distance computed IN THE QUANTIZED
+ * DOMAIN with exact integer accumulation — distance_func_narrow.go:330-450;
+ * the /mul^2 rescale happens at gather, ivfflat/search.go:1062-1077).
+ * Data layout: bytes [g][dpad/4][64][4] (same interleave as f32, byte
+ * elements — one coalesced 256 B read per wave step). Query tile: quantized
+ * bytes in LDS. int32 accumulators are exact for dpad <= 33024 (l2sq term
+ * <= 65025/elem). */
+template <int METRIC, bool UNSIGNED>
+__global__ __launch_bounds__(256) void scan_i8_kernel(
+    const uint8_t* __restrict__ packed, const uint8_t* __restrict__ queries_q,
+    const int32_t* __restrict__ qnorms_i /* [nq] sum sq, cos only */,
+    int dpad,
+    const int64_t* __restrict__ j_databaseg,
+    const int32_t* __restrict__ j_gstart, const int32_t* __restrict__ j_gcount,
+    const int32_t* __restrict__ j_rows, const int32_t* __restrict__ j_nq,
+    const int32_t* __restrict__ j_qbase,
+    const int32_t* __restrict__ qslot_query,
+    const int64_t* __restrict__ qslot_outbase,
+    const int64_t* __restrict__ j_slot_base,
+    const uint32_t* __restrict__ filter_bitset,
+    float* __restrict__ dists_out) {
+    constexpr int QT = 8;
+    extern __shared__ float ldsf[];
+    uint8_t* ldsq = (uint8_t*)ldsf;            /* [QT][dpad] bytes */
+    int32_t* ldsn = (int32_t*)(ldsq + QT * dpad);
+
+    int j;
+    {
+        const int nwg = gridDim.x, bid = blockIdx.x;
+        const int q8 = nwg >> 3, r8 = nwg & 7;
+        const int xcd = bid & 7, idx = bid >> 3;
+        j = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+        if (j >= nwg) j = bid;
+    }
+    const int nq = j_nq[j];
+    const int qbase = j_qbase[j];
+    for (int t = 0; t < QT; ++t) {
+        if (t < nq) {
+            const int q = qslot_query[qbase + t];
+            const uint8_t* src = queries_q + (int64_t)q * dpad;
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = src[e];
+            if (METRIC == KM_COS && threadIdx.x == 0) ldsn[t] = qnorms_i[q];
+        } else {
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = 0;
+            if (METRIC == KM_COS && threadIdx.x == 0) ldsn[t] = 0;
+        }
+    }
+    __syncthreads();
+
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int gstart = j_gstart[j], gcount = j_gcount[j];
+    const int rows = j_rows[j];
+    const int64_t baseg = j_databaseg[j];
+    const int d4 = dpad >> 2;
+
+    const auto sx = [](uint8_t b) -> int {
+        return UNSIGNED ? (int)b : (int)(int8_t)b;
+    };
+
+    for (int gp = gstart + 2 * wave; gp < gstart + gcount; gp += 8) {
+        const int g0 = gp;
+        const bool has1 = (gp + 1) < (gstart + gcount);
+        const int g1 = has1 ? gp + 1 : gp;
+        const uchar4* __restrict__ d0 =
+            (const uchar4*)(packed + (baseg + g0) * (int64_t)64 * dpad) + lane;
+        const uchar4* __restrict__ d1 =
+            (const uchar4*)(packed + (baseg + g1) * (int64_t)64 * dpad) + lane;
+        int32_t acc0[QT], acc1[QT];
+#pragma unroll
+        for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0;
+        int32_t rn0 = 0, rn1 = 0;
+#pragma unroll 2
+        for (int q4 = 0; q4 < d4; ++q4) {
+            const uchar4 x0 = d0[q4 * 64];
+            const uchar4 x1 = d1[q4 * 64];
+            if (METRIC == KM_COS) {
+                rn0 += sx(x0.x) * sx(x0.x) + sx(x0.y) * sx(x0.y) +
+                       sx(x0.z) * sx(x0.z) + sx(x0.w) * sx(x0.w);
+                rn1 += sx(x1.x) * sx(x1.x) + sx(x1.y) * sx(x1.y) +
+                       sx(x1.z) * sx(x1.z) + sx(x1.w) * sx(x1.w);
+            }
+#pragma unroll
+            for (int t = 0; t < QT; ++t) {
+                const uchar4 qv = ((const uchar4*)(ldsq + t * dpad))[q4];
+                if (METRIC == KM_L2SQ) {
+                    int e;
+                    e = sx(x0.x) - sx(qv.x); acc0[t] += e * e;
+                    e = sx(x0.y) - sx(qv.y); acc0[t] += e * e;
+                    e = sx(x0.z) - sx(qv.z); acc0[t] += e * e;
+                    e = sx(x0.w) - sx(qv.w); acc0[t] += e * e;
+                    e = sx(x1.x) - sx(qv.x); acc1[t] += e * e;
+                    e = sx(x1.y) - sx(qv.y); acc1[t] += e * e;
+                    e = sx(x1.z) - sx(qv.z); acc1[t] += e * e;
+                    e = sx(x1.w) - sx(qv.w); acc1[t] += e * e;
+                } else if (METRIC == KM_IP || METRIC == KM_COS) {
+                    acc0[t] += sx(x0.x) * sx(qv.x) + sx(x0.y) * sx(qv.y) +
+                               sx(x0.z) * sx(qv.z) + sx(x0.w) * sx(qv.w);
+                    acc1[t] += sx(x1.x) * sx(qv.x) + sx(x1.y) * sx(qv.y) +
+                               sx(x1.z) * sx(qv.z) + sx(x1.w) * sx(qv.w);
+                } else { /* KM_L1 */
+                    acc0[t] += abs(sx(x0.x) - sx(qv.x)) +
+                               abs(sx(x0.y) - sx(qv.y)) +
+                               abs(sx(x0.z) - sx(qv.z)) +
+                               abs(sx(x0.w) - sx(qv.w));
+                    acc1[t] += abs(sx(x1.x) - sx(qv.x)) +
+                               abs(sx(x1.y) - sx(qv.y)) +
+                               abs(sx(x1.z) - sx(qv.z)) +
+                               abs(sx(x1.w) - sx(qv.w));
+                }
+            }
+        }
+        const int row0 = g0 * 64 + lane;
+        const int row1 = g1 * 64 + lane;
+        bool pass0 = true, pass1 = true;
+        if (filter_bitset) {
+            const int64_t sb = j_slot_base[j];
+            if (row0 < rows) {
+                const int64_t gs = sb + row0;
+                pass0 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+            if (has1 && row1 < rows) {
+                const int64_t gs = sb + row1;
+                pass1 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+        }
+#pragma unroll
+        for (int t = 0; t < QT; ++t) {
+            if (t >= nq) break;
+            const int64_t ob = qslot_outbase[qbase + t];
+            if (row0 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = (float)(-(int64_t)acc0[t]);
+                else if (METRIC == KM_COS) {
+                    /* distance_func_narrow.go:444-449: int sums -> double */
+                    const double denom =
+                        sqrt((double)rn0) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc0[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else dist = (float)acc0[t];
+                dists_out[ob + row0] = pass0 ? dist : FLT_MAX;
+            }
+            if (has1 && row1 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = (float)(-(int64_t)acc1[t]);
+                else if (METRIC == KM_COS) {
+                    const double denom =
+                        sqrt((double)rn1) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc1[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else dist = (float)acc1[t];
+                dists_out[ob + row1] = pass1 ? dist : FLT_MAX;
+            }
+        }
+    }
+}
+
+template <int METRIC>
+static void launch_scan_i8_m(bool uns, const uint8_t* packed,
+                             const uint8_t* queries_q, const int32_t* qnorms,
+                             int dpad, const ScanJobs& jb, float* dists_out,
+                             hipStream_t stream,
+                             const uint32_t* filter_bitset) {
+    const dim3 grid(jb.njobs), block(256);
+    const size_t shmem = 8 * dpad + 8 * 4 + 16;
+#define I8ARGS packed, queries_q, qnorms, dpad, jb.databaseg, jb.gstart,     jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase,     jb.slot_base, filter_bitset, dists_out
+    if (uns)
+        hipLaunchKernelGGL((scan_i8_kernel<METRIC, true>), grid, block, shmem,
+                           stream, I8ARGS);
+    else
+        hipLaunchKernelGGL((scan_i8_kernel<METRIC, false>), grid, block,
+                           shmem, stream, I8ARGS);
+#undef I8ARGS
+}
+
+void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
+                    const uint8_t* queries_q, const int32_t* qnorms, int dpad,
+                    const ScanJobs& jb, float* dists_out, hipStream_t stream,
+                    const uint32_t* filter_bitset) {
+    if (!jb.njobs) return;
+    switch (metric) {
+    case KM_L2SQ: launch_scan_i8_m<KM_L2SQ>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_IP:   launch_scan_i8_m<KM_IP>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_COS:  launch_scan_i8_m<KM_COS>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    default:      launch_scan_i8_m<KM_L1>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    }
+}
+
+/* quantize f32 rows -> int8/uint8 with the reference's exact semantics:
+ * two separate f32 roundings (product, then sum — quantizer.go:165-176),
+ * round half away from zero, clamp, NaN -> 0 (float16.go:227-239). */
+template <bool UNSIGNED>
+__global__ void quantize_kernel(const float* __restrict__ in, int64_t n_elems,
+                                float fmul, float fadd,
+                                uint8_t* __restrict__ out) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n_elems) return;
+    const float x = in[i];
+    float m = x * fmul;
+    asm volatile("" : "+v"(m)); /* block FMA fusion of the two f32 ops */
+    const float sum = m + fadd;
+    float dist;
+    if (sum != sum) dist = 0.f;
+    else dist = (float)round((double)sum);
+    const float lo = UNSIGNED ? 0.f : -128.f, hi = UNSIGNED ? 255.f : 127.f;
+    dist = dist < lo ? lo : (dist > hi ? hi : dist);
+    out[i] = UNSIGNED ? (uint8_t)dist : (uint8_t)(int8_t)dist;
+}
+
+void launch_quantize(bool uns, const float* in, int64_t n_elems, float fmul,
+                     float fadd, uint8_t* out, hipStream_t stream) {
+    if (!n_elems) return;
+    const int block = 256;
+    const int64_t grid = (n_elems + block - 1) / block;
+    if (uns)
+        hipLaunchKernelGGL((quantize_kernel<true>), dim3((uint32_t)grid),
+                           dim3(block), 0, stream, in, n_elems, fmul, fadd, out);
+    else
+        hipLaunchKernelGGL((quantize_kernel<false>), dim3((uint32_t)grid),
+                           dim3(block), 0, stream, in, n_elems, fmul, fadd, out);
+}
+
+/* int32 sum-of-squares per quantized query (cos, narrow path) */
+template <bool UNSIGNED>
+__global__ void qnorm_i8_kernel(const uint8_t* __restrict__ q, int nq,
+                                int dpad, int32_t* __restrict__ out) {
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= nq) return;
+    const uint8_t* row = q + (int64_t)i * dpad;
+    int32_t s = 0;
+    for (int e = 0; e < dpad; ++e) {
+        const int v = UNSIGNED ? (int)row[e] : (int)(int8_t)row[e];
+        s += v * v;
+    }
+    out[i] = s;
+}
+
+void launch_qnorms_i8(bool uns, const uint8_t* q, int nq, int dpad,
+                      int32_t* out, hipStream_t stream) {
+    if (!nq) return;
+    const int grid = (nq + 255) / 256;
+    if (uns)
+        hipLaunchKernelGGL((qnorm_i8_kernel<true>), dim3(grid), dim3(256), 0,
+                           stream, q, nq, dpad, out);
+    else
+        hipLaunchKernelGGL((qnorm_i8_kernel<false>), dim3(grid), dim3(256), 0,
+                           stream, q, nq, dpad, out);
+}
+
 /* --------------------------- pairwise 1xN --------------------------------
  * The SQL distance-builtin batch (metric.PairwiseDistanceLaunch,
  * pkg/sql/plan/function/func_binary.go:9127 const-query 1xN) and the legacy

@@ -14,6 +14,7 @@
 
 #include <algorithm>
 #include <cfloat>
+#include <cmath>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -117,7 +118,10 @@ struct IvfIndex {
     distance_type_t metric = DistanceType_L2Expanded;
     int kmetric = KM_L2SQ;
     uint64_t cap = 0, count = 0;
-    double quant_mul = 1.0; /* identity for f32 (quantizer round 2) */
+    quantization_t qtype = Quantization_F32;
+    double quant_mul = 1.0, quant_add = 0.0; /* identity until trained */
+    double quant_min = 0.0, quant_max = 0.0;
+    uint8_t* d_packed_q = nullptr;  /* quantized interleaved entries */
     bool built = false, started = false;
 
     hipStream_t stream = nullptr;
@@ -148,7 +152,7 @@ struct IvfIndex {
     DevBuf w_queries, w_qnorms, w_cdists, w_csel_slots, w_csel_dists;
     DevBuf w_jobs, w_qslots, w_cand, w_sel_slots, w_sel_dists;
     DevBuf w_probe_lists, w_probe_offs, w_qoffs, w_out_ids, w_out_dists;
-    DevBuf w_filter;
+    DevBuf w_filter, w_queries_q, w_qnorms_i;
     std::vector<int64_t> h_id_by_slot;  /* kept for the membership filter */
     std::unordered_map<int64_t, int64_t> id_to_slot; /* lazy */
 
@@ -187,7 +191,7 @@ struct IvfIndex {
         }
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_staging, (void*)d_packed, (void*)d_cent_packed,
-                       (void*)d_cent_rows, (void*)d_cnorms,
+                       (void*)d_cent_rows, (void*)d_cnorms, (void*)d_packed_q,
                        (void*)d_id_by_slot, (void*)d_list_slot_base})
             if (p) (void)hipFree(p);
         if (stream) (void)hipStreamDestroy(stream);
@@ -482,6 +486,8 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const int nqslots = (int)qslot_query.size();
 
     /* 4. list scan into the ragged candidate buffer */
+    const bool narrow = ix->qtype != Quantization_F32;
+    const bool uns = ix->qtype == Quantization_UINT8;
     const uint32_t* d_filter = nullptr;
     if (filter_words) {
         const size_t fw = (size_t)((ix->count + 31) / 32);
@@ -517,8 +523,29 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         jb.slot_base = (int64_t*)up(jslot_base.data(), njobs * b64);
         jb.njobs = njobs;
         HIP_CHECK(hipEventRecord(ix->perf.ev[2], s));
-        launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
-                    ix->w_cand.as<float>(), s, d_filter);
+        if (narrow) {
+            /* quantize the padded f32 queries with the entry map (the
+             * search_quantize contract: query narrowed the same way as the
+             * entries — quantizer.go ApplyInt8/entry SQL) */
+            ix->w_queries_q.ensure((size_t)nq * dpad);
+            launch_quantize(uns, d_q, (int64_t)nq * dpad,
+                            (float)ix->quant_mul, (float)ix->quant_add,
+                            ix->w_queries_q.as<uint8_t>(), s);
+            KCHECK("query-quantize");
+            const int32_t* d_qni = nullptr;
+            if (ix->kmetric == KM_COS) {
+                ix->w_qnorms_i.ensure(nq * 4);
+                launch_qnorms_i8(uns, ix->w_queries_q.as<uint8_t>(), (int)nq,
+                                 dpad, ix->w_qnorms_i.as<int32_t>(), s);
+                d_qni = ix->w_qnorms_i.as<int32_t>();
+            }
+            launch_scan_i8(ix->kmetric, uns, ix->d_packed_q,
+                           ix->w_queries_q.as<uint8_t>(), d_qni, dpad, jb,
+                           ix->w_cand.as<float>(), s, d_filter);
+        } else {
+            launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
+                        ix->w_cand.as<float>(), s, d_filter);
+        }
         KCHECK("list-scan");
         HIP_CHECK(hipEventRecord(ix->perf.ev[3], s));
     }
@@ -650,8 +677,11 @@ gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
                                       quantization_t btype, quantization_t qtype,
                                       const int64_t* /*ids*/, void* errmsg) {
     try {
-        if (btype != Quantization_F32 || qtype != Quantization_F32)
-            throw std::runtime_error("round 1 wires F32/F32 only");
+        if (btype != Quantization_F32 ||
+            (qtype != Quantization_F32 && qtype != Quantization_INT8 &&
+             qtype != Quantization_UINT8))
+            throw std::runtime_error(
+                "wired combos: F32 base with F32/INT8/UINT8 storage");
         auto ix = std::make_unique<IvfIndex>();
         ix->device = (devices && device_count > 0) ? devices[0] : 0;
         HIP_CHECK(hipSetDevice(ix->device));
@@ -661,6 +691,7 @@ gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
         ix->metric = metric;
         ix->kmetric = metric_kind(metric);
         ix->nlist = build_params.n_lists ? build_params.n_lists : 1024;
+        ix->qtype = qtype;
         ix->cap = total_count;
         ix->h_ids.reserve(total_count);
         if (total_count)
@@ -761,6 +792,21 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
         /* pack entries */
         pack_lists(ix, ix->d_staging, ix->count, slot_rows, loffsets,
                    &ix->d_packed, &ix->list_gbase, &ix->total_groups);
+        if (ix->qtype != Quantization_F32) {
+            /* quantize the PACKED f32 image elementwise (same layout: the
+             * byte interleave [g][dpad/4][64][4] equals the float4
+             * interleave with byte elements), then drop the f32 copy —
+             * narrow storage is the point (search.go:1062 quant domain). */
+            const int64_t n_elems = ix->total_groups * 64 * (int64_t)ix->dpad;
+            HIP_CHECK(hipMalloc(&ix->d_packed_q, (size_t)n_elems));
+            launch_quantize(ix->qtype == Quantization_UINT8, ix->d_packed,
+                            n_elems, (float)ix->quant_mul,
+                            (float)ix->quant_add, ix->d_packed_q, ix->stream);
+            HIP_CHECK(hipStreamSynchronize(ix->stream));
+            KCHECK("quantize");
+            HIP_CHECK(hipFree(ix->d_packed));
+            ix->d_packed = nullptr;
+        }
         /* group -> slot metadata kept for save_dir's unpack */
         ix->group_slotbase.resize(ix->total_groups);
         ix->group_valid.resize(ix->total_groups);
@@ -1099,6 +1145,9 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
         auto ix = IX(h);
         std::lock_guard<std::mutex> lk(ix->mu);
         if (!ix->built) throw std::runtime_error("index not built");
+        if (ix->qtype != Quantization_F32)
+            throw std::runtime_error(
+                "save_dir for narrow storage not yet wired");
         HIP_CHECK(hipSetDevice(ix->device));
         std::string d(dir);
         auto wfile = [&](const std::string& name, const void* p, size_t n) {
@@ -1237,6 +1286,44 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
         }
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_load_dir", e.what());
+    }
+}
+
+/* ivf_flat_c.h:93-94 set/get_quantizer: trains the affine scalar quantizer
+ * from [min,max] with the reference formulas (quantizer.go:90-100 Int8Params
+ * mul=255/rng, add=-min*mul-128; :232-240 Uint8Params without the -128
+ * shift). Call before build; search rescales raw scores by 1/mul^2
+ * (search.go:1062-1077). */
+void gpu_ivf_flat_set_quantizer(gpu_ivf_flat_c h, float min, float max,
+                                void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (ix->qtype == Quantization_F32)
+            throw std::runtime_error("index storage is f32 (no quantizer)");
+        const double rng = (double)max - (double)min;
+        ix->quant_min = min;
+        ix->quant_max = max;
+        if (!(rng > 0) || std::isinf(rng)) {
+            ix->quant_mul = 1.0;
+            ix->quant_add = 0.0;
+        } else {
+            ix->quant_mul = 255.0 / rng;
+            ix->quant_add = -(double)min * ix->quant_mul +
+                            (ix->qtype == Quantization_INT8 ? -128.0 : 0.0);
+        }
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_set_quantizer", e.what());
+    }
+}
+
+void gpu_ivf_flat_get_quantizer(gpu_ivf_flat_c h, float* min, float* max,
+                                void* errmsg) {
+    try {
+        auto ix = IX(h);
+        *min = (float)ix->quant_min;
+        *max = (float)ix->quant_max;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_get_quantizer", e.what());
     }
 }
 

@@ -97,6 +97,10 @@ def lib() -> ct.CDLL:
     L.gpu_ivf_flat_get_n_list.argtypes = [ct.c_void_p]
     L.gpu_ivf_flat_set_batch_window.argtypes = [ct.c_void_p, ct.c_int64,
                                                 ct.c_void_p]
+    L.gpu_ivf_flat_set_quantizer.argtypes = [ct.c_void_p, ct.c_float,
+                                             ct.c_float, ct.c_void_p]
+    L.gpu_ivf_flat_get_quantizer.argtypes = [ct.c_void_p, ct.c_void_p,
+                                             ct.c_void_p, ct.c_void_p]
     L.gpu_ivf_flat_save_dir.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_void_p]
     L.gpu_ivf_flat_load_dir.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_int,
                                         ct.c_void_p]
@@ -177,8 +181,10 @@ class IvfFlatIndex:
     SearchAsync / SearchWait), with the MatrixOne-CPU-path ingestion shape
     (explicit centroids + entry assignments from the hidden tables)."""
 
+    QTYPES = {"f32": 0, "int8": 2, "uint8": 3}  # quantization_t
+
     def __init__(self, dim: int, nlist: int, metric: str = "l2sq",
-                 capacity: int = 0, device: int = 0):
+                 capacity: int = 0, device: int = 0, qtype: str = "f32"):
         if metric not in METRICS:
             raise ValueError(f"metric {metric!r} not in {sorted(METRICS)}")
         self.dim, self.nlist, self.metric = dim, nlist, metric
@@ -187,8 +193,8 @@ class IvfFlatIndex:
         bp = BuildParams(nlist, True, 0.5)
         dev = (ct.c_int * 1)(device)
         self._h = lib().gpu_ivf_flat_new_empty(
-            capacity, dim, METRICS[metric], bp, dev, 1, 0, 0, 0, 0, None,
-            err.ref)
+            capacity, dim, METRICS[metric], bp, dev, 1, 0, 0, 0,
+            self.QTYPES[qtype], None, err.ref)
         err.check("new_empty")
         if not self._h:
             raise MoannError("gpu_ivf_flat_new_empty returned NULL")
@@ -248,6 +254,11 @@ class IvfFlatIndex:
                                         SearchParams(nprobe), err.ref)
         err.check("search")
         return self._read_result(res, nq, k, "search")
+
+    def set_quantizer(self, vmin: float, vmax: float):
+        err = _Err()
+        lib().gpu_ivf_flat_set_quantizer(self._h, vmin, vmax, err.ref)
+        err.check("set_quantizer")
 
     def save_dir(self, path: str):
         err = _Err()

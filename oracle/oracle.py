@@ -208,3 +208,108 @@ def usearch_exact_search(base: np.ndarray, queries: np.ndarray, k: int,
     if err.value:
         raise RuntimeError(err.value.decode())
     return keys, dists
+
+
+# ---------------- int8/uint8 scalar quantization (quantizer.go) -------------
+
+def int8_params(vmin, vmax):
+    """quantizer.go:90-100 Int8Params: mul=255/rng, add=-min*mul-128."""
+    rng = vmax - vmin
+    if not (rng > 0) or np.isinf(rng):
+        return 1.0, 0.0
+    mul = 255.0 / rng
+    return mul, -vmin * mul + (-128.0)
+
+
+def uint8_params(vmin, vmax):
+    """quantizer.go:232-240 Uint8Params: add folds only the -min offset."""
+    rng = vmax - vmin
+    if not (rng > 0) or np.isinf(rng):
+        return 1.0, 0.0
+    mul = 255.0 / rng
+    return mul, -vmin * mul
+
+
+def _round_half_away(x):
+    """math.Round semantics (ties away from zero) in float64."""
+    return np.copysign(np.floor(np.abs(x) + 0.5), x)
+
+
+def quantize(vecs_f32, mul, add, unsigned=False):
+    """ApplyInt8/ApplyUint8 (quantizer.go:165-176): TWO separate float32
+    roundings (product then sum — the entry SQL evaluates them as two f32
+    ops), then round-half-away-from-zero + clamp (Float32ToInt8/Uint8,
+    container/types/float16.go:227-239). NaN -> 0."""
+    v = np.asarray(vecs_f32, dtype=np.float32)
+    if mul == 1.0 and add == 0.0:
+        s = v
+    else:
+        m = (v * np.float32(mul)).astype(np.float32)
+        s = (m + np.float32(add)).astype(np.float32)
+    r = _round_half_away(s.astype(np.float64))
+    r = np.where(np.isnan(s), 0.0, r)
+    if unsigned:
+        return np.clip(r, 0, 255).astype(np.uint8)
+    return np.clip(r, -128, 127).astype(np.int8)
+
+
+def _int_metric_dists(metric, rows_i, q_i):
+    """Integer-domain distances (distance_func_narrow.go:330-450): exact
+    int64 accumulation, MO conventions."""
+    a = rows_i.astype(np.int64)
+    b = q_i.astype(np.int64)
+    if metric in (METRIC_L2, METRIC_L2SQ):
+        d = a - b
+        return (d * d).sum(1).astype(np.float64)
+    if metric == METRIC_IP:
+        return (-(a * b).sum(1)).astype(np.float64)
+    if metric == METRIC_L1:
+        return np.abs(a - b).sum(1).astype(np.float64)
+    dot = (a * b).sum(1).astype(np.float64)
+    na2 = (a * a).sum(1).astype(np.float64)
+    nb2 = float((b * b).sum())
+    denom = np.sqrt(na2) * np.sqrt(nb2)
+    sim = np.where(denom > 0, dot / np.maximum(denom, 1e-300), 0.0)
+    sim = np.clip(sim, -1.0, 1.0)
+    return np.where(denom > 0, 1.0 - sim, 1.0)
+
+
+def ivf_search_quantized(idx: "IvfIndex", metric, queries, probe, k,
+                         mul, add, unsigned=False, orig_l2=False):
+    """Quantized-entry IVF probe: centroid rank in f32 (LoadCentroids keeps
+    f32 centroids), entries AND query quantized by the same affine map,
+    integer-domain distance, then scoreFromQuantized (raw/mul^2 for every
+    metric — ivfflat/search.go:1062-1077) + the sqrt transform."""
+    queries = np.ascontiguousarray(queries, dtype=np.float32)
+    nq = queries.shape[0]
+    # f32 centroid ranking via the C oracle (identical to the f32 path)
+    _, _, probe_ids = idx.search(metric, queries, probe, 1, want_probe=True)
+    codes = quantize(idx.vecs, mul, add, unsigned)
+    out_ids = np.full((nq, k), -1, dtype=np.int64)
+    out_d = np.full((nq, k), np.float32(np.finfo(np.float32).max))
+    inv_mul2 = 1.0 / (mul * mul) if mul not in (0.0, 1.0) else 1.0
+    for qi in range(nq):
+        q_i = quantize(queries[qi], mul, add, unsigned)
+        cand_rows, cand_d = [], []
+        for l in probe_ids[qi]:
+            if l < 0:
+                continue
+            lo, hi = idx.list_offsets[l], idx.list_offsets[l + 1]
+            rows = idx.slot_rows[lo:hi]
+            if len(rows) == 0:
+                continue
+            cand_rows.append(rows)
+            cand_d.append(_int_metric_dists(metric, codes[rows], q_i))
+        if not cand_rows:
+            continue
+        rows = np.concatenate(cand_rows)
+        d = np.concatenate(cand_d)
+        order = np.lexsort((rows, d))[:k]
+        sel = rows[order]
+        raw = d[order] * inv_mul2
+        if orig_l2 and metric in (METRIC_L2, METRIC_L2SQ):
+            raw = np.sqrt(raw)
+        out_ids[qi, :len(order)] = (idx.ids[sel] if idx.ids is not None
+                                    else sel)
+        out_d[qi, :len(order)] = raw.astype(np.float32)
+    return out_ids, out_d

@@ -307,3 +307,42 @@ def test_pairwise_distance():
         want = np.array([orc.distance(om, rows[i], q) for i in range(500)])
         np.testing.assert_allclose(got, want, rtol=2e-5, atol=2e-5,
                                    err_msg=metric)
+
+
+@pytest.mark.parametrize("qtype,metric", [
+    ("int8", "l2"), ("int8", "ip"), ("uint8", "l2sq"), ("int8", "cos"),
+    ("int8", "l1")])
+def test_quantized_storage(qtype, metric):
+    """INT8/UINT8 storage (the reference's wired btype/qtype combos): entries
+    and query quantized by the same affine map, EXACT integer distances in
+    the quantized domain, raw/mul^2 rescale + sqrt transform — vs the oracle
+    quantized restatement."""
+    from matrixone_amd import engine
+    mmap = {"l2sq": orc.METRIC_L2SQ, "l2": orc.METRIC_L2,
+            "ip": orc.METRIC_IP, "cos": orc.METRIC_COS, "l1": orc.METRIC_L1}
+    rng = np.random.Generator(np.random.PCG64(900))
+    n, d, nlist, nq, probe, k = 12000, 96, 32, 16, 6, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    vmin, vmax = np.percentile(vecs, 0.1), np.percentile(vecs, 99.9)
+    uns = qtype == "uint8"
+    mul, add = (orc.uint8_params(vmin, vmax) if uns
+                else orc.int8_params(vmin, vmax))
+
+    idx = orc.IvfIndex(cents, vecs, assign)
+    ref_ids, ref_d = orc.ivf_search_quantized(
+        idx, mmap[metric], queries, probe, k, mul, add, unsigned=uns,
+        orig_l2=(metric == "l2"))
+
+    ix = engine.IvfFlatIndex(d, nlist, metric=metric, capacity=n, qtype=qtype)
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.set_quantizer(vmin, vmax)
+    ix.build()
+    ids, dists = ix.search(queries, k, probe)
+    _assert_parity(ids, dists, ref_ids, ref_d, ctx=f"{qtype}/{metric}")
+    ix.close()
