@@ -180,6 +180,10 @@ def main():
     ap.add_argument("--recall-queries", type=int, default=256)
     ap.add_argument("--cpu-baseline-queries", type=int, default=48)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--qtype", default="f32", choices=["f32", "int8", "uint8"],
+                    help="storage type; the BASELINE metric is f32 — narrow "
+                         "runs are secondary datapoints and report their own "
+                         "dtype")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -214,7 +218,8 @@ def main():
     need_host = (rank == 0 and world == 1 and not args.skip_cpu_baseline)
     host_vecs = data.cpu().numpy() if need_host else None
     ix = engine.IvfFlatIndex(args.dim, args.nlist, metric="l2sq",
-                             capacity=args.rows, device=local_rank)
+                             capacity=args.rows, device=local_rank,
+                             qtype=args.qtype)
     CH = 1 << 21
     for c0 in range(0, args.rows, CH):
         if host_vecs is not None:
@@ -223,6 +228,10 @@ def main():
             ix.add(data[c0:c0 + CH].cpu().numpy())
     ix.set_centroids(cent_np)
     ix.set_assignments(assign_np)
+    if args.qtype != "f32":
+        lo = float(torch.quantile(data[:65536].reshape(-1).float(), 0.001))
+        hi = float(torch.quantile(data[:65536].reshape(-1).float(), 0.999))
+        ix.set_quantizer(lo, hi)
     ix.build()
     log(f"[bench] build done in {time.perf_counter() - t_build0:.1f}s")
 
@@ -293,7 +302,9 @@ def main():
     perf = ix.perf()
 
     # ---- roofline (list-scan kernel, HIP events on the index stream) ------
-    scan_gbs = (perf["scan_bytes"] / 1e9) / max(perf["scan_ms"] / 1e3, 1e-9)
+    elem = 1 if args.qtype != "f32" else 4
+    scan_gbs = (perf["scan_bytes"] / 4 * elem / 1e9) / max(
+        perf["scan_ms"] / 1e3, 1e-9)
     roofline = {
         "bound": "hbm",
         "achieved": round(scan_gbs, 1),
@@ -327,7 +338,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "f32",
+            "dtype": args.qtype,
             "data": "synthetic",
             "config": {
                 "workload": f"ivfflat-l2-{args.rows}x{args.dim}-"

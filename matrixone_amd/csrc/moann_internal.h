@@ -95,8 +95,13 @@ void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
                     const uint8_t* queries_q, const int32_t* qnorms, int dpad,
                     const ScanJobs& jb, float* dists_out, hipStream_t stream,
                     const uint32_t* filter_bitset = nullptr);
-void launch_quantize(bool uns, const float* in, int64_t n_elems, float fmul,
-                     float fadd, uint8_t* out, hipStream_t stream);
+void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
+                          int in_stride, int dim, int dpad, float fmul,
+                          float fadd, uint8_t* out, hipStream_t stream);
+void launch_pq_pack(const uint8_t* codes_rowmajor, int nsub,
+                    const int64_t* group_rowbase, const int32_t* group_valid,
+                    const int64_t* slot_rows, int64_t ngroups, uint8_t* packed,
+                    hipStream_t stream);
 void launch_qnorms_i8(bool uns, const uint8_t* q, int nq, int dpad,
                       int32_t* out, hipStream_t stream);
 

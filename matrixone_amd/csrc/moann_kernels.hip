@@ -823,36 +823,50 @@ void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
 
 /* quantize f32 rows -> int8/uint8 with the reference's exact semantics:
  * two separate f32 roundings (product, then sum — quantizer.go:165-176),
- * round half away from zero, clamp, NaN -> 0 (float16.go:227-239). */
+ * round half away from zero, clamp, NaN -> 0 (float16.go:227-239).
+ * Pad columns (dim <= c < dpad) emit 0 so padded dims stay identity for
+ * every metric (q(0) is generally nonzero — quantizing pads would poison
+ * IP/cosine). in: [rows][in_stride] f32; out: [rows][dpad] bytes. */
 template <bool UNSIGNED>
-__global__ void quantize_kernel(const float* __restrict__ in, int64_t n_elems,
-                                float fmul, float fadd,
-                                uint8_t* __restrict__ out) {
+__global__ void quantize_rows_kernel(const float* __restrict__ in,
+                                     int64_t nrows, int in_stride, int dim,
+                                     int dpad, float fmul, float fadd,
+                                     uint8_t* __restrict__ out) {
     const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n_elems) return;
-    const float x = in[i];
+    if (i >= nrows * (int64_t)dpad) return;
+    const int64_t r = i / dpad;
+    const int c = (int)(i % dpad);
+    if (c >= dim) {
+        out[i] = 0;
+        return;
+    }
+    const float x = in[r * (int64_t)in_stride + c];
     float m = x * fmul;
     asm volatile("" : "+v"(m)); /* block FMA fusion of the two f32 ops */
     const float sum = m + fadd;
-    float dist;
-    if (sum != sum) dist = 0.f;
-    else dist = (float)round((double)sum);
+    float v;
+    if (sum != sum) v = 0.f;
+    else v = (float)round((double)sum);
     const float lo = UNSIGNED ? 0.f : -128.f, hi = UNSIGNED ? 255.f : 127.f;
-    dist = dist < lo ? lo : (dist > hi ? hi : dist);
-    out[i] = UNSIGNED ? (uint8_t)dist : (uint8_t)(int8_t)dist;
+    v = v < lo ? lo : (v > hi ? hi : v);
+    out[i] = UNSIGNED ? (uint8_t)v : (uint8_t)(int8_t)v;
 }
 
-void launch_quantize(bool uns, const float* in, int64_t n_elems, float fmul,
-                     float fadd, uint8_t* out, hipStream_t stream) {
-    if (!n_elems) return;
+void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
+                          int in_stride, int dim, int dpad, float fmul,
+                          float fadd, uint8_t* out, hipStream_t stream) {
+    const int64_t total = nrows * (int64_t)dpad;
+    if (!total) return;
     const int block = 256;
-    const int64_t grid = (n_elems + block - 1) / block;
+    const int64_t grid = (total + block - 1) / block;
     if (uns)
-        hipLaunchKernelGGL((quantize_kernel<true>), dim3((uint32_t)grid),
-                           dim3(block), 0, stream, in, n_elems, fmul, fadd, out);
+        hipLaunchKernelGGL((quantize_rows_kernel<true>), dim3((uint32_t)grid),
+                           dim3(block), 0, stream, in, nrows, in_stride, dim,
+                           dpad, fmul, fadd, out);
     else
-        hipLaunchKernelGGL((quantize_kernel<false>), dim3((uint32_t)grid),
-                           dim3(block), 0, stream, in, n_elems, fmul, fadd, out);
+        hipLaunchKernelGGL((quantize_rows_kernel<false>), dim3((uint32_t)grid),
+                           dim3(block), 0, stream, in, nrows, in_stride, dim,
+                           dpad, fmul, fadd, out);
 }
 
 /* int32 sum-of-squares per quantized query (cos, narrow path) */

@@ -528,9 +528,10 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
              * search_quantize contract: query narrowed the same way as the
              * entries — quantizer.go ApplyInt8/entry SQL) */
             ix->w_queries_q.ensure((size_t)nq * dpad);
-            launch_quantize(uns, d_q, (int64_t)nq * dpad,
-                            (float)ix->quant_mul, (float)ix->quant_add,
-                            ix->w_queries_q.as<uint8_t>(), s);
+            launch_quantize_rows(uns, d_q, (int64_t)nq, dpad, (int)ix->dim,
+                                 dpad, (float)ix->quant_mul,
+                                 (float)ix->quant_add,
+                                 ix->w_queries_q.as<uint8_t>(), s);
             KCHECK("query-quantize");
             const int32_t* d_qni = nullptr;
             if (ix->kmetric == KM_COS) {
@@ -792,21 +793,6 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
         /* pack entries */
         pack_lists(ix, ix->d_staging, ix->count, slot_rows, loffsets,
                    &ix->d_packed, &ix->list_gbase, &ix->total_groups);
-        if (ix->qtype != Quantization_F32) {
-            /* quantize the PACKED f32 image elementwise (same layout: the
-             * byte interleave [g][dpad/4][64][4] equals the float4
-             * interleave with byte elements), then drop the f32 copy —
-             * narrow storage is the point (search.go:1062 quant domain). */
-            const int64_t n_elems = ix->total_groups * 64 * (int64_t)ix->dpad;
-            HIP_CHECK(hipMalloc(&ix->d_packed_q, (size_t)n_elems));
-            launch_quantize(ix->qtype == Quantization_UINT8, ix->d_packed,
-                            n_elems, (float)ix->quant_mul,
-                            (float)ix->quant_add, ix->d_packed_q, ix->stream);
-            HIP_CHECK(hipStreamSynchronize(ix->stream));
-            KCHECK("quantize");
-            HIP_CHECK(hipFree(ix->d_packed));
-            ix->d_packed = nullptr;
-        }
         /* group -> slot metadata kept for save_dir's unpack */
         ix->group_slotbase.resize(ix->total_groups);
         ix->group_valid.resize(ix->total_groups);
@@ -817,6 +803,46 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
                 ix->group_valid[g] = (int32_t)std::min<int64_t>(
                     64, (loffsets[l + 1] - loffsets[l]) - gi * 64);
             }
+        if (ix->qtype != Quantization_F32) {
+            /* quantize the staged rows (pad columns emit 0 so padded dims
+             * stay identity), byte-pack into [g][dpad/4][64][4], and drop
+             * the f32 copy — storage stays narrow (search.go:1062 quantized
+             * domain). */
+            const bool uns = ix->qtype == Quantization_UINT8;
+            DevBuf d_qrows, d_rb2, d_gv2, d_sr2;
+            d_qrows.ensure((size_t)ix->count * ix->dpad);
+            launch_quantize_rows(uns, ix->d_staging, (int64_t)ix->count,
+                                 (int)ix->dim, (int)ix->dim, (int)ix->dpad,
+                                 (float)ix->quant_mul, (float)ix->quant_add,
+                                 d_qrows.as<uint8_t>(), ix->stream);
+            KCHECK("quantize-rows");
+            d_rb2.ensure(ix->total_groups * 8);
+            d_gv2.ensure(ix->total_groups * 4);
+            d_sr2.ensure(std::max<size_t>(8, slot_rows.size() * 8));
+            HIP_CHECK(hipMemcpyAsync(d_rb2.ptr, ix->group_slotbase.data(),
+                                     ix->total_groups * 8,
+                                     hipMemcpyHostToDevice, ix->stream));
+            HIP_CHECK(hipMemcpyAsync(d_gv2.ptr, ix->group_valid.data(),
+                                     ix->total_groups * 4,
+                                     hipMemcpyHostToDevice, ix->stream));
+            /* pack wants slot->row; group_slotbase already holds the slot
+             * base per group, slot_rows maps slot->row */
+            if (!slot_rows.empty())
+                HIP_CHECK(hipMemcpyAsync(d_sr2.ptr, slot_rows.data(),
+                                         slot_rows.size() * 8,
+                                         hipMemcpyHostToDevice, ix->stream));
+            HIP_CHECK(hipMalloc(&ix->d_packed_q,
+                                (size_t)ix->total_groups * 64 * ix->dpad));
+            launch_pq_pack(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                           d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
+                           d_sr2.as<int64_t>(), ix->total_groups,
+                           ix->d_packed_q, ix->stream);
+            HIP_CHECK(hipStreamSynchronize(ix->stream));
+            KCHECK("quantize-pack");
+            HIP_CHECK(hipFree(ix->d_packed));
+            ix->d_packed = nullptr;
+        }
+
 
         /* id map: slot -> entry id */
         std::vector<int64_t> id_by_slot(std::max<uint64_t>(1, ix->count));

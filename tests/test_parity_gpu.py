@@ -346,3 +346,30 @@ def test_quantized_storage(qtype, metric):
     ids, dists = ix.search(queries, k, probe)
     _assert_parity(ids, dists, ref_ids, ref_d, ctx=f"{qtype}/{metric}")
     ix.close()
+
+
+def test_quantized_storage_padded_dim():
+    """dim not divisible by 4: padded columns must stay identity under
+    quantization (q(0) != 0 would poison IP/cos sums)."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(910))
+    n, d, nlist, nq, probe, k = 4000, 67, 16, 8, 4, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    vmin, vmax = float(vecs.min()), float(vecs.max())
+    mul, add = orc.int8_params(vmin, vmax)
+    idx = orc.IvfIndex(cents, vecs, assign)
+    ref_ids, ref_d = orc.ivf_search_quantized(
+        idx, orc.METRIC_IP, queries, probe, k, mul, add)
+    ix = engine.IvfFlatIndex(d, nlist, metric="ip", capacity=n, qtype="int8")
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.set_quantizer(vmin, vmax)
+    ix.build()
+    ids, dists = ix.search(queries, k, probe)
+    _assert_parity(ids, dists, ref_ids, ref_d, ctx="int8-ip-d67")
+    ix.close()
