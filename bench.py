@@ -190,9 +190,10 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
+    backend = os.environ.get("MOANN_BENCH_BACKEND", "nccl")
     if distributed:
         import torch.distributed as dist
-        dist.init_process_group("nccl")
+        dist.init_process_group(backend)
     torch.cuda.set_device(local_rank)
     dev = torch.device("cuda", local_rank)
 
@@ -251,13 +252,13 @@ def main():
         off = rank * args.rows
         ids_off = np.where(ids_r >= 0, ids_r + off, -1)
         from matrixone_amd.distributed import allgather_merge
-        m_ids, _ = allgather_merge(ids_off, dists_r, args.k, device=dev)
+        m_ids, _ = allgather_merge(ids_off, dists_r, args.k)
         # global GT: merge per-shard exact GT by distance
         gt_dist = gt_exact_dists(data, queries[:nrq], gt_local)
         g_ids, _ = allgather_merge(
             np.where(gt_local.cpu().numpy() >= 0,
                      gt_local.cpu().numpy() + off, -1),
-            gt_dist.cpu().numpy().astype(np.float32), args.k, device=dev)
+            gt_dist.cpu().numpy().astype(np.float32), args.k)
         recall = float(np.mean([
             len(set(m_ids[i]).intersection(g_ids[i])) / args.k
             for i in range(nrq)]))
@@ -274,7 +275,7 @@ def main():
             from matrixone_amd.distributed import allgather_merge
             off = rank * args.rows
             ids = np.where(ids >= 0, ids + off, -1)
-            ids, dists = allgather_merge(ids, dists, args.k, device=dev)
+            ids, dists = allgather_merge(ids, dists, args.k)
         return ids
 
     for i in range(args.warmup):
@@ -294,7 +295,8 @@ def main():
     elapsed = time.perf_counter() - t0
     if distributed:
         import torch.distributed as dist
-        t = torch.tensor([elapsed], device=dev)
+        t = (torch.tensor([elapsed], device=dev)
+             if dist.get_backend() == "nccl" else torch.tensor([elapsed]))
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

@@ -180,10 +180,7 @@ def test_duplicate_vectors_ties():
     ix.close()
 
 
-def test_hnsw_placeholder_absent():
-    """HNSW GPU path is SURVEY.md §8 row a3 — lands in a later commit of this
-    round / round 2; this marker keeps the scope honest in CI output."""
-    pytest.skip("HNSW GPU probe not yet implemented (§8a3, planned)")
+# (HNSW GPU coverage lives in tests/test_hnsw_gpu.py)
 
 
 def test_filtered_search():
