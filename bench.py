@@ -230,8 +230,9 @@ def main():
     ix.set_centroids(cent_np)
     ix.set_assignments(assign_np)
     if args.qtype != "f32":
-        lo = float(torch.quantile(data[:65536].reshape(-1).float(), 0.001))
-        hi = float(torch.quantile(data[:65536].reshape(-1).float(), 0.999))
+        sample = data[:2048].reshape(-1).float()  # quantile caps at 2^24
+        lo = float(torch.quantile(sample, 0.001))
+        hi = float(torch.quantile(sample, 0.999))
         ix.set_quantizer(lo, hi)
     ix.build()
     log(f"[bench] build done in {time.perf_counter() - t_build0:.1f}s")
