@@ -194,8 +194,11 @@ def main():
     if distributed:
         import torch.distributed as dist
         dist.init_process_group(backend)
-    torch.cuda.set_device(local_rank)
-    dev = torch.device("cuda", local_rank)
+    # modulo lets N-rank validation run on a box with fewer GPUs (the
+    # driver's 8-GPU run maps ranks to distinct devices unchanged)
+    devno = local_rank % max(1, torch.cuda.device_count())
+    torch.cuda.set_device(devno)
+    dev = torch.device("cuda", devno)
 
     from matrixone_amd import engine
     from matrixone_amd.distributed import merge_topk_torch
@@ -219,7 +222,7 @@ def main():
     need_host = (rank == 0 and world == 1 and not args.skip_cpu_baseline)
     host_vecs = data.cpu().numpy() if need_host else None
     ix = engine.IvfFlatIndex(args.dim, args.nlist, metric="l2sq",
-                             capacity=args.rows, device=local_rank,
+                             capacity=args.rows, device=devno,
                              qtype=args.qtype)
     CH = 1 << 21
     for c0 in range(0, args.rows, CH):

@@ -116,98 +116,6 @@ __global__ __launch_bounds__(256) void scan_kernel(
         for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0.f;
         float rn0 = 0.f, rn1 = 0.f;
 
-        /* GP==4 (selected as MOANN_QT=9): the two-group QT=8 shape with a
-         * 3-deep PAIR ring in inline asm, ordered by form (ii) — the
-         * counted wait NAMES the ring registers as "+v" operands, so
-         * ordering comes from data dependence and the LDS reads stay free
-         * to schedule (no sched_barrier fence, unlike the GP==1 ring).
-         * 6 dwordx4 in flight; vmcnt(4) retires the oldest pair. */
-        if (GP == 4) {
-            float4 ra0, rb0, ra1, rb1, ra2, rb2;
-#define MOANN_PISS(A, B, idx)                                               \
-            {                                                               \
-                const int ii = (idx) < d4 ? (idx) : d4 - 1;                 \
-                const float4* p0 = d0 + ii * 64;                            \
-                const float4* p1 = d1 + ii * 64;                            \
-                asm volatile("global_load_dwordx4 %0, %2, off\n\t"          \
-                             "global_load_dwordx4 %1, %3, off"              \
-                             : "=v"(A), "=v"(B) : "v"(p0), "v"(p1));        \
-            }
-#define MOANN_PBODY(A, B, idx)                                              \
-            {                                                               \
-                const float4 x0 = A;                                        \
-                const float4 x1 = B;                                        \
-                const int q4 = (idx);                                       \
-                if (METRIC == KM_COS) {                                     \
-                    rn0 += x0.x * x0.x + x0.y * x0.y + x0.z * x0.z +        \
-                           x0.w * x0.w;                                     \
-                    rn1 += x1.x * x1.x + x1.y * x1.y + x1.z * x1.z +        \
-                           x1.w * x1.w;                                     \
-                }                                                           \
-                _Pragma("unroll")                                           \
-                for (int t = 0; t < QT; ++t) {                              \
-                    const float4 qv =                                       \
-                        ((const float4*)(ldsq + t * dpad))[q4];             \
-                    if (METRIC == KM_L2SQ) {                                \
-                        float e;                                            \
-                        e = x0.x - qv.x; acc0[t] = fmaf(e, e, acc0[t]);     \
-                        e = x0.y - qv.y; acc0[t] = fmaf(e, e, acc0[t]);     \
-                        e = x0.z - qv.z; acc0[t] = fmaf(e, e, acc0[t]);     \
-                        e = x0.w - qv.w; acc0[t] = fmaf(e, e, acc0[t]);     \
-                        e = x1.x - qv.x; acc1[t] = fmaf(e, e, acc1[t]);     \
-                        e = x1.y - qv.y; acc1[t] = fmaf(e, e, acc1[t]);     \
-                        e = x1.z - qv.z; acc1[t] = fmaf(e, e, acc1[t]);     \
-                        e = x1.w - qv.w; acc1[t] = fmaf(e, e, acc1[t]);     \
-                    } else if (METRIC == KM_IP || METRIC == KM_COS) {       \
-                        acc0[t] = fmaf(x0.x, qv.x, acc0[t]);                \
-                        acc0[t] = fmaf(x0.y, qv.y, acc0[t]);                \
-                        acc0[t] = fmaf(x0.z, qv.z, acc0[t]);                \
-                        acc0[t] = fmaf(x0.w, qv.w, acc0[t]);                \
-                        acc1[t] = fmaf(x1.x, qv.x, acc1[t]);                \
-                        acc1[t] = fmaf(x1.y, qv.y, acc1[t]);                \
-                        acc1[t] = fmaf(x1.z, qv.z, acc1[t]);                \
-                        acc1[t] = fmaf(x1.w, qv.w, acc1[t]);                \
-                    } else {                                                \
-                        acc0[t] += fabsf(x0.x - qv.x) + fabsf(x0.y - qv.y) +\
-                                   fabsf(x0.z - qv.z) + fabsf(x0.w - qv.w); \
-                        acc1[t] += fabsf(x1.x - qv.x) + fabsf(x1.y - qv.y) +\
-                                   fabsf(x1.z - qv.z) + fabsf(x1.w - qv.w); \
-                    }                                                       \
-                }                                                           \
-            }
-#define MOANN_PCONS(A, B, idx)                                              \
-            {                                                               \
-                asm volatile("s_waitcnt vmcnt(4)"                           \
-                             : "+v"(A.x), "+v"(A.y), "+v"(A.z), "+v"(A.w),  \
-                               "+v"(B.x), "+v"(B.y), "+v"(B.z), "+v"(B.w)); \
-                MOANN_PBODY(A, B, idx)                                      \
-            }
-            MOANN_PISS(ra0, rb0, 0)
-            MOANN_PISS(ra1, rb1, 1)
-            MOANN_PISS(ra2, rb2, 2)
-            int q4 = 0;
-            for (; q4 + 3 <= d4; q4 += 3) {
-                MOANN_PCONS(ra0, rb0, q4)
-                MOANN_PISS(ra0, rb0, q4 + 3)
-                MOANN_PCONS(ra1, rb1, q4 + 1)
-                MOANN_PISS(ra1, rb1, q4 + 4)
-                MOANN_PCONS(ra2, rb2, q4 + 2)
-                MOANN_PISS(ra2, rb2, q4 + 5)
-            }
-            asm volatile(
-                "s_waitcnt vmcnt(0)"
-                : "+v"(ra0.x), "+v"(ra0.y), "+v"(ra0.z), "+v"(ra0.w),
-                  "+v"(rb0.x), "+v"(rb0.y), "+v"(rb0.z), "+v"(rb0.w),
-                  "+v"(ra1.x), "+v"(ra1.y), "+v"(ra1.z), "+v"(ra1.w),
-                  "+v"(rb1.x), "+v"(rb1.y), "+v"(rb1.z), "+v"(rb1.w),
-                  "+v"(ra2.x), "+v"(ra2.y), "+v"(ra2.z), "+v"(ra2.w),
-                  "+v"(rb2.x), "+v"(rb2.y), "+v"(rb2.z), "+v"(rb2.w));
-            if (q4 < d4) { MOANN_PBODY(ra0, rb0, q4) }
-            if (q4 + 1 < d4) { MOANN_PBODY(ra1, rb1, q4 + 1) }
-#undef MOANN_PISS
-#undef MOANN_PBODY
-#undef MOANN_PCONS
-        } else
         /* GP==1: hand-rolled 3-deep load ring in inline asm. hipcc will not
          * software-pipeline this loop at QT=16 register pressure (it emits a
          * full vmcnt(0) drain per iteration); the asm ring keeps 2 loads in
@@ -392,10 +300,6 @@ static void launch_scan_qt(int qt, const float* packed, const float* queries,
     case 16:
         hipLaunchKernelGGL((scan_kernel<METRIC, 16, 1>), grid, block,
                            (16 * dpad + 16) * sizeof(float), stream, ARGS);
-        break;
-    case 9: /* QT=8 with the form-(ii) asm pair ring (A/B via MOANN_QT=9) */
-        hipLaunchKernelGGL((scan_kernel<METRIC, 8, 4>), grid, block,
-                           (8 * dpad + 8) * sizeof(float), stream, ARGS);
         break;
     case 8:
         hipLaunchKernelGGL((scan_kernel<METRIC, 8, 2>), grid, block,
