@@ -306,6 +306,10 @@ def main():
 
     qps = args.steps * args.batch / elapsed
     perf = ix.perf()
+    log(f"[bench] perf: rank {perf['rank_ms']/args.steps:.3f} ms/step, "
+        f"scan {perf['scan_ms']/args.steps:.3f}, "
+        f"select {perf['select_ms']/args.steps:.3f}, "
+        f"step total {elapsed/args.steps*1e3:.3f}")
 
     # ---- roofline (list-scan kernel, HIP events on the index stream) ------
     elem = 1 if args.qtype != "f32" else 4
