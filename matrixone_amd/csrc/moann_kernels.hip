@@ -301,6 +301,10 @@ static void launch_scan_qt(int qt, const float* packed, const float* queries,
         hipLaunchKernelGGL((scan_kernel<METRIC, 16, 1>), grid, block,
                            (16 * dpad + 16) * sizeof(float), stream, ARGS);
         break;
+    case 12:
+        hipLaunchKernelGGL((scan_kernel<METRIC, 12, 2>), grid, block,
+                           (12 * dpad + 12) * sizeof(float), stream, ARGS);
+        break;
     case 8:
         hipLaunchKernelGGL((scan_kernel<METRIC, 8, 2>), grid, block,
                            (8 * dpad + 8) * sizeof(float), stream, ARGS);

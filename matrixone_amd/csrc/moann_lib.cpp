@@ -202,7 +202,7 @@ int pick_qt(uint32_t dpad) {
     /* MOANN_QT overrides for A/B tuning */
     if (const char* e = getenv("MOANN_QT")) {
         const int v = atoi(e);
-        if (v == 1 || v == 2 || v == 4 || v == 8 || v == 16) return v;
+        if (v == 1 || v == 2 || v == 4 || v == 8 || v == 12 || v == 16) return v;
     }
     /* QT query rows in LDS; <= 52 KiB keeps >=3 blocks/CU resident.
      * QT=16 halves list re-reads when many queries probe one list (the
