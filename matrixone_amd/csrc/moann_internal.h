@@ -43,6 +43,14 @@ void launch_scan(int metric, int qt, const float* packed, const float* queries,
                  float* dists_out, hipStream_t stream,
                  const uint32_t* filter_bitset = nullptr);
 
+/* Hand-scheduled asm specialization of the scan for the flagship shape
+ * (scan_asm768.hip): METRIC=L2SQ, dpad=768, QT=16, one group per wave.
+ * launch_scan dispatches here itself when qt==16 fits; exposed for tests. */
+void launch_scan_asm768(const float* packed, const float* queries,
+                        int dpad, const ScanJobs& jobs, float* dists_out,
+                        hipStream_t stream,
+                        const uint32_t* filter_bitset = nullptr);
+
 /* Per-query ascending top-k select (radix select + in-LDS bitonic sort).
  * Candidates for query q live at dists[off(q) .. off(q)+count(q)) where
  * off/count come from qoffs (ragged, [nq+1]) or uniform n (qoffs==nullptr).

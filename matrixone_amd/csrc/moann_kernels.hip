@@ -20,6 +20,8 @@
 #include <hip/hip_runtime.h>
 #include <cfloat>
 #include <cstdint>
+#include <cstdlib>
+#include <cstring>
 
 #include "moann_internal.h"
 
@@ -330,6 +332,19 @@ void launch_scan(int metric, int qt, const float* packed, const float* queries,
                  float* dists_out, hipStream_t stream,
                  const uint32_t* filter_bitset) {
     if (jb.njobs == 0) return;
+    /* flagship shape -> hand-scheduled asm kernel (scan_asm768.hip);
+     * MOANN_SCAN=generic forces the compiler-scheduled path for A/B */
+    if (metric == KM_L2SQ && qt == 16 && dpad == 768) {
+        static const bool generic = [] {
+            const char* e = getenv("MOANN_SCAN");
+            return e && strcmp(e, "generic") == 0;
+        }();
+        if (!generic) {
+            launch_scan_asm768(packed, queries, dpad, jb, dists_out, stream,
+                               filter_bitset);
+            return;
+        }
+    }
     switch (metric) {
     case KM_L2SQ: launch_scan_qt<KM_L2SQ>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
     case KM_IP:   launch_scan_qt<KM_IP>(qt, packed, queries, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;

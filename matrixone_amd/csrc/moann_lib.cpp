@@ -198,11 +198,18 @@ struct IvfIndex {
     }
 };
 
-int pick_qt(uint32_t dpad) {
+int pick_qt(int kmetric, uint32_t dpad, bool narrow) {
     /* MOANN_QT overrides for A/B tuning */
     if (const char* e = getenv("MOANN_QT")) {
         const int v = atoi(e);
         if (v == 1 || v == 2 || v == 4 || v == 8 || v == 12 || v == 16) return v;
+    }
+    /* flagship f32 shape: the hand-scheduled QT=16 asm kernel
+     * (scan_asm768.hip) — lowest tile re-read multiplicity AND a pipelined
+     * load ring. MOANN_SCAN=generic drops back to the QT=8 policy below. */
+    if (!narrow && kmetric == KM_L2SQ && dpad == 768) {
+        const char* e = getenv("MOANN_SCAN");
+        if (!(e && strcmp(e, "generic") == 0)) return 16;
     }
     /* QT query rows in LDS; <= 52 KiB keeps >=3 blocks/CU resident.
      * QT=16 halves list re-reads when many queries probe one list (the
@@ -317,7 +324,8 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
 
     /* 2. centroid rank: scan the centroid matrix (one "list" of nlist rows),
      * then top-probe per query (ivfflat/search.go:265-308 rankCentroids). */
-    const int qt = pick_qt(ix->dpad);
+    const int qt = pick_qt(ix->kmetric, ix->dpad,
+                           ix->qtype != Quantization_F32);
     if (use_mfma_rank) {
         ix->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         HIP_CHECK(hipEventRecord(ix->perf.ev[0], s));
