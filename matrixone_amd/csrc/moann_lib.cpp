@@ -457,38 +457,52 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
             }
     }
 
-    /* jobs: per list, query tiles of qt; split groups if too few jobs */
-    int64_t tiles_total = 0;
-    for (uint32_t l = 0; l < ix->nlist; ++l)
-        if (lcount[l]) tiles_total += (lcount[l] + qt - 1) / qt;
-    int split = 1;
-    if (tiles_total > 0 && tiles_total < 2048)
-        split = (int)std::min<int64_t>(64, (2048 + tiles_total - 1) / tiles_total);
+    /* jobs: per list, query tiles of qt x group CHUNKS.
+     * Chunking serves two ends at once:
+     *  - tail balance: a job is <= ~3.5 MB of list data, so hub lists
+     *    (many rows AND many query tiles) become many small jobs instead
+     *    of one straggler workgroup;
+     *  - L2 reuse: jobs are emitted chunk-OUTER, tile-INNER, and the
+     *    XCD-aware remap in the kernel gives each XCD a contiguous job
+     *    range — so the ~96 co-resident jobs of an XCD are different
+     *    query tiles over the SAME <=3.5 MB chunk (< 4 MB per-XCD L2),
+     *    and the tile re-read multiplicity is served from L2, not HBM.
+     * CHUNK_G=18 groups = 18*64 rows * dpad*4 B (3.45 MB at dpad=768);
+     * MOANN_CHUNKG overrides for tuning. */
+    static const int64_t CHUNK_G = [] {
+        const char* e = getenv("MOANN_CHUNKG");
+        const int v = e ? atoi(e) : 0;
+        return (int64_t)(v > 0 ? v : 18);
+    }();
 
     std::vector<int64_t> databaseg, qslot_outbase, jslot_base;
     std::vector<int32_t> gstart, gcount, rows, jnq, qbase, qslot_query;
+    std::vector<int32_t> tile_qb, tile_nq; /* per-tile scratch, reused */
     for (uint32_t l = 0; l < ix->nlist; ++l) {
         if (!lcount[l]) continue;
         const int64_t lg = (ix->list_rows[l] + 63) / 64;
-        const int64_t chunk = std::max<int64_t>(1, (lg + split - 1) / split);
+        tile_qb.clear();
+        tile_nq.clear();
         for (int64_t t0 = 0; t0 < lcount[l]; t0 += qt) {
             const int tn = (int)std::min<int64_t>(qt, lcount[l] - t0);
-            const int32_t qb = (int32_t)qslot_query.size();
+            tile_qb.push_back((int32_t)qslot_query.size());
+            tile_nq.push_back(tn);
             for (int u = 0; u < qt; ++u) {
                 const int64_t pi = lbase[l] + t0 + std::min<int64_t>(u, tn - 1);
                 qslot_query.push_back(pair_q[pi]);
                 qslot_outbase.push_back(pair_ob[pi]);
             }
-            for (int64_t g0 = 0; g0 < lg; g0 += chunk) {
+        }
+        for (int64_t g0 = 0; g0 < lg; g0 += CHUNK_G)
+            for (size_t ti = 0; ti < tile_qb.size(); ++ti) {
                 jslot_base.push_back(ix->list_slot_base[l]);
                 databaseg.push_back(ix->list_gbase[l]);
                 gstart.push_back((int32_t)g0);
-                gcount.push_back((int32_t)std::min<int64_t>(chunk, lg - g0));
+                gcount.push_back((int32_t)std::min<int64_t>(CHUNK_G, lg - g0));
                 rows.push_back(ix->list_rows[l]);
-                jnq.push_back(tn);
-                qbase.push_back(qb);
+                jnq.push_back(tile_nq[ti]);
+                qbase.push_back(tile_qb[ti]);
             }
-        }
     }
     const int njobs = (int)databaseg.size();
     const int nqslots = (int)qslot_query.size();
