@@ -467,12 +467,14 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
      *    range — so the ~96 co-resident jobs of an XCD are different
      *    query tiles over the SAME <=3.5 MB chunk (< 4 MB per-XCD L2),
      *    and the tile re-read multiplicity is served from L2, not HBM.
-     * CHUNK_G=18 groups = 18*64 rows * dpad*4 B (3.45 MB at dpad=768);
+     * CHUNK_G=8 groups = 8*64 rows * dpad*4 B (1.5 MB at dpad=768; measured best
+     * of {6,8,10,12,14,18,36} end-to-end — small chunks also win on tail
+     * balance);
      * MOANN_CHUNKG overrides for tuning. */
     static const int64_t CHUNK_G = [] {
         const char* e = getenv("MOANN_CHUNKG");
         const int v = e ? atoi(e) : 0;
-        return (int64_t)(v > 0 ? v : 18);
+        return (int64_t)(v > 0 ? v : 8);
     }();
 
     std::vector<int64_t> databaseg, qslot_outbase, jslot_base;
