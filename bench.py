@@ -308,6 +308,7 @@ def main():
     perf = ix.perf()
     log(f"[bench] perf: rank {perf['rank_ms']/args.steps:.3f} ms/step, "
         f"scan {perf['scan_ms']/args.steps:.3f}, "
+        f"other {perf.get('other_ms', 0.0)/args.steps:.3f}, "
         f"select {perf['select_ms']/args.steps:.3f}, "
         f"step total {elapsed/args.steps*1e3:.3f}")
 

@@ -375,21 +375,35 @@ __global__ __launch_bounds__(256) void topk_kernel(
     const int64_t count = (qoffs ? qoffs[q + 1] : (int64_t)(q + 1) * uniform_n) - off;
     const int tid = threadIdx.x;
     const int m = (int)(count < (int64_t)k ? count : k);
+    int nkeys;
 
     if (count > (int64_t)k) {
-        /* radix select for the exact threshold */
-        if (tid == 0) { st[0] = 0; st[1] = 0; st[3] = 0; st[4] = 0; }
+        /* radix select for the exact threshold. Each full-array read is
+         * 4x unrolled so every wave keeps 4 loads in flight (a single
+         * strided scalar load per wave is latency-bound at ~1.3 TB/s;
+         * measured r01_fetch_10M_asm_cg8.csv). */
+        if (tid == 0) { st[0] = 0; st[1] = 0; st[3] = 0; st[4] = 0; st[5] = 0; }
         __syncthreads();
         for (int level = 0; level < 4 && !st[4]; ++level) {
             const int shift = 24 - 8 * level;
             for (int b = tid; b < 256; b += blockDim.x) hist[b] = 0;
             __syncthreads();
             const uint32_t pref = st[0];
-            for (int64_t i = tid; i < count; i += blockDim.x) {
-                const uint32_t u = f2u(dists[off + i]);
-                if (level == 0 || (u >> (shift + 8)) == (pref >> (shift + 8)))
-                    atomicAdd(&hist[(u >> shift) & 255], 1u);
+            const int64_t step = blockDim.x;
+            int64_t i = tid;
+#define MOANN_HISTO(u)                                                                  if (level == 0 || ((u) >> (shift + 8)) == (pref >> (shift + 8)))                atomicAdd(&hist[((u) >> shift) & 255], 1u);
+            for (; i + 3 * step < count; i += 4 * step) {
+                const uint32_t u0 = f2u(dists[off + i]);
+                const uint32_t u1 = f2u(dists[off + i + step]);
+                const uint32_t u2 = f2u(dists[off + i + 2 * step]);
+                const uint32_t u3 = f2u(dists[off + i + 3 * step]);
+                MOANN_HISTO(u0) MOANN_HISTO(u1) MOANN_HISTO(u2) MOANN_HISTO(u3)
             }
+            for (; i < count; i += step) {
+                const uint32_t u = f2u(dists[off + i]);
+                MOANN_HISTO(u)
+            }
+#undef MOANN_HISTO
             __syncthreads();
             if (tid == 0) {
                 const uint32_t need = (uint32_t)k - st[1];
@@ -405,9 +419,21 @@ __global__ __launch_bounds__(256) void topk_kernel(
                     st[1] += hist[b];
                     st[0] |= (shift ? ((1u << shift) - 1u) : 0u);
                     st[3] = 0; st[4] = 1;
+                } else if (st[1] + hist[b] <= (uint32_t)sortP) {
+                    /* boundary GROUP fits in the sort buffer: collect the
+                     * whole prefix group and let the bitonic sort truncate.
+                     * Exact (every candidate strictly below the group is
+                     * kept) and deterministic (full keys are distinct), and
+                     * it usually ends the scan at level 0 or 1 instead of
+                     * re-reading the array for levels 2-3. */
+                    st[2] = st[0];
+                    st[3] = hist[b];
+                    st[5] = (uint32_t)shift;
+                    st[4] = 1;
                 } else if (level == 3) {
                     st[2] = st[0];
                     st[3] = need - cum;
+                    st[5] = 0;
                 }
             }
             __syncthreads();
@@ -415,31 +441,44 @@ __global__ __launch_bounds__(256) void topk_kernel(
         const uint32_t c_sure = st[1];
         const uint32_t tieval = st[2];
         const uint32_t need_ties = st[3];
-        const int early = st[4];
-        /* strict-below bound: early-exit keeps everything <= prefix (low bits
-         * already forced to 1), full depth keeps u < prefix exactly */
-        const uint64_t below = (uint64_t)st[0] + (early ? 1u : 0u);
+        const uint32_t tie_shift = st[5];
+        const int exact = st[4] && !need_ties;
+        /* strict-below bound: exact-exit keeps everything <= prefix (low
+         * bits already forced to 1); group/tie modes keep u < prefix (low
+         * bits zero, so < excludes the whole boundary group) */
+        const uint64_t below = (uint64_t)st[0] + (exact ? 1u : 0u);
         if (tid == 0) { ctr[0] = 0; ctr[1] = 0; }
         __syncthreads();
-        for (int64_t i = tid; i < count; i += blockDim.x) {
-            const uint32_t u = f2u(dists[off + i]);
-            if ((uint64_t)u < below) {
-                const uint32_t pos = atomicAdd(&ctr[0], 1u);
-                keys[pos] = ((uint64_t)u << 32) | (uint32_t)i;
-            } else if (need_ties && u == tieval) {
-                const uint32_t t = atomicAdd(&ctr[1], 1u);
-                if (t < need_ties)
-                    keys[c_sure + t] = ((uint64_t)u << 32) | (uint32_t)i;
+        {
+            const int64_t step = blockDim.x;
+            int64_t i = tid;
+#define MOANN_COLLECT(u, idx)                                                           if ((uint64_t)(u) < below) {                                                    const uint32_t pos = atomicAdd(&ctr[0], 1u);                                keys[pos] = ((uint64_t)(u) << 32) | (uint32_t)(idx);                    } else if (need_ties &&                                                                ((u) >> tie_shift) == (tieval >> tie_shift)) {                       const uint32_t t = atomicAdd(&ctr[1], 1u);                                  if (t < need_ties)                                                              keys[c_sure + t] = ((uint64_t)(u) << 32) | (uint32_t)(idx);             }
+            for (; i + 3 * step < count; i += 4 * step) {
+                const uint32_t u0 = f2u(dists[off + i]);
+                const uint32_t u1 = f2u(dists[off + i + step]);
+                const uint32_t u2 = f2u(dists[off + i + 2 * step]);
+                const uint32_t u3 = f2u(dists[off + i + 3 * step]);
+                MOANN_COLLECT(u0, i) MOANN_COLLECT(u1, i + step)
+                MOANN_COLLECT(u2, i + 2 * step) MOANN_COLLECT(u3, i + 3 * step)
             }
+            for (; i < count; i += step) {
+                const uint32_t u = f2u(dists[off + i]);
+                MOANN_COLLECT(u, i)
+            }
+#undef MOANN_COLLECT
         }
         __syncthreads();
+        nkeys = (int)(c_sure + need_ties);
     } else {
         for (int64_t i = tid; i < count; i += blockDim.x)
             keys[i] = ((uint64_t)f2u(dists[off + i]) << 32) | (uint32_t)i;
         __syncthreads();
+        nkeys = m;
     }
 
-    for (int i = m + tid; i < sortP; i += blockDim.x) keys[i] = ~0ULL;
+    /* pad from the number of COLLECTED keys (group-collect gathers up to
+     * sortP entries, more than k) and let the sort truncate to k */
+    for (int i = nkeys + tid; i < sortP; i += blockDim.x) keys[i] = ~0ULL;
     __syncthreads();
 
     /* bitonic sort ascending over sortP u64 keys */

@@ -304,6 +304,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         (ix->quant_mul != 0.0 && ix->quant_mul != 1.0)
             ? 1.0 / (ix->quant_mul * ix->quant_mul) : 1.0;
 
+    const auto t_all0 = std::chrono::steady_clock::now();
     /* 1. queries to device, padded to dpad */
     ix->w_queries.ensure((size_t)nq * dpad * 4);
     if ((int)ix->dim != dpad)
@@ -610,21 +611,26 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     HIP_CHECK(hipGetLastError());
 
     /* perf accounting */
-    float ms = 0;
-    HIP_CHECK(hipEventElapsedTime(&ms, ix->perf.ev[0], ix->perf.ev[1]));
+    float r_ms = 0, sc_ms = 0, sel_ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&r_ms, ix->perf.ev[0], ix->perf.ev[1]));
     ix->perf.c.rank_launches++;
-    ix->perf.c.rank_ms += ms;
+    ix->perf.c.rank_ms += r_ms;
     ix->perf.c.rank_flops += 2ull * nq * ix->nlist * ix->dim;
-    HIP_CHECK(hipEventElapsedTime(&ms, ix->perf.ev[2], ix->perf.ev[3]));
+    HIP_CHECK(hipEventElapsedTime(&sc_ms, ix->perf.ev[2], ix->perf.ev[3]));
     uint64_t unique_rows = 0;
     for (uint32_t l = 0; l < ix->nlist; ++l)
         if (lcount[l]) unique_rows += (uint64_t)ix->list_rows[l];
     ix->perf.c.scan_launches++;
-    ix->perf.c.scan_ms += ms;
+    ix->perf.c.scan_ms += sc_ms;
     ix->perf.c.scan_bytes += unique_rows * ix->dim * 4;
     ix->perf.c.scan_rows += (uint64_t)total_cand;
-    HIP_CHECK(hipEventElapsedTime(&ms, ix->perf.ev[4], ix->perf.ev[5]));
-    ix->perf.c.select_ms += ms;
+    HIP_CHECK(hipEventElapsedTime(&sel_ms, ix->perf.ev[4], ix->perf.ev[5]));
+    ix->perf.c.select_ms += sel_ms;
+    /* everything the three event windows miss: H2D/D2H copies, the probe
+     * sync, host job build, gather — per-call wall minus kernel windows */
+    const double wall = std::chrono::duration<double, std::milli>(
+        std::chrono::steady_clock::now() - t_all0).count();
+    ix->perf.c.other_ms += wall - r_ms - sc_ms - sel_ms;
     return res;
 }
 
