@@ -271,10 +271,18 @@ def main():
     # ---- timed region ------------------------------------------------------
     nbatches = max(1, queries.shape[0] // args.batch)
 
-    def step(i):
+    # depth-2 pipelined stepping (moann_ivf_flat_search_submit): batch i+1's
+    # centroid rank + host job build overlap batch i's list scan. Disable
+    # with MOANN_BENCH_PIPELINE=0 for A/B. Every batch is fully computed and
+    # collected inside the timed region.
+    pipeline = os.environ.get("MOANN_BENCH_PIPELINE", "1") != "0"
+
+    def submit(i):
         b0 = (i % nbatches) * args.batch
         q = queries[b0:b0 + args.batch].contiguous()
-        ids, dists = ix.search_device(q, args.k, args.nprobe)
+        return ix.search_submit(q, args.k, args.nprobe)
+
+    def merge(ids, dists):
         if distributed:
             from matrixone_amd.distributed import allgather_merge
             off = rank * args.rows
@@ -282,16 +290,32 @@ def main():
             ids, dists = allgather_merge(ids, dists, args.k)
         return ids
 
-    for i in range(args.warmup):
-        step(i)
+    def step(i):
+        b0 = (i % nbatches) * args.batch
+        q = queries[b0:b0 + args.batch].contiguous()
+        ids, dists = ix.search_device(q, args.k, args.nprobe)
+        return merge(ids, dists)
+
+    def run_steps(n):
+        if not pipeline:
+            for i in range(n):
+                step(i)
+            return
+        t = submit(0)
+        for i in range(n):
+            nxt = submit(i + 1) if i + 1 < n else None
+            ids, dists = ix.search_collect(t)
+            merge(ids, dists)
+            t = nxt
+
+    run_steps(args.warmup)
     ix.perf_reset()
     if distributed:
         import torch.distributed as dist
         dist.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        step(i)
+    run_steps(args.steps)
     torch.cuda.synchronize()
     if distributed:
         import torch.distributed as dist

@@ -159,6 +159,19 @@ gpu_ivf_flat_search_res_t gpu_ivf_flat_search(gpu_ivf_flat_c index_c,
                                               ivf_flat_search_params_t search_params,
                                               void* errmsg);
 
+/* moann extension — pipelined batch submit/wait. Submit runs the centroid
+ * rank and the host job build immediately and enqueues the list scan and
+ * result readback without waiting; a second submit overlaps its rank/build
+ * with the first submit's scan (two internal HIP streams). Collect with
+ * moann_ivf_flat_result_wait or any gpu_ivf_flat_get_*; at most two
+ * submits in flight, FIFO. queries_on_device: 0 = host f32 rows, 1 =
+ * device pointer. */
+gpu_ivf_flat_search_res_t moann_ivf_flat_search_submit(
+    gpu_ivf_flat_c index_c, const void* queries, int queries_on_device,
+    uint64_t num_queries, uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, void* errmsg);
+int moann_ivf_flat_result_wait(gpu_ivf_flat_result_c result_c, void* errmsg);
+
 /* moann extension: queries already resident in HBM (device pointer, caller
  * synchronised).  Same semantics otherwise. */
 gpu_ivf_flat_search_res_t moann_ivf_flat_search_device(gpu_ivf_flat_c index_c,

@@ -43,6 +43,12 @@ inline void moann_set_errmsg(void* errmsg, const char* context,
 struct MoannResult {
     std::vector<int64_t> neighbors;
     std::vector<float> distances;
+    /* async pipeline (moann_ivf_flat_search_submit): while !collected the
+     * payload lives in the owning SearchCtx's pinned bounce buffer and
+     * done_ev marks the D2H completion on that ctx's stream. */
+    void* owner = nullptr;        /* SearchCtx* (IVF lib) or null          */
+    hipEvent_t done_ev = nullptr; /* owned by the result; null => sync    */
+    bool collected = true;
 };
 
 /* grow-only device buffer (cuvs_worker.hpp:50-150 shape, minus RMM) */

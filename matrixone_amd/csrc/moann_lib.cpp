@@ -111,6 +111,48 @@ struct PerfState {
     }
 };
 
+/* One in-flight search pipeline: its own stream, perf events, grow-only
+ * workspaces and a pinned result bounce buffer. IvfIndex keeps two, so
+ * consecutive moann_ivf_flat_search_submit calls overlap: batch B's rank +
+ * host job build run while batch A's list scan still owns the other stream
+ * (the GPU otherwise idles ~0.8 ms per step; dynamic_batching.hpp's
+ * overlapping-worker precedent, redesigned around two HIP streams). */
+struct SearchCtx {
+    hipStream_t stream = nullptr;
+    hipEvent_t ev[6] {}; /* rank0,rank1, scan0,scan1, sel0,sel1 */
+    bool made = false;
+    DevBuf w_queries, w_qnorms, w_cdists, w_csel_slots, w_csel_dists;
+    DevBuf w_jobs, w_qslots, w_cand, w_sel_slots, w_sel_dists;
+    DevBuf w_probe_lists, w_probe_offs, w_qoffs, w_out_ids, w_out_dists;
+    DevBuf w_filter, w_queries_q, w_qnorms_i;
+    void* h_out = nullptr;   /* pinned [ids int64][dists f32] bounce */
+    size_t h_out_bytes = 0;
+    MoannResult* pending = nullptr; /* submitted, not yet collected */
+    void* parent = nullptr;         /* IvfIndex*, for perf accumulation */
+    void ensure(void* ix) {
+        parent = ix;
+        if (made) return;
+        HIP_CHECK(hipStreamCreate(&stream));
+        for (auto& e : ev) HIP_CHECK(hipEventCreate(&e));
+        made = true;
+    }
+    void ensure_hout(size_t need) {
+        if (need <= h_out_bytes) return;
+        if (h_out) (void)hipHostFree(h_out);
+        h_out = nullptr;
+        h_out_bytes = 0;
+        HIP_CHECK(hipHostMalloc(&h_out, need));
+        h_out_bytes = need;
+    }
+    ~SearchCtx() {
+        if (made) {
+            (void)hipStreamDestroy(stream);
+            for (auto& e : ev) (void)hipEventDestroy(e);
+        }
+        if (h_out) (void)hipHostFree(h_out);
+    }
+};
+
 struct IvfIndex {
     /* config */
     int device = 0;
@@ -148,11 +190,9 @@ struct IvfIndex {
     int64_t total_groups = 0;
     int64_t cent_groups = 0;
 
-    /* per-call grow-only workspaces */
-    DevBuf w_queries, w_qnorms, w_cdists, w_csel_slots, w_csel_dists;
-    DevBuf w_jobs, w_qslots, w_cand, w_sel_slots, w_sel_dists;
-    DevBuf w_probe_lists, w_probe_offs, w_qoffs, w_out_ids, w_out_dists;
-    DevBuf w_filter, w_queries_q, w_qnorms_i;
+    /* per-call workspaces: two pipeline contexts (submit/wait overlap) */
+    SearchCtx sctx[2];
+    uint64_t sparity = 0;
     std::vector<int64_t> h_id_by_slot;  /* kept for the membership filter */
     std::unordered_map<int64_t, int64_t> id_to_slot; /* lazy */
 
@@ -280,12 +320,18 @@ void pack_lists(IvfIndex* ix, const float* d_rows, int64_t nrows,
 
 /* ------------------------- the search pipeline -------------------------- */
 
+double collect_result(MoannResult* res); /* below; returns kernel-window ms */
+
 std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                                    bool queries_on_device, uint64_t nq,
                                    uint32_t limit, uint32_t n_probes,
-                                   const uint32_t* filter_words = nullptr) {
+                                   const uint32_t* filter_words = nullptr,
+                                   bool sync_end = true) {
     std::lock_guard<std::mutex> lk(ix->mu);
     HIP_CHECK(hipSetDevice(ix->device));
+    SearchCtx* cx = &ix->sctx[ix->sparity++ & 1];
+    cx->ensure(ix);
+    if (cx->pending) collect_result(cx->pending); /* FIFO reuse guard */
     auto res = std::make_unique<Result>();
     res->neighbors.assign((size_t)nq * limit, -1);
     res->distances.assign((size_t)nq * limit, FLT_MAX);
@@ -297,7 +343,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     uint32_t probe = std::min<uint32_t>(n_probes ? n_probes : 1, ix->nlist);
     if (probe > 4096) throw std::runtime_error("n_probes > 4096 unsupported");
     const int dpad = (int)ix->dpad;
-    const hipStream_t s = ix->stream;
+    const hipStream_t s = cx->stream;
     ix->perf.ensure();
     const int do_sqrt = ix->metric == DistanceType_L2SqrtExpanded;
     const double inv_mul2 =
@@ -306,21 +352,21 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
 
     const auto t_all0 = std::chrono::steady_clock::now();
     /* 1. queries to device, padded to dpad */
-    ix->w_queries.ensure((size_t)nq * dpad * 4);
+    cx->w_queries.ensure((size_t)nq * dpad * 4);
     if ((int)ix->dim != dpad)
-        HIP_CHECK(hipMemsetAsync(ix->w_queries.ptr, 0, (size_t)nq * dpad * 4, s));
-    HIP_CHECK(hipMemcpy2DAsync(ix->w_queries.ptr, (size_t)dpad * 4, queries,
+        HIP_CHECK(hipMemsetAsync(cx->w_queries.ptr, 0, (size_t)nq * dpad * 4, s));
+    HIP_CHECK(hipMemcpy2DAsync(cx->w_queries.ptr, (size_t)dpad * 4, queries,
                                (size_t)ix->dim * 4, (size_t)ix->dim * 4, nq,
                                queries_on_device ? hipMemcpyDeviceToDevice
                                                  : hipMemcpyHostToDevice, s));
-    const float* d_q = ix->w_queries.as<float>();
+    const float* d_q = cx->w_queries.as<float>();
     const float* d_qn = nullptr;
     const bool use_mfma_rank = ix->rank_mfma && ix->kmetric != KM_L1;
     if (ix->kmetric == KM_COS || use_mfma_rank) {
-        ix->w_qnorms.ensure(nq * 4);
-        launch_qnorms(d_q, (int)nq, dpad, ix->w_qnorms.as<float>(), s);
+        cx->w_qnorms.ensure(nq * 4);
+        launch_qnorms(d_q, (int)nq, dpad, cx->w_qnorms.as<float>(), s);
         KCHECK("qnorms");
-        d_qn = ix->w_qnorms.as<float>();
+        d_qn = cx->w_qnorms.as<float>();
     }
 
     /* 2. centroid rank: scan the centroid matrix (one "list" of nlist rows),
@@ -328,21 +374,21 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const int qt = pick_qt(ix->kmetric, ix->dpad,
                            ix->qtype != Quantization_F32);
     if (use_mfma_rank) {
-        ix->w_cdists.ensure((size_t)nq * ix->nlist * 4);
-        HIP_CHECK(hipEventRecord(ix->perf.ev[0], s));
+        cx->w_cdists.ensure((size_t)nq * ix->nlist * 4);
+        HIP_CHECK(hipEventRecord(cx->ev[0], s));
         launch_rank_gemm(ix->kmetric, d_q, ix->d_cent_rows, d_qn,
                          ix->d_cnorms, (int)nq, (int)ix->nlist, dpad,
-                         ix->w_cdists.as<float>(), s);
+                         cx->w_cdists.as<float>(), s);
         KCHECK("rank-gemm");
-        HIP_CHECK(hipEventRecord(ix->perf.ev[1], s));
-        ix->w_csel_slots.ensure((size_t)nq * probe * 4);
-        ix->w_csel_dists.ensure((size_t)nq * probe * 4);
-        launch_topk(ix->w_cdists.as<float>(), nullptr, ix->nlist, (int)nq,
-                    (int)probe, ix->w_csel_slots.as<int32_t>(),
-                    ix->w_csel_dists.as<float>(), s);
+        HIP_CHECK(hipEventRecord(cx->ev[1], s));
+        cx->w_csel_slots.ensure((size_t)nq * probe * 4);
+        cx->w_csel_dists.ensure((size_t)nq * probe * 4);
+        launch_topk(cx->w_cdists.as<float>(), nullptr, ix->nlist, (int)nq,
+                    (int)probe, cx->w_csel_slots.as<int32_t>(),
+                    cx->w_csel_dists.as<float>(), s);
         KCHECK("rank-topk");
     } else {
-        ix->w_cdists.ensure((size_t)nq * ix->nlist * 4);
+        cx->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         /* tile queries, chunk centroid groups to fill the chip */
         const int ntiles = (int)((nq + qt - 1) / qt);
         const int64_t cgroups = ix->cent_groups;
@@ -380,8 +426,8 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
         size_t need = al8(njobs * b64) + 5 * al8(njobs * b32) +
                       al8(nqslots * b32) + al8(nqslots * b64);
-        ix->w_jobs.ensure(need);
-        uint8_t* p = ix->w_jobs.as<uint8_t>();
+        cx->w_jobs.ensure(need);
+        uint8_t* p = cx->w_jobs.as<uint8_t>();
         ScanJobs jb;
         auto up = [&](const void* src, size_t bytes) {
             HIP_CHECK(hipMemcpyAsync(p, src, bytes, hipMemcpyHostToDevice, s));
@@ -399,23 +445,23 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         jb.qslot_outbase = (int64_t*)up(qslot_outbase.data(), nqslots * b64);
         jb.slot_base = nullptr;
         jb.njobs = njobs;
-        HIP_CHECK(hipEventRecord(ix->perf.ev[0], s));
+        HIP_CHECK(hipEventRecord(cx->ev[0], s));
         launch_scan(ix->kmetric, qt, ix->d_cent_packed, d_q, d_qn, dpad, jb,
-                    ix->w_cdists.as<float>(), s);
+                    cx->w_cdists.as<float>(), s);
         KCHECK("rank-scan");
-        HIP_CHECK(hipEventRecord(ix->perf.ev[1], s));
-        ix->w_csel_slots.ensure((size_t)nq * probe * 4);
-        ix->w_csel_dists.ensure((size_t)nq * probe * 4);
-        launch_topk(ix->w_cdists.as<float>(), nullptr, ix->nlist, (int)nq,
-                    (int)probe, ix->w_csel_slots.as<int32_t>(),
-                    ix->w_csel_dists.as<float>(), s);
+        HIP_CHECK(hipEventRecord(cx->ev[1], s));
+        cx->w_csel_slots.ensure((size_t)nq * probe * 4);
+        cx->w_csel_dists.ensure((size_t)nq * probe * 4);
+        launch_topk(cx->w_cdists.as<float>(), nullptr, ix->nlist, (int)nq,
+                    (int)probe, cx->w_csel_slots.as<int32_t>(),
+                    cx->w_csel_dists.as<float>(), s);
         KCHECK("rank-topk");
     }
 
     /* 3. D2H the probe selection; build list-scan jobs on the host
      * (findCentroids -> scanEntries shape, search.go:917-1005). */
     std::vector<int32_t> h_probe((size_t)nq * probe);
-    HIP_CHECK(hipMemcpyAsync(h_probe.data(), ix->w_csel_slots.ptr,
+    HIP_CHECK(hipMemcpyAsync(h_probe.data(), cx->w_csel_slots.ptr,
                              h_probe.size() * 4, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     HIP_CHECK(hipGetLastError());
@@ -516,19 +562,19 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const uint32_t* d_filter = nullptr;
     if (filter_words) {
         const size_t fw = (size_t)((ix->count + 31) / 32);
-        ix->w_filter.ensure(fw * 4);
-        HIP_CHECK(hipMemcpyAsync(ix->w_filter.ptr, filter_words, fw * 4,
+        cx->w_filter.ensure(fw * 4);
+        HIP_CHECK(hipMemcpyAsync(cx->w_filter.ptr, filter_words, fw * 4,
                                  hipMemcpyHostToDevice, s));
-        d_filter = ix->w_filter.as<uint32_t>();
+        d_filter = cx->w_filter.as<uint32_t>();
     }
-    ix->w_cand.ensure(std::max<int64_t>(1, total_cand) * 4);
+    cx->w_cand.ensure(std::max<int64_t>(1, total_cand) * 4);
     {
         const size_t b64 = sizeof(int64_t), b32 = sizeof(int32_t);
         auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
         size_t need = 2 * al8((size_t)njobs * b64) + 5 * al8((size_t)njobs * b32) +
                       al8((size_t)nqslots * b32) + al8((size_t)nqslots * b64) + 64;
-        ix->w_qslots.ensure(need);
-        uint8_t* p = ix->w_qslots.as<uint8_t>();
+        cx->w_qslots.ensure(need);
+        uint8_t* p = cx->w_qslots.as<uint8_t>();
         auto up = [&](const void* src, size_t bytes) {
             void* r = p;
             if (bytes)
@@ -547,91 +593,127 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         jb.qslot_outbase = (int64_t*)up(qslot_outbase.data(), nqslots * b64);
         jb.slot_base = (int64_t*)up(jslot_base.data(), njobs * b64);
         jb.njobs = njobs;
-        HIP_CHECK(hipEventRecord(ix->perf.ev[2], s));
+        HIP_CHECK(hipEventRecord(cx->ev[2], s));
         if (narrow) {
             /* quantize the padded f32 queries with the entry map (the
              * search_quantize contract: query narrowed the same way as the
              * entries — quantizer.go ApplyInt8/entry SQL) */
-            ix->w_queries_q.ensure((size_t)nq * dpad);
+            cx->w_queries_q.ensure((size_t)nq * dpad);
             launch_quantize_rows(uns, d_q, (int64_t)nq, dpad, (int)ix->dim,
                                  dpad, (float)ix->quant_mul,
                                  (float)ix->quant_add,
-                                 ix->w_queries_q.as<uint8_t>(), s);
+                                 cx->w_queries_q.as<uint8_t>(), s);
             KCHECK("query-quantize");
             const int32_t* d_qni = nullptr;
             if (ix->kmetric == KM_COS) {
-                ix->w_qnorms_i.ensure(nq * 4);
-                launch_qnorms_i8(uns, ix->w_queries_q.as<uint8_t>(), (int)nq,
-                                 dpad, ix->w_qnorms_i.as<int32_t>(), s);
-                d_qni = ix->w_qnorms_i.as<int32_t>();
+                cx->w_qnorms_i.ensure(nq * 4);
+                launch_qnorms_i8(uns, cx->w_queries_q.as<uint8_t>(), (int)nq,
+                                 dpad, cx->w_qnorms_i.as<int32_t>(), s);
+                d_qni = cx->w_qnorms_i.as<int32_t>();
             }
             launch_scan_i8(ix->kmetric, uns, ix->d_packed_q,
-                           ix->w_queries_q.as<uint8_t>(), d_qni, dpad, jb,
-                           ix->w_cand.as<float>(), s, d_filter);
+                           cx->w_queries_q.as<uint8_t>(), d_qni, dpad, jb,
+                           cx->w_cand.as<float>(), s, d_filter);
         } else {
             launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
-                        ix->w_cand.as<float>(), s, d_filter);
+                        cx->w_cand.as<float>(), s, d_filter);
         }
         KCHECK("list-scan");
-        HIP_CHECK(hipEventRecord(ix->perf.ev[3], s));
+        HIP_CHECK(hipEventRecord(cx->ev[3], s));
     }
 
     /* 5. per-query top-limit + id gather + transform */
-    ix->w_qoffs.ensure((nq + 1) * 8);
-    HIP_CHECK(hipMemcpyAsync(ix->w_qoffs.ptr, qoffs.data(), (nq + 1) * 8,
+    cx->w_qoffs.ensure((nq + 1) * 8);
+    HIP_CHECK(hipMemcpyAsync(cx->w_qoffs.ptr, qoffs.data(), (nq + 1) * 8,
                              hipMemcpyHostToDevice, s));
-    ix->w_sel_slots.ensure((size_t)nq * limit * 4);
-    ix->w_sel_dists.ensure((size_t)nq * limit * 4);
-    HIP_CHECK(hipEventRecord(ix->perf.ev[4], s));
-    launch_topk(ix->w_cand.as<float>(), ix->w_qoffs.as<int64_t>(), 0, (int)nq,
-                (int)limit, ix->w_sel_slots.as<int32_t>(),
-                ix->w_sel_dists.as<float>(), s);
+    cx->w_sel_slots.ensure((size_t)nq * limit * 4);
+    cx->w_sel_dists.ensure((size_t)nq * limit * 4);
+    HIP_CHECK(hipEventRecord(cx->ev[4], s));
+    launch_topk(cx->w_cand.as<float>(), cx->w_qoffs.as<int64_t>(), 0, (int)nq,
+                (int)limit, cx->w_sel_slots.as<int32_t>(),
+                cx->w_sel_dists.as<float>(), s);
     KCHECK("cand-topk");
-    HIP_CHECK(hipEventRecord(ix->perf.ev[5], s));
-    ix->w_probe_lists.ensure((size_t)nq * probe * 4);
-    ix->w_probe_offs.ensure(probe_offs.size() * 8);
-    HIP_CHECK(hipMemcpyAsync(ix->w_probe_lists.ptr, h_probe.data(),
+    HIP_CHECK(hipEventRecord(cx->ev[5], s));
+    cx->w_probe_lists.ensure((size_t)nq * probe * 4);
+    cx->w_probe_offs.ensure(probe_offs.size() * 8);
+    HIP_CHECK(hipMemcpyAsync(cx->w_probe_lists.ptr, h_probe.data(),
                              h_probe.size() * 4, hipMemcpyHostToDevice, s));
-    HIP_CHECK(hipMemcpyAsync(ix->w_probe_offs.ptr, probe_offs.data(),
+    HIP_CHECK(hipMemcpyAsync(cx->w_probe_offs.ptr, probe_offs.data(),
                              probe_offs.size() * 8, hipMemcpyHostToDevice, s));
-    ix->w_out_ids.ensure((size_t)nq * limit * 8);
-    ix->w_out_dists.ensure((size_t)nq * limit * 4);
-    launch_gather(ix->w_sel_slots.as<int32_t>(), ix->w_sel_dists.as<float>(),
-                  ix->w_probe_lists.as<int32_t>(),
-                  ix->w_probe_offs.as<int64_t>(), ix->d_list_slot_base,
+    cx->w_out_ids.ensure((size_t)nq * limit * 8);
+    cx->w_out_dists.ensure((size_t)nq * limit * 4);
+    launch_gather(cx->w_sel_slots.as<int32_t>(), cx->w_sel_dists.as<float>(),
+                  cx->w_probe_lists.as<int32_t>(),
+                  cx->w_probe_offs.as<int64_t>(), ix->d_list_slot_base,
                   ix->d_id_by_slot, (int)probe, (int)nq, (int)limit, do_sqrt,
-                  inv_mul2, ix->w_out_ids.as<int64_t>(),
-                  ix->w_out_dists.as<float>(), s);
+                  inv_mul2, cx->w_out_ids.as<int64_t>(),
+                  cx->w_out_dists.as<float>(), s);
     KCHECK("gather");
-    HIP_CHECK(hipMemcpyAsync(res->neighbors.data(), ix->w_out_ids.ptr,
+    /* D2H through the ctx's pinned bounce (pageable-async degrades to a
+     * host-blocking copy on ROCm, which would serialize the pipeline) */
+    cx->ensure_hout((size_t)nq * limit * 12);
+    int64_t* ho_ids = (int64_t*)cx->h_out;
+    float* ho_dists = (float*)(ho_ids + (size_t)nq * limit);
+    HIP_CHECK(hipMemcpyAsync(ho_ids, cx->w_out_ids.ptr,
                              (size_t)nq * limit * 8, hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipMemcpyAsync(res->distances.data(), ix->w_out_dists.ptr,
+    HIP_CHECK(hipMemcpyAsync(ho_dists, cx->w_out_dists.ptr,
                              (size_t)nq * limit * 4, hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(hipGetLastError());
 
-    /* perf accounting */
-    float r_ms = 0, sc_ms = 0, sel_ms = 0;
-    HIP_CHECK(hipEventElapsedTime(&r_ms, ix->perf.ev[0], ix->perf.ev[1]));
+    /* host-known perf counters (kernel window times land at collect) */
     ix->perf.c.rank_launches++;
-    ix->perf.c.rank_ms += r_ms;
     ix->perf.c.rank_flops += 2ull * nq * ix->nlist * ix->dim;
-    HIP_CHECK(hipEventElapsedTime(&sc_ms, ix->perf.ev[2], ix->perf.ev[3]));
     uint64_t unique_rows = 0;
     for (uint32_t l = 0; l < ix->nlist; ++l)
         if (lcount[l]) unique_rows += (uint64_t)ix->list_rows[l];
     ix->perf.c.scan_launches++;
-    ix->perf.c.scan_ms += sc_ms;
     ix->perf.c.scan_bytes += unique_rows * ix->dim * 4;
     ix->perf.c.scan_rows += (uint64_t)total_cand;
-    HIP_CHECK(hipEventElapsedTime(&sel_ms, ix->perf.ev[4], ix->perf.ev[5]));
-    ix->perf.c.select_ms += sel_ms;
-    /* everything the three event windows miss: H2D/D2H copies, the probe
-     * sync, host job build, gather — per-call wall minus kernel windows */
-    const double wall = std::chrono::duration<double, std::milli>(
-        std::chrono::steady_clock::now() - t_all0).count();
-    ix->perf.c.other_ms += wall - r_ms - sc_ms - sel_ms;
+
+    res->owner = cx;
+    res->collected = false;
+    HIP_CHECK(hipEventCreate(&res->done_ev));
+    HIP_CHECK(hipEventRecord(res->done_ev, s));
+    cx->pending = res.get();
+    /* submit-side wall: under pipelining this is the NON-overlapped host
+     * cost (H2D, probe sync, job build); kernel windows come from cx->ev */
+    if (sync_end) {
+        const double kms = collect_result(res.get());
+        const double wall = std::chrono::duration<double, std::milli>(
+            std::chrono::steady_clock::now() - t_all0).count();
+        ix->perf.c.other_ms += wall - kms;
+    } else {
+        const double wall = std::chrono::duration<double, std::milli>(
+            std::chrono::steady_clock::now() - t_all0).count();
+        ix->perf.c.other_ms += wall;
+    }
     return res;
+}
+
+/* Wait for a submitted search, copy the pinned payload into the result
+ * vectors and fold the kernel event windows into the perf counters.
+ * Serialized by the submit/wait discipline (the submit path collects a
+ * ctx's pending result before reusing it). */
+double collect_result(MoannResult* res) {
+    if (res->collected) return 0.0;
+    auto cx = (SearchCtx*)res->owner;
+    auto ix = (IvfIndex*)cx->parent;
+    HIP_CHECK(hipEventSynchronize(res->done_ev));
+    HIP_CHECK(hipGetLastError());
+    const size_t n = res->neighbors.size();
+    const int64_t* ho_ids = (const int64_t*)cx->h_out;
+    const float* ho_dists = (const float*)(ho_ids + n);
+    memcpy(res->neighbors.data(), ho_ids, n * 8);
+    memcpy(res->distances.data(), ho_dists, n * 4);
+    float r_ms = 0, sc_ms = 0, sel_ms = 0;
+    (void)hipEventElapsedTime(&r_ms, cx->ev[0], cx->ev[1]);
+    (void)hipEventElapsedTime(&sc_ms, cx->ev[2], cx->ev[3]);
+    (void)hipEventElapsedTime(&sel_ms, cx->ev[4], cx->ev[5]);
+    ix->perf.c.rank_ms += r_ms;
+    ix->perf.c.scan_ms += sc_ms;
+    ix->perf.c.select_ms += sel_ms;
+    res->collected = true;
+    if (cx->pending == res) cx->pending = nullptr;
+    return (double)r_ms + sc_ms + sel_ms;
 }
 
 void batcher_loop(IvfIndex* ix) {
@@ -996,6 +1078,42 @@ gpu_ivf_flat_search_res_t moann_ivf_flat_search_device(gpu_ivf_flat_c h,
     return out;
 }
 
+/* Pipelined submit: phase 1 (query H2D, centroid rank, probe top-k, probe
+ * D2H) and the host job build run NOW on one of two per-index pipeline
+ * contexts; the list scan / select / gather / result D2H are ENQUEUED on
+ * that context's stream and the call returns without waiting. Collect with
+ * moann_ivf_flat_result_wait (or any get_*); at most two submits in
+ * flight, collected FIFO — a third submit collects the oldest itself. */
+gpu_ivf_flat_search_res_t moann_ivf_flat_search_submit(
+    gpu_ivf_flat_c h, const void* queries, int queries_on_device,
+    uint64_t num_queries, uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t sp, void* errmsg) {
+    gpu_ivf_flat_search_res_t out {nullptr};
+    try {
+        auto ix = IX(h);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        out.result_ptr = run_search(ix, (const float*)queries,
+                                    queries_on_device != 0, num_queries,
+                                    limit, sp.n_probes, nullptr,
+                                    /*sync_end=*/false)
+                             .release();
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_search_submit", e.what());
+    }
+    return out;
+}
+
+int moann_ivf_flat_result_wait(gpu_ivf_flat_result_c r, void* errmsg) {
+    try {
+        collect_result((Result*)r);
+        return 0;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_result_wait", e.what());
+        return -1;
+    }
+}
+
 uint64_t gpu_ivf_flat_search_async(gpu_ivf_flat_c h, const void* queries_data,
                                    uint64_t num_queries,
                                    uint32_t query_dimension, uint32_t limit,
@@ -1074,6 +1192,7 @@ gpu_ivf_flat_search_res_t gpu_ivf_flat_search_wait(gpu_ivf_flat_c h,
 void gpu_ivf_flat_get_neighbors(gpu_ivf_flat_result_c r, uint64_t n,
                                 int64_t* neighbors) {
     auto res = (Result*)r;
+    if (!res->collected) collect_result(res);
     const uint64_t m = std::min<uint64_t>(n, res->neighbors.size());
     memcpy(neighbors, res->neighbors.data(), m * 8);
 }
@@ -1081,11 +1200,23 @@ void gpu_ivf_flat_get_neighbors(gpu_ivf_flat_result_c r, uint64_t n,
 void gpu_ivf_flat_get_distances(gpu_ivf_flat_result_c r, uint64_t n,
                                 float* distances) {
     auto res = (Result*)r;
+    if (!res->collected) collect_result(res);
     const uint64_t m = std::min<uint64_t>(n, res->distances.size());
     memcpy(distances, res->distances.data(), m * 4);
 }
 
-void gpu_ivf_flat_free_result(gpu_ivf_flat_result_c r) { delete (Result*)r; }
+void gpu_ivf_flat_free_result(gpu_ivf_flat_result_c r) {
+    auto res = (Result*)r;
+    if (res && res->done_ev) {
+        if (!res->collected) {
+            (void)hipEventSynchronize(res->done_ev);
+            auto cx = (SearchCtx*)res->owner;
+            if (cx && cx->pending == res) cx->pending = nullptr;
+        }
+        (void)hipEventDestroy(res->done_ev);
+    }
+    delete res;
+}
 
 uint64_t gpu_ivf_flat_cap(gpu_ivf_flat_c h) { return IX(h)->cap; }
 uint64_t gpu_ivf_flat_len(gpu_ivf_flat_c h) { return IX(h)->count; }

@@ -108,6 +108,12 @@ def lib() -> ct.CDLL:
     L.moann_ivf_flat_search_filtered.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
         SearchParams, ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    L.moann_ivf_flat_search_submit.restype = SearchRes
+    L.moann_ivf_flat_search_submit.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_int, ct.c_uint64, ct.c_uint32,
+        ct.c_uint32, SearchParams, ct.c_void_p]
+    L.moann_ivf_flat_result_wait.restype = ct.c_int
+    L.moann_ivf_flat_result_wait.argtypes = [ct.c_void_p, ct.c_void_p]
     L.moann_ivf_flat_slots_of_ids.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
     L.moann_brute_force_search.argtypes = [
@@ -313,6 +319,34 @@ class IvfFlatIndex:
             SearchParams(nprobe), err.ref)
         err.check("search_device")
         return self._read_result(res, nq, k, "search_device")
+
+    def search_submit(self, queries_tensor, k: int, nprobe: int):
+        """Pipelined submit (moann_ivf_flat_search_submit): centroid rank +
+        host job build run now, the list scan is enqueued; a second submit
+        overlaps with this one's scan. At most 2 in flight, collect FIFO
+        with search_collect."""
+        nq, qdim = queries_tensor.shape
+        err = _Err()
+        res = lib().moann_ivf_flat_search_submit(
+            self._h, ct.c_void_p(queries_tensor.data_ptr()), 1, nq, qdim, k,
+            SearchParams(nprobe), err.ref)
+        err.check("search_submit")
+        if not res.result_ptr:
+            raise MoannError("search_submit returned NULL result")
+        return (res.result_ptr, nq, k)
+
+    def search_collect(self, ticket):
+        """Wait for a search_submit ticket and return (ids, dists)."""
+        ptr, nq, k = ticket
+        err = _Err()
+        rc = lib().moann_ivf_flat_result_wait(ptr, err.ref)
+        err.check("result_wait")
+        if rc != 0:
+            raise MoannError("result_wait failed")
+
+        class _R:
+            result_ptr = ptr
+        return self._read_result(_R, nq, k, "search_collect")
 
     def set_batch_window(self, window_us: int):
         """ivf_flat_c.h:91 — fuse concurrent search_async calls arriving
