@@ -518,11 +518,15 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
      * of {6,8,10,12,14,18,36} end-to-end — small chunks also win on tail
      * balance);
      * MOANN_CHUNKG overrides for tuning. */
-    static const int64_t CHUNK_G = [] {
+    static const int64_t CHUNK_G_BASE = [] {
         const char* e = getenv("MOANN_CHUNKG");
         const int v = e ? atoi(e) : 0;
         return (int64_t)(v > 0 ? v : 8);
     }();
+    /* the chunk target is BYTES (~1.5 MB of list data); byte storage packs
+     * 4x the rows per byte, so scale the group count accordingly */
+    const int64_t CHUNK_G = ix->qtype != Quantization_F32 ? CHUNK_G_BASE * 4
+                                                          : CHUNK_G_BASE;
 
     std::vector<int64_t> databaseg, qslot_outbase, jslot_base;
     std::vector<int32_t> gstart, gcount, rows, jnq, qbase, qslot_query;
