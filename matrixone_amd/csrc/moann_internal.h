@@ -100,9 +100,14 @@ bool launch_rank_gemm(int metric, const float* queries, const float* cents,
 
 /* Quantized (int8/uint8) list scan + helpers. */
 void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
-                    const uint8_t* queries_q, const int32_t* qnorms, int dpad,
+                    const uint8_t* queries_q, const int32_t* qnorms,
+                    const int32_t* rownorms, int dpad,
                     const ScanJobs& jb, float* dists_out, hipStream_t stream,
                     const uint32_t* filter_bitset = nullptr);
+/* per-(group,lane) sum-of-squares over the packed byte rows (build-time,
+ * feeds the dot-form scan's rn + qn - 2*dot L2) */
+void launch_rownorms_i8(bool uns, const uint8_t* packed, int64_t ngroups,
+                        int dpad, int32_t* out, hipStream_t stream);
 void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
                           int in_stride, int dim, int dpad, float fmul,
                           float fadd, uint8_t* out, hipStream_t stream);

@@ -848,34 +848,271 @@ __global__ __launch_bounds__(256) void scan_i8_kernel(
     }
 }
 
+/* Dot-form byte scan (dpad % 16 == 0): one dwordx4 load = 16 dims of the
+ * row per lane (1 KiB per wave step, same stream shape as the f32 kernel
+ * — the uchar4 kernel's 256 B loads were MLP-starved at ~1 TB/s), and the
+ * arithmetic uses the CDNA dot/sad byte ops:
+ *   L2sq:  |x-q|^2 = rn + qn - 2*dot(x,q)  (exact in int32: bounds
+ *          255^2*dpad <= 5.1e7 at dpad<=768; v_dot4_{i32_i8,u32_u8})
+ *   IP/cos: dot(x,q) directly (cos norms precomputed, same ints)
+ *   L1:    v_sad_u8 (signed values biased by XOR 0x80 per byte — two's
+ *          complement int8 + 128 == x ^ 0x80, |x-q| unchanged)
+ * Row norms rn come from launch_rownorms_i8 at build time; query norms qn
+ * from launch_qnorms_i8. Results are bit-identical to scan_i8_kernel
+ * (exact integer arithmetic both ways); that kernel stays the path for
+ * dpad % 16 != 0. */
+template <int METRIC, bool UNSIGNED>
+__global__ __launch_bounds__(256) void scan_i8_dot_kernel(
+    const uint8_t* __restrict__ packed, const uint8_t* __restrict__ queries_q,
+    const int32_t* __restrict__ qnorms_i, /* [nq] sum sq (l2/cos) */
+    const int32_t* __restrict__ rownorms, /* [ngroups*64] sum sq (l2/cos) */
+    int dpad,
+    const int64_t* __restrict__ j_databaseg,
+    const int32_t* __restrict__ j_gstart, const int32_t* __restrict__ j_gcount,
+    const int32_t* __restrict__ j_rows, const int32_t* __restrict__ j_nq,
+    const int32_t* __restrict__ j_qbase,
+    const int32_t* __restrict__ qslot_query,
+    const int64_t* __restrict__ qslot_outbase,
+    const int64_t* __restrict__ j_slot_base,
+    const uint32_t* __restrict__ filter_bitset,
+    float* __restrict__ dists_out) {
+    constexpr int QT = 8;
+    constexpr bool NEED_N = METRIC == KM_L2SQ || METRIC == KM_COS;
+    extern __shared__ float ldsf[];
+    uint8_t* ldsq = (uint8_t*)ldsf;            /* [QT][dpad] bytes */
+    int32_t* ldsn = (int32_t*)(ldsq + QT * dpad);
+
+    int j;
+    {
+        const int nwg = gridDim.x, bid = blockIdx.x;
+        const int q8 = nwg >> 3, r8 = nwg & 7;
+        const int xcd = bid & 7, idx = bid >> 3;
+        j = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+        if (j >= nwg) j = bid;
+    }
+    const int nq = j_nq[j];
+    const int qbase = j_qbase[j];
+    for (int t = 0; t < QT; ++t) {
+        if (t < nq) {
+            const int q = qslot_query[qbase + t];
+            const uint8_t* src = queries_q + (int64_t)q * dpad;
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = src[e];
+            if (NEED_N && threadIdx.x == 0) ldsn[t] = qnorms_i[q];
+        } else {
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = 0;
+            if (NEED_N && threadIdx.x == 0) ldsn[t] = 0;
+        }
+    }
+    __syncthreads();
+
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int gstart = j_gstart[j], gcount = j_gcount[j];
+    const int rows = j_rows[j];
+    const int64_t baseg = j_databaseg[j];
+    const int d16 = dpad >> 4;
+
+    for (int gp = gstart + 2 * wave; gp < gstart + gcount; gp += 8) {
+        const int g0 = gp;
+        const bool has1 = (gp + 1) < (gstart + gcount);
+        const int g1 = has1 ? gp + 1 : gp;
+        const uint4* __restrict__ d0 =
+            (const uint4*)(packed + (baseg + g0) * (int64_t)64 * dpad) + lane;
+        const uint4* __restrict__ d1 =
+            (const uint4*)(packed + (baseg + g1) * (int64_t)64 * dpad) + lane;
+
+        int32_t acc0[QT], acc1[QT];
+#pragma unroll
+        for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0;
+
+#pragma unroll 2
+        for (int j16 = 0; j16 < d16; ++j16) {
+            uint4 x0 = d0[j16 * 64];
+            uint4 x1 = d1[j16 * 64];
+            if (METRIC == KM_L1 && !UNSIGNED) {
+                /* bias int8 -> u8 so v_sad_u8 applies */
+                x0.x ^= 0x80808080u; x0.y ^= 0x80808080u;
+                x0.z ^= 0x80808080u; x0.w ^= 0x80808080u;
+                x1.x ^= 0x80808080u; x1.y ^= 0x80808080u;
+                x1.z ^= 0x80808080u; x1.w ^= 0x80808080u;
+            }
+#pragma unroll
+            for (int t = 0; t < QT; ++t) {
+                uint4 qv = *(const uint4*)(ldsq + t * dpad + j16 * 16);
+                if (METRIC == KM_L1) {
+                    if (!UNSIGNED) {
+                        qv.x ^= 0x80808080u; qv.y ^= 0x80808080u;
+                        qv.z ^= 0x80808080u; qv.w ^= 0x80808080u;
+                    }
+                    acc0[t] = (int32_t)__builtin_amdgcn_sad_u8(x0.x, qv.x, (uint32_t)acc0[t]);
+                    acc0[t] = (int32_t)__builtin_amdgcn_sad_u8(x0.y, qv.y, (uint32_t)acc0[t]);
+                    acc0[t] = (int32_t)__builtin_amdgcn_sad_u8(x0.z, qv.z, (uint32_t)acc0[t]);
+                    acc0[t] = (int32_t)__builtin_amdgcn_sad_u8(x0.w, qv.w, (uint32_t)acc0[t]);
+                    acc1[t] = (int32_t)__builtin_amdgcn_sad_u8(x1.x, qv.x, (uint32_t)acc1[t]);
+                    acc1[t] = (int32_t)__builtin_amdgcn_sad_u8(x1.y, qv.y, (uint32_t)acc1[t]);
+                    acc1[t] = (int32_t)__builtin_amdgcn_sad_u8(x1.z, qv.z, (uint32_t)acc1[t]);
+                    acc1[t] = (int32_t)__builtin_amdgcn_sad_u8(x1.w, qv.w, (uint32_t)acc1[t]);
+                } else if (UNSIGNED) {
+                    acc0[t] = (int32_t)__builtin_amdgcn_udot4(x0.x, qv.x, (uint32_t)acc0[t], false);
+                    acc0[t] = (int32_t)__builtin_amdgcn_udot4(x0.y, qv.y, (uint32_t)acc0[t], false);
+                    acc0[t] = (int32_t)__builtin_amdgcn_udot4(x0.z, qv.z, (uint32_t)acc0[t], false);
+                    acc0[t] = (int32_t)__builtin_amdgcn_udot4(x0.w, qv.w, (uint32_t)acc0[t], false);
+                    acc1[t] = (int32_t)__builtin_amdgcn_udot4(x1.x, qv.x, (uint32_t)acc1[t], false);
+                    acc1[t] = (int32_t)__builtin_amdgcn_udot4(x1.y, qv.y, (uint32_t)acc1[t], false);
+                    acc1[t] = (int32_t)__builtin_amdgcn_udot4(x1.z, qv.z, (uint32_t)acc1[t], false);
+                    acc1[t] = (int32_t)__builtin_amdgcn_udot4(x1.w, qv.w, (uint32_t)acc1[t], false);
+                } else {
+                    acc0[t] = __builtin_amdgcn_sdot4((int32_t)x0.x, (int32_t)qv.x, acc0[t], false);
+                    acc0[t] = __builtin_amdgcn_sdot4((int32_t)x0.y, (int32_t)qv.y, acc0[t], false);
+                    acc0[t] = __builtin_amdgcn_sdot4((int32_t)x0.z, (int32_t)qv.z, acc0[t], false);
+                    acc0[t] = __builtin_amdgcn_sdot4((int32_t)x0.w, (int32_t)qv.w, acc0[t], false);
+                    acc1[t] = __builtin_amdgcn_sdot4((int32_t)x1.x, (int32_t)qv.x, acc1[t], false);
+                    acc1[t] = __builtin_amdgcn_sdot4((int32_t)x1.y, (int32_t)qv.y, acc1[t], false);
+                    acc1[t] = __builtin_amdgcn_sdot4((int32_t)x1.z, (int32_t)qv.z, acc1[t], false);
+                    acc1[t] = __builtin_amdgcn_sdot4((int32_t)x1.w, (int32_t)qv.w, acc1[t], false);
+                }
+            }
+        }
+
+        const int row0 = g0 * 64 + lane;
+        const int row1 = g1 * 64 + lane;
+        const int32_t rn0 = NEED_N ? rownorms[(baseg + g0) * 64 + lane] : 0;
+        const int32_t rn1 = NEED_N ? rownorms[(baseg + g1) * 64 + lane] : 0;
+        bool pass0 = true, pass1 = true;
+        if (filter_bitset) {
+            const int64_t sb = j_slot_base[j];
+            if (row0 < rows) {
+                const int64_t gs = sb + row0;
+                pass0 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+            if (has1 && row1 < rows) {
+                const int64_t gs = sb + row1;
+                pass1 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+        }
+#pragma unroll
+        for (int t = 0; t < QT; ++t) {
+            if (t >= nq) break;
+            const int64_t ob = qslot_outbase[qbase + t];
+            if (row0 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = (float)(-(int64_t)acc0[t]);
+                else if (METRIC == KM_COS) {
+                    const double denom =
+                        sqrt((double)rn0) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc0[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else if (METRIC == KM_L2SQ)
+                    dist = (float)(rn0 + ldsn[t] - 2 * acc0[t]);
+                else dist = (float)acc0[t]; /* L1 */
+                dists_out[ob + row0] = pass0 ? dist : FLT_MAX;
+            }
+            if (has1 && row1 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = (float)(-(int64_t)acc1[t]);
+                else if (METRIC == KM_COS) {
+                    const double denom =
+                        sqrt((double)rn1) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc1[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else if (METRIC == KM_L2SQ)
+                    dist = (float)(rn1 + ldsn[t] - 2 * acc1[t]);
+                else dist = (float)acc1[t];
+                dists_out[ob + row1] = pass1 ? dist : FLT_MAX;
+            }
+        }
+    }
+}
+
+/* per-(group,lane) byte-row sum of squares, same wave layout as the scan */
+template <bool UNSIGNED>
+__global__ __launch_bounds__(256) void rownorms_i8_kernel(
+    const uint8_t* __restrict__ packed, int64_t ngroups, int dpad,
+    int32_t* __restrict__ out) {
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int64_t g = (int64_t)blockIdx.x * 4 + wave;
+    if (g >= ngroups) return;
+    const uchar4* __restrict__ d =
+        (const uchar4*)(packed + g * (int64_t)64 * dpad) + lane;
+    int32_t acc = 0;
+    const int d4 = dpad >> 2;
+    for (int q4 = 0; q4 < d4; ++q4) {
+        const uchar4 v = d[q4 * 64];
+        const int a = UNSIGNED ? (int)v.x : (int)(int8_t)v.x;
+        const int b = UNSIGNED ? (int)v.y : (int)(int8_t)v.y;
+        const int c = UNSIGNED ? (int)v.z : (int)(int8_t)v.z;
+        const int e = UNSIGNED ? (int)v.w : (int)(int8_t)v.w;
+        acc += a * a + b * b + c * c + e * e;
+    }
+    out[g * 64 + lane] = acc;
+}
+
+void launch_rownorms_i8(bool uns, const uint8_t* packed, int64_t ngroups,
+                        int dpad, int32_t* out, hipStream_t stream) {
+    const dim3 grid((uint32_t)((ngroups + 3) / 4)), block(256);
+    if (uns)
+        hipLaunchKernelGGL((rownorms_i8_kernel<true>), grid, block, 0, stream,
+                           packed, ngroups, dpad, out);
+    else
+        hipLaunchKernelGGL((rownorms_i8_kernel<false>), grid, block, 0, stream,
+                           packed, ngroups, dpad, out);
+}
+
 template <int METRIC>
 static void launch_scan_i8_m(bool uns, const uint8_t* packed,
                              const uint8_t* queries_q, const int32_t* qnorms,
-                             int dpad, const ScanJobs& jb, float* dists_out,
+                             const int32_t* rownorms, int dpad,
+                             const ScanJobs& jb, float* dists_out,
                              hipStream_t stream,
                              const uint32_t* filter_bitset) {
     const dim3 grid(jb.njobs), block(256);
     const size_t shmem = 8 * dpad + 8 * 4 + 16;
-#define I8ARGS packed, queries_q, qnorms, dpad, jb.databaseg, jb.gstart,     jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase,     jb.slot_base, filter_bitset, dists_out
-    if (uns)
+    /* dot-form kernel needs 16-dim-aligned rows and the precomputed row
+     * norms; falls back to the uchar4 kernel otherwise (odd dims) */
+    const bool dot = (dpad & 15) == 0 && rownorms != nullptr;
+#define I8ARGS packed, queries_q, qnorms, dpad, jb.databaseg, jb.gstart, \
+    jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase, \
+    jb.slot_base, filter_bitset, dists_out
+#define I8DARGS packed, queries_q, qnorms, rownorms, dpad, jb.databaseg, \
+    jb.gstart, jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, \
+    jb.qslot_outbase, jb.slot_base, filter_bitset, dists_out
+    if (dot) {
+        if (uns)
+            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, true>), grid,
+                               block, shmem, stream, I8DARGS);
+        else
+            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, false>), grid,
+                               block, shmem, stream, I8DARGS);
+    } else if (uns)
         hipLaunchKernelGGL((scan_i8_kernel<METRIC, true>), grid, block, shmem,
                            stream, I8ARGS);
     else
         hipLaunchKernelGGL((scan_i8_kernel<METRIC, false>), grid, block,
                            shmem, stream, I8ARGS);
 #undef I8ARGS
+#undef I8DARGS
 }
 
 void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
-                    const uint8_t* queries_q, const int32_t* qnorms, int dpad,
+                    const uint8_t* queries_q, const int32_t* qnorms,
+                    const int32_t* rownorms, int dpad,
                     const ScanJobs& jb, float* dists_out, hipStream_t stream,
                     const uint32_t* filter_bitset) {
     if (!jb.njobs) return;
     switch (metric) {
-    case KM_L2SQ: launch_scan_i8_m<KM_L2SQ>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
-    case KM_IP:   launch_scan_i8_m<KM_IP>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
-    case KM_COS:  launch_scan_i8_m<KM_COS>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
-    default:      launch_scan_i8_m<KM_L1>(uns, packed, queries_q, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_L2SQ: launch_scan_i8_m<KM_L2SQ>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_IP:   launch_scan_i8_m<KM_IP>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_COS:  launch_scan_i8_m<KM_COS>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    default:      launch_scan_i8_m<KM_L1>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
     }
 }
 

@@ -164,6 +164,7 @@ struct IvfIndex {
     double quant_mul = 1.0, quant_add = 0.0; /* identity until trained */
     double quant_min = 0.0, quant_max = 0.0;
     uint8_t* d_packed_q = nullptr;  /* quantized interleaved entries */
+    int32_t* d_rownorm_q = nullptr; /* [total_groups*64] byte-row |x|^2 */
     bool built = false, started = false;
 
     hipStream_t stream = nullptr;
@@ -232,6 +233,7 @@ struct IvfIndex {
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_staging, (void*)d_packed, (void*)d_cent_packed,
                        (void*)d_cent_rows, (void*)d_cnorms, (void*)d_packed_q,
+                       (void*)d_rownorm_q,
                        (void*)d_id_by_slot, (void*)d_list_slot_base})
             if (p) (void)hipFree(p);
         if (stream) (void)hipStreamDestroy(stream);
@@ -609,14 +611,16 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                                  cx->w_queries_q.as<uint8_t>(), s);
             KCHECK("query-quantize");
             const int32_t* d_qni = nullptr;
-            if (ix->kmetric == KM_COS) {
+            if (ix->kmetric == KM_COS || ix->kmetric == KM_L2SQ) {
+                /* cos denominators; also qn of the dot-form L2 */
                 cx->w_qnorms_i.ensure(nq * 4);
                 launch_qnorms_i8(uns, cx->w_queries_q.as<uint8_t>(), (int)nq,
                                  dpad, cx->w_qnorms_i.as<int32_t>(), s);
                 d_qni = cx->w_qnorms_i.as<int32_t>();
             }
             launch_scan_i8(ix->kmetric, uns, ix->d_packed_q,
-                           cx->w_queries_q.as<uint8_t>(), d_qni, dpad, jb,
+                           cx->w_queries_q.as<uint8_t>(), d_qni,
+                           ix->d_rownorm_q, dpad, jb,
                            cx->w_cand.as<float>(), s, d_filter);
         } else {
             launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
@@ -947,12 +951,25 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
                 HIP_CHECK(hipMemcpyAsync(d_sr2.ptr, slot_rows.data(),
                                          slot_rows.size() * 8,
                                          hipMemcpyHostToDevice, ix->stream));
+            if (ix->d_rownorm_q) {
+                (void)hipFree(ix->d_rownorm_q);
+                ix->d_rownorm_q = nullptr;
+            }
             HIP_CHECK(hipMalloc(&ix->d_packed_q,
                                 (size_t)ix->total_groups * 64 * ix->dpad));
             launch_pq_pack(d_qrows.as<uint8_t>(), (int)ix->dpad,
                            d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
                            d_sr2.as<int64_t>(), ix->total_groups,
                            ix->d_packed_q, ix->stream);
+            if ((ix->dpad & 15) == 0) {
+                /* row norms for the dot-form byte scan (rn + qn - 2*dot) */
+                HIP_CHECK(hipMalloc(&ix->d_rownorm_q,
+                                    (size_t)ix->total_groups * 64 * 4));
+                launch_rownorms_i8(ix->qtype == Quantization_UINT8,
+                                   ix->d_packed_q, ix->total_groups,
+                                   (int)ix->dpad, ix->d_rownorm_q,
+                                   ix->stream);
+            }
             HIP_CHECK(hipStreamSynchronize(ix->stream));
             KCHECK("quantize-pack");
             HIP_CHECK(hipFree(ix->d_packed));
