@@ -106,6 +106,11 @@ void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
                     const uint32_t* filter_bitset = nullptr);
 /* per-(group,lane) sum-of-squares over the packed byte rows (build-time,
  * feeds the dot-form scan's rn + qn - 2*dot L2) */
+void launch_bytes_pack16(const uint8_t* rows_q, int dpad,
+                         const int64_t* group_rowbase,
+                         const int32_t* group_valid,
+                         const int64_t* slot_rows, int64_t ngroups,
+                         uint8_t* packed, hipStream_t stream);
 void launch_rownorms_i8(bool uns, const uint8_t* packed, int64_t ngroups,
                         int dpad, int32_t* out, hipStream_t stream);
 void launch_quantize_rows(bool uns, const float* in, int64_t nrows,

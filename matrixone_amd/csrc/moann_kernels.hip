@@ -1033,6 +1033,47 @@ __global__ __launch_bounds__(256) void scan_i8_dot_kernel(
     }
 }
 
+/* pack quantized byte rows into 16-byte units: [g][dpad/16][64][16] — the
+ * layout the dot-form scan streams with 16 B/lane dwordx4 loads (the
+ * uchar4 [g][dpad/4][64][4] layout only gives 4 B/lane per 256 B block).
+ * Used when dpad % 16 == 0; thread per output uint4. */
+__global__ void bytes_pack16_kernel(const uint8_t* __restrict__ rows_q,
+                                    int dpad,
+                                    const int64_t* __restrict__ group_rowbase,
+                                    const int32_t* __restrict__ group_valid,
+                                    const int64_t* __restrict__ slot_rows,
+                                    int64_t ngroups,
+                                    uint8_t* __restrict__ packed) {
+    const int d16 = dpad >> 4;
+    const int64_t nunits = ngroups * (int64_t)d16 * 64;
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= nunits) return;
+    const int lane = (int)(idx & 63);
+    const int64_t rest = idx >> 6;
+    const int m16 = (int)(rest % d16);
+    const int64_t g = rest / d16;
+    uint4 v = {0, 0, 0, 0};
+    if (lane < group_valid[g]) {
+        const int64_t row = slot_rows[group_rowbase[g] + lane];
+        v = *(const uint4*)(rows_q + row * (int64_t)dpad + 16 * m16);
+    }
+    ((uint4*)packed)[idx] = v;
+}
+
+void launch_bytes_pack16(const uint8_t* rows_q, int dpad,
+                         const int64_t* group_rowbase,
+                         const int32_t* group_valid,
+                         const int64_t* slot_rows, int64_t ngroups,
+                         uint8_t* packed, hipStream_t stream) {
+    const int64_t nunits = ngroups * (int64_t)(dpad >> 4) * 64;
+    if (!nunits) return;
+    const int block = 256;
+    const int64_t grid = (nunits + block - 1) / block;
+    hipLaunchKernelGGL(bytes_pack16_kernel, dim3((uint32_t)grid), dim3(block),
+                       0, stream, rows_q, dpad, group_rowbase, group_valid,
+                       slot_rows, ngroups, packed);
+}
+
 /* per-(group,lane) byte-row sum of squares, same wave layout as the scan */
 template <bool UNSIGNED>
 __global__ __launch_bounds__(256) void rownorms_i8_kernel(
@@ -1041,17 +1082,24 @@ __global__ __launch_bounds__(256) void rownorms_i8_kernel(
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
     const int64_t g = (int64_t)blockIdx.x * 4 + wave;
     if (g >= ngroups) return;
-    const uchar4* __restrict__ d =
-        (const uchar4*)(packed + g * (int64_t)64 * dpad) + lane;
+    /* layout16: [g][dpad/16][64][16] (bytes_pack16_kernel) */
+    const uint4* __restrict__ d =
+        (const uint4*)(packed + g * (int64_t)64 * dpad) + lane;
     int32_t acc = 0;
-    const int d4 = dpad >> 2;
-    for (int q4 = 0; q4 < d4; ++q4) {
-        const uchar4 v = d[q4 * 64];
-        const int a = UNSIGNED ? (int)v.x : (int)(int8_t)v.x;
-        const int b = UNSIGNED ? (int)v.y : (int)(int8_t)v.y;
-        const int c = UNSIGNED ? (int)v.z : (int)(int8_t)v.z;
-        const int e = UNSIGNED ? (int)v.w : (int)(int8_t)v.w;
-        acc += a * a + b * b + c * c + e * e;
+    const int d16 = dpad >> 4;
+    for (int m = 0; m < d16; ++m) {
+        const uint4 v = d[m * 64];
+        if (UNSIGNED) {
+            acc = (int32_t)__builtin_amdgcn_udot4(v.x, v.x, (uint32_t)acc, false);
+            acc = (int32_t)__builtin_amdgcn_udot4(v.y, v.y, (uint32_t)acc, false);
+            acc = (int32_t)__builtin_amdgcn_udot4(v.z, v.z, (uint32_t)acc, false);
+            acc = (int32_t)__builtin_amdgcn_udot4(v.w, v.w, (uint32_t)acc, false);
+        } else {
+            acc = __builtin_amdgcn_sdot4((int32_t)v.x, (int32_t)v.x, acc, false);
+            acc = __builtin_amdgcn_sdot4((int32_t)v.y, (int32_t)v.y, acc, false);
+            acc = __builtin_amdgcn_sdot4((int32_t)v.z, (int32_t)v.z, acc, false);
+            acc = __builtin_amdgcn_sdot4((int32_t)v.w, (int32_t)v.w, acc, false);
+        }
     }
     out[g * 64 + lane] = acc;
 }

@@ -957,10 +957,17 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
             }
             HIP_CHECK(hipMalloc(&ix->d_packed_q,
                                 (size_t)ix->total_groups * 64 * ix->dpad));
-            launch_pq_pack(d_qrows.as<uint8_t>(), (int)ix->dpad,
-                           d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
-                           d_sr2.as<int64_t>(), ix->total_groups,
-                           ix->d_packed_q, ix->stream);
+            if ((ix->dpad & 15) == 0)
+                /* 16-byte-unit layout for the dot-form scan */
+                launch_bytes_pack16(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                                    d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
+                                    d_sr2.as<int64_t>(), ix->total_groups,
+                                    ix->d_packed_q, ix->stream);
+            else
+                launch_pq_pack(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                               d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
+                               d_sr2.as<int64_t>(), ix->total_groups,
+                               ix->d_packed_q, ix->stream);
             if ((ix->dpad & 15) == 0) {
                 /* row norms for the dot-form byte scan (rn + qn - 2*dot) */
                 HIP_CHECK(hipMalloc(&ix->d_rownorm_q,
