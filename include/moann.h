@@ -232,6 +232,32 @@ void moann_pairwise_distance(distance_type_t metric, const float* query,
                              const float* rows, uint64_t n, uint32_t dim,
                              int device, float* out_dists, void* errmsg);
 
+/* --- docfilter membership evaluation (moann extension) -------------------
+ * Turns a serialized MatrixOne membership filter into a key bitset for the
+ * filtered-search kernels. Kinds match cgo/usearchex.h:30-33 (the tag bytes
+ * of pkg/common/docfilter); payload formats: bloom = bloomfilter_t flat
+ * bytes (cgo/bloom.h:36-42), croaring = roaring64 PORTABLE serialization
+ * (cgo/croaring.c:99-112), cbitmap = [base u64][nbits u64][words]
+ * (cgo/cbitmap.c:159-201), sorted64 = [count u64][sorted u64s]
+ * (cgo/sorted64.c:26-57). Unknown kinds fail closed with an error. */
+#define MOANN_FILTER_BLOOM 0
+#define MOANN_FILTER_CROARING 2
+#define MOANN_FILTER_CBITMAP 3
+#define MOANN_FILTER_SORTED64 4
+
+/* Evaluate membership of nkeys int64 keys; out_words = (nkeys+31)/32 u32
+ * little-endian bitset words, bit i set iff keys[i] is a member. */
+int moann_docfilter_eval(int kind, const void* payload, uint64_t payload_len,
+                         const int64_t* keys, uint64_t nkeys,
+                         uint32_t* out_words, void* errmsg);
+
+/* Same, over an IVF index's entries in GLOBAL SLOT order — the bitset
+ * moann_ivf_flat_search_filtered consumes ((len(index)+31)/32 words). */
+int moann_ivf_flat_filter_from_docfilter(gpu_ivf_flat_c index_c, int kind,
+                                         const void* payload,
+                                         uint64_t payload_len,
+                                         uint32_t* out_words, void* errmsg);
+
 /* --- perf counters (roofline evidence; DESIGN.md measurement section) ----- */
 
 typedef struct {

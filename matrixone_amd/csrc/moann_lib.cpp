@@ -1246,6 +1246,22 @@ void gpu_ivf_flat_free_result(gpu_ivf_flat_result_c r) {
     delete res;
 }
 
+int moann_ivf_flat_filter_from_docfilter(gpu_ivf_flat_c h, int kind,
+                                         const void* payload,
+                                         uint64_t payload_len,
+                                         uint32_t* out_words, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (!ix->built) throw std::runtime_error("index not built");
+        return moann_docfilter_eval(kind, payload, payload_len,
+                                    ix->h_id_by_slot.data(), ix->count,
+                                    out_words, errmsg);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_filter_from_docfilter", e.what());
+        return -1;
+    }
+}
+
 uint64_t gpu_ivf_flat_cap(gpu_ivf_flat_c h) { return IX(h)->cap; }
 uint64_t gpu_ivf_flat_len(gpu_ivf_flat_c h) { return IX(h)->count; }
 uint32_t gpu_ivf_flat_get_n_list(gpu_ivf_flat_c h) { return IX(h)->nlist; }

@@ -114,6 +114,14 @@ def lib() -> ct.CDLL:
         ct.c_uint32, SearchParams, ct.c_void_p]
     L.moann_ivf_flat_result_wait.restype = ct.c_int
     L.moann_ivf_flat_result_wait.argtypes = [ct.c_void_p, ct.c_void_p]
+    L.moann_docfilter_eval.restype = ct.c_int
+    L.moann_docfilter_eval.argtypes = [
+        ct.c_int, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_uint64,
+        ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_flat_filter_from_docfilter.restype = ct.c_int
+    L.moann_ivf_flat_filter_from_docfilter.argtypes = [
+        ct.c_void_p, ct.c_int, ct.c_void_p, ct.c_uint64, ct.c_void_p,
+        ct.c_void_p]
     L.moann_ivf_flat_slots_of_ids.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
     L.moann_brute_force_search.argtypes = [
@@ -177,6 +185,31 @@ def device_count() -> int:
 def _as_f32(a) -> np.ndarray:
     arr = np.ascontiguousarray(a, dtype=np.float32)
     return arr
+
+
+FILTER_BLOOM = 0
+FILTER_CROARING = 2
+FILTER_CBITMAP = 3
+FILTER_SORTED64 = 4
+
+
+def docfilter_eval(kind: int, payload: bytes, keys) -> np.ndarray:
+    """Evaluate a serialized MatrixOne membership filter (docfilter tag
+    `kind`, cgo/usearchex.h:30-33) over int64 keys -> bool array.
+    Payload formats per include/moann.h."""
+    keys = np.ascontiguousarray(keys, dtype=np.int64)
+    nw = (len(keys) + 31) // 32
+    out = np.zeros(max(1, nw), dtype=np.uint32)
+    buf = ct.create_string_buffer(bytes(payload), len(payload))
+    err = _Err()
+    rc = lib().moann_docfilter_eval(kind, ct.cast(buf, ct.c_void_p),
+                                    len(payload), keys.ctypes.data,
+                                    len(keys), out.ctypes.data, err.ref)
+    err.check("docfilter_eval")
+    if rc != 0:
+        raise MoannError("docfilter_eval failed")
+    idx = np.arange(len(keys))
+    return ((out[idx >> 5] >> (idx & 31)) & 1).astype(bool)
 
 
 class IvfFlatIndex:
@@ -347,6 +380,30 @@ class IvfFlatIndex:
         class _R:
             result_ptr = ptr
         return self._read_result(_R, nq, k, "search_collect")
+
+    def docfilter_bitset(self, kind: int, payload: bytes) -> np.ndarray:
+        """Slot bitset from a serialized docfilter: membership of every
+        entry's id, in global slot order — feed to search_filtered."""
+        n = len(self)
+        nw = (n + 31) // 32
+        out = np.zeros(max(1, nw), dtype=np.uint32)
+        buf = ct.create_string_buffer(bytes(payload), len(payload))
+        err = _Err()
+        rc = lib().moann_ivf_flat_filter_from_docfilter(
+            self._h, kind, ct.cast(buf, ct.c_void_p), len(payload),
+            out.ctypes.data, err.ref)
+        err.check("docfilter_bitset")
+        if rc != 0:
+            raise MoannError("docfilter_bitset failed")
+        return out
+
+    def search_docfiltered(self, queries, k: int, nprobe: int, kind: int,
+                           payload: bytes):
+        """Filtered search gated by a serialized docfilter (the
+        usearchex_filtered_search_with_membership shape, usearchex.c:51-70,
+        evaluated host-side into the slot bitset the kernels consume)."""
+        return self.search_filtered(queries, k, nprobe,
+                                    self.docfilter_bitset(kind, payload))
 
     def set_batch_window(self, window_us: int):
         """ivf_flat_c.h:91 — fuse concurrent search_async calls arriving

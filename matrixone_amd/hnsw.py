@@ -99,6 +99,7 @@ class HnswIndex:
         keys = np.ascontiguousarray(g.keys.astype(np.int64))
         L.moann_hnsw_set_keys(self._h, keys.ctypes.data, err.ref)
         err.check("set_keys")
+        self._gkeys = keys
         offs0, nbrs0 = g.adj[0]
         offs0 = np.ascontiguousarray(offs0, dtype=np.int64)
         nbrs0 = np.ascontiguousarray(nbrs0, dtype=np.uint32)
@@ -150,6 +151,20 @@ class HnswIndex:
                             out_dists.ctypes.data, err.ref)
         err.check("hnsw_search")
         return out_ids, out_dists
+
+    def search_docfiltered(self, queries, ef: int, k: int, kind: int,
+                           payload: bytes):
+        """Filtered beam search gated by a serialized docfilter — the
+        usearchex_filtered_search_with_membership shape (usearchex.c:51-70):
+        each graph key (MO pk) is tested against the filter host-side and
+        the passing set becomes the slot bitset the kernel consumes."""
+        bits = engine.docfilter_eval(kind, payload,
+                                     self._gkeys.astype(np.int64))
+        slots = np.nonzero(bits)[0]
+        words = np.zeros((len(self._gkeys) + 31) // 32, dtype=np.uint32)
+        np.bitwise_or.at(words, slots // 32,
+                         (np.uint32(1) << (slots % 32).astype(np.uint32)))
+        return self.search_filtered(queries, ef, k, words)
 
     def search_filtered(self, queries, ef: int, k: int,
                         slot_bitset: np.ndarray):

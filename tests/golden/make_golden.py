@@ -90,3 +90,50 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def make_docfilter_golden():
+    """Golden membership-filter payloads + expected membership, built by the
+    REFERENCE's own filter code (oracle/_ref/libref_filters.so — cgo/bloom.c,
+    cbitmap.c, sorted64.c, croaring.c + vendored CRoaring). The payloads
+    travel as fixtures so the GPU box (no /root/reference) can still pin
+    matrixone_amd's independent parsers bit-exactly."""
+    from oracle import docfilter as DF
+    rng = np.random.Generator(np.random.PCG64(20260915))
+    keys = rng.choice(1_000_000, 20000, replace=False).astype(np.int64)
+    probe = np.concatenate([
+        rng.choice(1_200_000, 20000, replace=False),
+        keys[:1000],
+        [0, 1, 999_999, 1_199_999],
+    ]).astype(np.int64)
+    out = {}
+    pay = DF.build_bloom(keys, nbits=1 << 18, k=5, seed=0x5EED_0F_B100)
+    out["bloom_payload"] = np.frombuffer(pay, dtype=np.uint8)
+    out["bloom_expect"] = DF.ref_bloom_test(pay, probe)
+    pc = DF.build_cbitmap(keys)
+    out["cbitmap_payload"] = np.frombuffer(pc, dtype=np.uint8)
+    out["cbitmap_expect"] = DF.ref_contains(DF.KIND_CBITMAP, pc, probe)
+    ps = DF.build_sorted64(keys)
+    out["sorted64_payload"] = np.frombuffer(ps, dtype=np.uint8)
+    out["sorted64_expect"] = DF.ref_contains(DF.KIND_SORTED64, ps, probe)
+    # croaring: mixed container shapes in one bitmap (array + bitset + runs
+    # + a high 32-bit bucket)
+    ck = np.concatenate([
+        keys[:4000],
+        np.arange(2_000_000, 2_070_000),          # bitset container
+        np.arange(5_000_000, 5_004_000),          # run container
+        rng.choice(10_000, 2000, replace=False) + (1 << 33),
+    ]).astype(np.int64)
+    pr = DF.build_croaring(ck)
+    cprobe = np.concatenate([probe, ck[::5], ck + 1,
+                             rng.integers(0, 2**34, 4000)]).astype(np.int64)
+    out["croaring_payload"] = np.frombuffer(pr, dtype=np.uint8)
+    out["croaring_expect"] = DF.ref_contains(DF.KIND_CROARING, pr, cprobe)
+    out["probe_keys"] = probe
+    out["croaring_probe_keys"] = cprobe
+    np.savez_compressed(os.path.join(HERE, "docfilter.npz"), **out)
+    print("docfilter.npz written")
+
+
+if __name__ == "__main__" and os.environ.get("MOANN_GOLDEN_DOCFILTER"):
+    make_docfilter_golden()

