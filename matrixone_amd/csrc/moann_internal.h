@@ -113,6 +113,11 @@ void launch_bytes_pack16(const uint8_t* rows_q, int dpad,
                          uint8_t* packed, hipStream_t stream);
 void launch_rownorms_i8(bool uns, const uint8_t* packed, int64_t ngroups,
                         int dpad, int32_t* out, hipStream_t stream);
+void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
+                         const int64_t* group_slotbase,
+                         const int32_t* group_valid, int64_t ngroups,
+                         int64_t slot_lo, int64_t slot_hi, uint8_t* out,
+                         hipStream_t stream);
 void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
                           int in_stride, int dim, int dpad, float fmul,
                           float fadd, uint8_t* out, hipStream_t stream);

@@ -680,6 +680,59 @@ void launch_unpack(const float* packed, int dim, int dpad,
                        ngroups, slot_lo, slot_hi, out);
 }
 
+/* byte-storage inverse of pack: packed bytes -> row-major [slot][dim]
+ * bytes, both layouts (layout16 = [g][dpad/16][64][16] when dpad%16==0,
+ * else the uchar4 [g][dpad/4][64][4]). save_dir for narrow storage. */
+template <bool L16>
+__global__ void unpack_bytes_kernel(const uint8_t* __restrict__ packed,
+                                    int dim, int dpad,
+                                    const int64_t* __restrict__ group_slotbase,
+                                    const int32_t* __restrict__ group_valid,
+                                    int64_t ngroups, int64_t slot_lo,
+                                    int64_t slot_hi,
+                                    uint8_t* __restrict__ out) {
+    const int U = L16 ? 16 : 4;
+    const int64_t nunits = ngroups * (int64_t)(dpad / U) * 64;
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= nunits) return;
+    const int lane = (int)(idx & 63);
+    const int64_t rest = idx >> 6;
+    const int m = (int)(rest % (dpad / U));
+    const int64_t g = rest / (dpad / U);
+    if (lane >= group_valid[g]) return;
+    const int64_t slot = group_slotbase[g] + lane;
+    if (slot < slot_lo || slot >= slot_hi) return;
+    const uint8_t* src = packed + idx * U;
+    uint8_t* dst = out + (slot - slot_lo) * (int64_t)dim;
+    const int base = m * U;
+#pragma unroll
+    for (int b = 0; b < U; ++b)
+        if (base + b < dim) dst[base + b] = src[b];
+}
+
+void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
+                         const int64_t* group_slotbase,
+                         const int32_t* group_valid, int64_t ngroups,
+                         int64_t slot_lo, int64_t slot_hi, uint8_t* out,
+                         hipStream_t stream) {
+    const bool l16 = (dpad & 15) == 0;
+    const int U = l16 ? 16 : 4;
+    const int64_t nunits = ngroups * (int64_t)(dpad / U) * 64;
+    if (!nunits) return;
+    const int block = 256;
+    const int64_t grid = (nunits + block - 1) / block;
+    if (l16)
+        hipLaunchKernelGGL(unpack_bytes_kernel<true>, dim3((uint32_t)grid),
+                           dim3(block), 0, stream, packed, dim, dpad,
+                           group_slotbase, group_valid, ngroups, slot_lo,
+                           slot_hi, out);
+    else
+        hipLaunchKernelGGL(unpack_bytes_kernel<false>, dim3((uint32_t)grid),
+                           dim3(block), 0, stream, packed, dim, dpad,
+                           group_slotbase, group_valid, ngroups, slot_lo,
+                           slot_hi, out);
+}
+
 /* -------------------------- quantized (int8/uint8) scan ------------------
  * Narrow-storage variant of the list scan (reference: int8/uint8 entries
  * quantized by q(x)=round(x*mul+add), distance computed IN THE QUANTIZED

@@ -1376,9 +1376,6 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
         auto ix = IX(h);
         std::lock_guard<std::mutex> lk(ix->mu);
         if (!ix->built) throw std::runtime_error("index not built");
-        if (ix->qtype != Quantization_F32)
-            throw std::runtime_error(
-                "save_dir for narrow storage not yet wired");
         HIP_CHECK(hipSetDevice(ix->device));
         std::string d(dir);
         auto wfile = [&](const std::string& name, const void* p, size_t n) {
@@ -1396,9 +1393,12 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
                      "{\n  \"format\": \"moann-ivf-flat-v1\",\n"
                      "  \"dimension\": %u,\n  \"n_lists\": %u,\n"
                      "  \"metric\": %d,\n  \"count\": %llu,\n"
+                     "  \"qtype\": %d,\n"
+                     "  \"quant_min\": %.17g,\n  \"quant_max\": %.17g,\n"
                      "  \"quant_mul\": %.17g\n}\n",
                      ix->dim, ix->nlist, (int)ix->metric,
-                     (unsigned long long)ix->count, ix->quant_mul);
+                     (unsigned long long)ix->count, (int)ix->qtype,
+                     ix->quant_min, ix->quant_max, ix->quant_mul);
             wfile("manifest.json", buf, strlen(buf));
         }
         wfile("centroids.f32", ix->h_centroids.data(),
@@ -1413,6 +1413,43 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
                     a[sidx] = (int32_t)l;
             wfile("assign.i32", a.data(), ix->count * 4);
         }
+        if (ix->qtype != Quantization_F32) {
+            /* narrow storage: the f32 staging was dropped at build; save the
+             * QUANTIZED bytes (vectors.u8). load_dir dequantizes with the
+             * manifest's quantizer and re-quantizes at build — an exact
+             * round trip (integer codes survive the f32 detour: the
+             * reconstruction error is << 0.5 code steps). */
+            FILE* f = fopen((d + "/vectors.u8").c_str(), "wb");
+            if (!f) throw std::runtime_error("cannot write vectors.u8");
+            DevBuf d_gsb, d_gv, d_tmp;
+            d_gsb.ensure(ix->total_groups * 8);
+            d_gv.ensure(ix->total_groups * 4);
+            HIP_CHECK(hipMemcpy(d_gsb.ptr, ix->group_slotbase.data(),
+                                ix->total_groups * 8, hipMemcpyHostToDevice));
+            HIP_CHECK(hipMemcpy(d_gv.ptr, ix->group_valid.data(),
+                                ix->total_groups * 4, hipMemcpyHostToDevice));
+            const int64_t CH = 1 << 21;
+            d_tmp.ensure((size_t)CH * ix->dim);
+            std::vector<uint8_t> hbuf((size_t)CH * ix->dim);
+            for (int64_t lo = 0; lo < (int64_t)ix->count; lo += CH) {
+                const int64_t hi = std::min<int64_t>(lo + CH, ix->count);
+                launch_unpack_bytes(ix->d_packed_q, (int)ix->dim,
+                                    (int)ix->dpad, d_gsb.as<int64_t>(),
+                                    d_gv.as<int32_t>(), ix->total_groups, lo,
+                                    hi, d_tmp.as<uint8_t>(), ix->stream);
+                HIP_CHECK(hipMemcpyAsync(hbuf.data(), d_tmp.ptr,
+                                         (size_t)(hi - lo) * ix->dim,
+                                         hipMemcpyDeviceToHost, ix->stream));
+                HIP_CHECK(hipStreamSynchronize(ix->stream));
+                HIP_CHECK(hipGetLastError());
+                const size_t n = (size_t)(hi - lo) * ix->dim;
+                if (fwrite(hbuf.data(), 1, n, f) != n) {
+                    fclose(f);
+                    throw std::runtime_error("short write vectors.u8");
+                }
+            }
+            fclose(f);
+        } else
         /* vectors.f32: unpack in chunks of groups */
         {
             FILE* f = fopen((d + "/vectors.f32").c_str(), "wb");
@@ -1487,6 +1524,17 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
                     "manifest does not match the created index");
             if (count > ix->cap)
                 throw std::runtime_error("manifest count exceeds capacity");
+            int qtype = (int)Quantization_F32;
+            double qmin = 0.0, qmax = 0.0;
+            if (const char* q = strstr(buf, "\"qtype\":"))
+                sscanf(q, "\"qtype\": %d", &qtype);
+            if (const char* q = strstr(buf, "\"quant_min\":"))
+                sscanf(q, "\"quant_min\": %lg", &qmin);
+            if (const char* q = strstr(buf, "\"quant_max\":"))
+                sscanf(q, "\"quant_max\": %lg", &qmax);
+            if (qtype != (int)ix->qtype)
+                throw std::runtime_error(
+                    "manifest storage type does not match the created index");
             std::vector<float> cents((size_t)nlist * dim);
             rfile("centroids.f32", cents.data(), cents.size() * 4);
             moann_ivf_flat_set_centroids(h, cents.data(), nlist, errmsg);
@@ -1495,15 +1543,37 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
             std::vector<int32_t> assign(count);
             rfile("assign.i32", assign.data(), count * 4);
             /* vectors streamed into add_chunk */
-            FILE* vf = fopen((d + "/vectors.f32").c_str(), "rb");
-            if (!vf) throw std::runtime_error("cannot read vectors.f32");
+            const bool narrow = qtype != (int)Quantization_F32;
+            const char* vname = narrow ? "/vectors.u8" : "/vectors.f32";
+            FILE* vf = fopen((d + vname).c_str(), "rb");
+            if (!vf)
+                throw std::runtime_error(std::string("cannot read ") + vname);
+            if (narrow) {
+                /* pin the quantizer BEFORE build so re-quantization uses
+                 * the saved parameters */
+                gpu_ivf_flat_set_quantizer(h, (float)qmin, (float)qmax,
+                                           errmsg);
+            }
+            const double mul = ix->quant_mul, add = ix->quant_add;
             const size_t CH = 1 << 20;
             std::vector<float> vbuf(CH * dim);
+            std::vector<uint8_t> qbuf(narrow ? CH * dim : 0);
             unsigned long long done = 0;
             while (done < count) {
                 const size_t n = (size_t)std::min<unsigned long long>(
                     CH, count - done);
-                if (fread(vbuf.data(), 4, n * dim, vf) != n * dim) {
+                if (narrow) {
+                    if (fread(qbuf.data(), 1, n * dim, vf) != n * dim) {
+                        fclose(vf);
+                        throw std::runtime_error("short read vectors.u8");
+                    }
+                    const bool uns = qtype == (int)Quantization_UINT8;
+                    for (size_t i = 0; i < n * dim; ++i) {
+                        const double q = uns ? (double)qbuf[i]
+                                             : (double)(int8_t)qbuf[i];
+                        vbuf[i] = (float)((q - add) / mul);
+                    }
+                } else if (fread(vbuf.data(), 4, n * dim, vf) != n * dim) {
                     fclose(vf);
                     throw std::runtime_error("short read vectors.f32");
                 }
