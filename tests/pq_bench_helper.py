@@ -1,0 +1,141 @@
+"""IVF-PQ measurement helper at the config-5 SHAPE (BASELINE.json configs[4]:
+IVF-PQ L2, d=1536, PQ96x8, nprobe=64, batch=4096) scaled to a 1-GPU row
+count (--rows; the named config is 100M sharded over 8 GPUs — the per-GPU
+slice is what one box can measure). Run under gpurun; writes one JSON line.
+
+Harness side (index BUILD, out of the judged search path): torch k-means
+for the coarse centroids, batched-bmm k-means for the 96 per-subspace
+codebooks over coarse residuals (oracle/pq.py conventions), chunked bmm
+encoding. The judged part is the ADC search (pq_scan LUT-in-LDS kernel).
+"""
+
+import argparse
+import json
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, ".")
+import bench  # noqa: E402  (kmeans_torch / assign_torch / make_mixture)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rows", type=int, default=4_000_000)
+    ap.add_argument("--dim", type=int, default=1536)
+    ap.add_argument("--nsub", type=int, default=96)
+    ap.add_argument("--nlist", type=int, default=4096)
+    ap.add_argument("--nprobe", type=int, default=64)
+    ap.add_argument("--batch", type=int, default=4096)
+    ap.add_argument("--k", type=int, default=10)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--gt-queries", type=int, default=512)
+    args = ap.parse_args()
+    dev = torch.device("cuda:0")
+    dsub = args.dim // args.nsub
+
+    t0 = time.perf_counter()
+    data = bench.make_mixture(args.rows, args.dim, 1, 0, dev)
+    qall = bench.make_mixture(args.batch + args.gt_queries, args.dim, 1, 7,
+                              dev)
+    cent = bench.kmeans_torch(data, args.nlist, iters=12, seed=3)
+
+    def assign_bf16(dat, cc, chunk=1 << 18):
+        # bf16 matmul (f32 accumulate) keeps the 4M x 4096 x 1536 coarse
+        # assignment in harness-time; borderline rows may flip lists, which
+        # only perturbs the index layout, never the measured search
+        out = torch.empty(dat.shape[0], dtype=torch.int32, device=dat.device)
+        cb16 = cc.to(torch.bfloat16)
+        cn = (cc * cc).sum(1)
+        for c0 in range(0, dat.shape[0], chunk):
+            blk = dat[c0:c0 + chunk]
+            ip = (blk.to(torch.bfloat16) @ cb16.T).float()
+            out[c0:c0 + chunk] = (cn[None, :] - 2.0 * ip).argmin(1).to(torch.int32)
+        return out
+
+    assign = assign_bf16(data, cent)
+    log = lambda m: print(m, file=sys.stderr, flush=True)
+    log(f"[pq] data+kmeans {time.perf_counter()-t0:.1f}s")
+
+    # ---- codebooks: batched k-means over residual subspaces --------------
+    t0 = time.perf_counter()
+    g = torch.Generator(device="cpu").manual_seed(11)
+    samp_idx = torch.randperm(args.rows, generator=g)[:262144].to(dev)
+    res = (data[samp_idx] - cent[assign[samp_idx].long()]) \
+        .view(-1, args.nsub, dsub).permute(1, 0, 2).contiguous()  # [S,n,ds]
+    nsamp = res.shape[1]
+    sel = torch.randperm(nsamp, generator=g)[:256].to(dev)
+    cb = res[:, sel, :].clone()                                   # [S,256,ds]
+    for _ in range(10):
+        d2 = (res * res).sum(-1, keepdim=True) \
+            - 2.0 * torch.bmm(res, cb.transpose(1, 2)) \
+            + (cb * cb).sum(-1)[:, None, :]
+        a = d2.argmin(-1)                                         # [S,n]
+        for s in range(args.nsub):
+            sums = torch.zeros(256, dsub, device=dev)
+            cnts = torch.zeros(256, device=dev)
+            sums.index_add_(0, a[s], res[s])
+            cnts.index_add_(0, a[s], torch.ones(nsamp, device=dev))
+            nz = cnts > 0
+            cb[s][nz] = sums[nz] / cnts[nz][:, None]
+    log(f"[pq] codebooks {time.perf_counter()-t0:.1f}s")
+
+    # ---- encode all rows (chunked bmm argmin) ----------------------------
+    t0 = time.perf_counter()
+    codes = torch.empty(args.rows, args.nsub, dtype=torch.uint8, device=dev)
+    cb2 = (cb * cb).sum(-1)[:, None, :]
+    CH = 1 << 19
+    for lo in range(0, args.rows, CH):
+        hi = min(lo + CH, args.rows)
+        r = (data[lo:hi] - cent[assign[lo:hi].long()]) \
+            .view(hi - lo, args.nsub, dsub).permute(1, 0, 2).contiguous()
+        d2 = -2.0 * torch.bmm(r, cb.transpose(1, 2)) + cb2
+        codes[lo:hi] = d2.argmin(-1).permute(1, 0).to(torch.uint8)
+    torch.cuda.synchronize()
+    log(f"[pq] encode {time.perf_counter()-t0:.1f}s")
+
+    # ---- exact GT on the f32 data (recall reference) ---------------------
+    gtq = qall[args.batch:args.batch + args.gt_queries]
+    gt = bench.exact_gt(data, gtq, args.k)
+
+    # ---- build the GPU index --------------------------------------------
+    from matrixone_amd import engine
+    t0 = time.perf_counter()
+    ix = engine.IvfPqIndex(args.dim, args.nsub, args.nlist,
+                           capacity=args.rows)
+    ix.set_centroids(cent.cpu().numpy())
+    ix.set_codebooks(cb.cpu().numpy())
+    ix.add_codes(codes.cpu().numpy())
+    ix.set_assignments(assign.cpu().numpy().astype(np.int32))
+    ix.build()
+    log(f"[pq] index build {time.perf_counter()-t0:.1f}s")
+
+    queries = qall[:args.batch].cpu().numpy()
+    ids, _ = ix.search(queries, args.k, args.nprobe)  # warmup
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        ix.search(queries, args.k, args.nprobe)
+    dt = (time.perf_counter() - t0) / args.steps
+    gids, _ = ix.search(gtq.cpu().numpy(), args.k, args.nprobe)
+    gtn = gt.cpu().numpy() if hasattr(gt, "cpu") else gt
+    recall = float(np.mean([
+        len(set(gids[i].tolist()) & set(gtn[i].tolist())) / args.k
+        for i in range(args.gt_queries)]))
+    out = {
+        "workload": f"ivfpq-l2-{args.rows}x{args.dim}-PQ{args.nsub}x8-"
+                    f"nlist{args.nlist}-nprobe{args.nprobe}-"
+                    f"batch{args.batch}-k{args.k}",
+        "qps": round(args.batch / dt, 1),
+        "ms_per_batch": round(dt * 1e3, 3),
+        "recall_at_10_vs_exact": round(recall, 4),
+        "code_bytes_per_row": args.nsub,
+        "note": "config-5 shape, 1-GPU slice; recall bound is PQ "
+                "quantization loss (ADC), not the kernel",
+    }
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
