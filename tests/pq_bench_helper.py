@@ -37,9 +37,10 @@ def main():
     dsub = args.dim // args.nsub
 
     t0 = time.perf_counter()
-    data = bench.make_mixture(args.rows, args.dim, 1, 0, dev)
-    qall = bench.make_mixture(args.batch + args.gt_queries, args.dim, 1, 7,
-                              dev)
+    data, queries_host = bench.make_mixture(args.rows, args.dim, 1, 0, dev)
+    need = args.batch + args.gt_queries
+    assert queries_host.shape[0] >= need
+    qall = queries_host[:need].to(dev)
     cent = bench.kmeans_torch(data, args.nlist, iters=12, seed=3)
 
     def assign_bf16(dat, cc, chunk=1 << 18):
