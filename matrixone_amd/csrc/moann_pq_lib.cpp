@@ -25,14 +25,18 @@
 #include "moann_internal.h"
 
 namespace moann {
-void launch_pq_scan(const uint8_t* codes, const float* queries,
-                    const float* cents, const float* atab, const float* btab,
-                    int nsub, int dpad, int njobs,
-                    const int64_t* j_databaseg,
-                    const int32_t* j_gcount, const int32_t* j_rows,
-                    const int32_t* j_query, const int32_t* j_list,
-                    const int64_t* j_outbase, float* dists_out,
-                    hipStream_t stream);
+void launch_pq_scan_q(const uint8_t* codes, const float* atab,
+                      const float* row_b, const float* qnorms,
+                      const float* cnorm2, const int32_t* csel_slots,
+                      const float* csel_dists, int probe,
+                      const int64_t* qoffs, const int64_t* probe_offs,
+                      const int64_t* list_gbase, const int32_t* list_rows,
+                      const int64_t* list_slot_base, int nsub, int nq,
+                      float* dists_out, hipStream_t stream);
+void launch_pq_rowb(bool w16, const uint8_t* codes, const float* btab,
+                    const int32_t* g2list, const int64_t* g2slot,
+                    const int32_t* g2valid, int64_t ngroups, int nsub,
+                    float* row_b, hipStream_t stream);
 void launch_pq_tables(const float* cents, const float* queries,
                       const float* cb, int nsub, int sd, int dpad,
                       int64_t nlist_or_nq, bool btab_mode, float* out,
@@ -65,7 +69,11 @@ struct PqIndex {
     float* d_cents_rows = nullptr;    /* [nlist][dpad] */
     float* d_cent_packed = nullptr;   /* interleaved (rank stage) */
     float* d_codebooks = nullptr;     /* [nsub][256][sd] */
-    float* d_btab = nullptr;          /* [nlist][nsub*256] ADC B table */
+    float* d_btab = nullptr;          /* [nlist][nsub*256] (build only) */
+    float* d_rowb = nullptr;          /* [count] per-row ADC constant */
+    float* d_cnorm2 = nullptr;        /* [nlist] |c_l|^2 */
+    int64_t* d_gbase_dev = nullptr;   /* [nlist] list_gbase */
+    int32_t* d_rows_dev = nullptr;    /* [nlist] list_rows */
     int64_t* d_id_by_slot = nullptr;
     int64_t* d_list_slot_base = nullptr;
     std::vector<int64_t> list_gbase;
@@ -75,7 +83,7 @@ struct PqIndex {
 
     MoannDevBuf w_queries, w_cdists, w_csel_slots, w_csel_dists, w_jobs;
     MoannDevBuf w_cand, w_sel_slots, w_sel_dists, w_probe_lists, w_probe_offs;
-    MoannDevBuf w_qoffs, w_out_ids, w_out_dists, w_atab;
+    MoannDevBuf w_qoffs, w_out_ids, w_out_dists, w_atab, w_qn;
 
     std::mutex mu;
     moann_perf_t perf {};
@@ -86,7 +94,8 @@ struct PqIndex {
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_codes, (void*)d_cents_rows,
                        (void*)d_cent_packed, (void*)d_codebooks,
-                       (void*)d_btab,
+                       (void*)d_btab, (void*)d_rowb, (void*)d_cnorm2,
+                       (void*)d_gbase_dev, (void*)d_rows_dev,
                        (void*)d_id_by_slot, (void*)d_list_slot_base})
             if (p) (void)hipFree(p);
         if (ev_made)
@@ -220,70 +229,11 @@ std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
     }
     const int64_t total_cand = qoffs[nq];
 
-    std::vector<int64_t> databaseg, outbase;
-    std::vector<int32_t> gcount, rows, jq, jl;
-    for (uint64_t q = 0; q < nq; ++q)
-        for (uint32_t r = 0; r < probe; ++r) {
-            const int32_t l = h_probe[q * probe + r];
-            if (l < 0 || ix->list_rows[l] == 0) continue;
-            databaseg.push_back(ix->list_gbase[l]);
-            gcount.push_back(
-                (int32_t)((ix->list_rows[l] + 63) / 64));
-            rows.push_back(ix->list_rows[l]);
-            jq.push_back((int32_t)q);
-            jl.push_back(l);
-            outbase.push_back(qoffs[q] + probe_offs[q * (probe + 1) + r]);
-        }
-    const int njobs = (int)databaseg.size();
-
     ix->w_cand.ensure(std::max<int64_t>(1, total_cand) * 4);
-    {
-        auto al8 = [](size_t b) { return (b + 7) & ~7ull; };
-        const size_t need = 2 * al8((size_t)njobs * 8) +
-                            4 * al8((size_t)njobs * 4) + 64;
-        MoannDevBuf& wj = ix->w_jobs;
-        wj.ensure(std::max(wj.bytes, need)); /* may still hold rank jobs —
-            rank stage completed before the sync above */
-        uint8_t* p = wj.as<uint8_t>();
-        auto up = [&](const void* src, size_t bytes) {
-            void* r = p;
-            if (bytes)
-                MOANN_HIP_CHECK(hipMemcpyAsync(p, src, bytes,
-                                               hipMemcpyHostToDevice, s));
-            p += (bytes + 7) & ~7ull;
-            return r;
-        };
-        const int64_t* d_dbg = (int64_t*)up(databaseg.data(), njobs * 8);
-        const int64_t* d_ob = (int64_t*)up(outbase.data(), njobs * 8);
-        const int32_t* d_gc = (int32_t*)up(gcount.data(), njobs * 4);
-        const int32_t* d_rw = (int32_t*)up(rows.data(), njobs * 4);
-        const int32_t* d_jq = (int32_t*)up(jq.data(), njobs * 4);
-        const int32_t* d_jl = (int32_t*)up(jl.data(), njobs * 4);
-        /* per-query A table for this batch (B is build-time) */
-        ix->w_atab.ensure((size_t)nq * ix->nsub * 256 * 4);
-        launch_pq_tables(nullptr, d_q, ix->d_codebooks, (int)ix->nsub,
-                         (int)ix->sd, dpad, (int64_t)nq, false,
-                         ix->w_atab.as<float>(), s);
-        MOANN_KCHECK("pq-atab");
-        MOANN_HIP_CHECK(hipEventRecord(ix->ev[0], s));
-        launch_pq_scan(ix->d_codes, d_q, ix->d_cents_rows,
-                       ix->w_atab.as<float>(), ix->d_btab,
-                       (int)ix->nsub, dpad, njobs, d_dbg, d_gc,
-                       d_rw, d_jq, d_jl, d_ob, ix->w_cand.as<float>(), s);
-        MOANN_KCHECK("pq-scan");
-        MOANN_HIP_CHECK(hipEventRecord(ix->ev[1], s));
-    }
-
-    /* 3. top-k + gather (shared kernels) */
+    /* offsets to device BEFORE the scan (the per-query kernel walks them) */
     ix->w_qoffs.ensure((nq + 1) * 8);
     MOANN_HIP_CHECK(hipMemcpyAsync(ix->w_qoffs.ptr, qoffs.data(),
                                    (nq + 1) * 8, hipMemcpyHostToDevice, s));
-    ix->w_sel_slots.ensure((size_t)nq * limit * 4);
-    ix->w_sel_dists.ensure((size_t)nq * limit * 4);
-    launch_topk(ix->w_cand.as<float>(), ix->w_qoffs.as<int64_t>(), 0, (int)nq,
-                (int)limit, ix->w_sel_slots.as<int32_t>(),
-                ix->w_sel_dists.as<float>(), s);
-    MOANN_KCHECK("pq-cand-topk");
     ix->w_probe_lists.ensure((size_t)nq * probe * 4);
     ix->w_probe_offs.ensure(probe_offs.size() * 8);
     MOANN_HIP_CHECK(hipMemcpyAsync(ix->w_probe_lists.ptr, h_probe.data(),
@@ -292,6 +242,37 @@ std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
     MOANN_HIP_CHECK(hipMemcpyAsync(ix->w_probe_offs.ptr, probe_offs.data(),
                                    probe_offs.size() * 8,
                                    hipMemcpyHostToDevice, s));
+    {
+        /* per-query A table + |q|^2 for this batch (row_b / |c|^2 are
+         * build-time; -2(q.c_l) falls out of the rank stage's distances) */
+        ix->w_atab.ensure((size_t)nq * ix->nsub * 256 * 4);
+        launch_pq_tables(nullptr, d_q, ix->d_codebooks, (int)ix->nsub,
+                         (int)ix->sd, dpad, (int64_t)nq, false,
+                         ix->w_atab.as<float>(), s);
+        MOANN_KCHECK("pq-atab");
+        ix->w_qn.ensure(nq * 4);
+        launch_qnorms(d_q, (int)nq, dpad, ix->w_qn.as<float>(), s);
+        MOANN_KCHECK("pq-qnorms");
+        MOANN_HIP_CHECK(hipEventRecord(ix->ev[0], s));
+        launch_pq_scan_q(ix->d_codes, ix->w_atab.as<float>(), ix->d_rowb,
+                         ix->w_qn.as<float>(), ix->d_cnorm2,
+                         ix->w_csel_slots.as<int32_t>(),
+                         ix->w_csel_dists.as<float>(), (int)probe,
+                         ix->w_qoffs.as<int64_t>(),
+                         ix->w_probe_offs.as<int64_t>(), ix->d_gbase_dev,
+                         ix->d_rows_dev, ix->d_list_slot_base,
+                         (int)ix->nsub, (int)nq, ix->w_cand.as<float>(), s);
+        MOANN_KCHECK("pq-scan");
+        MOANN_HIP_CHECK(hipEventRecord(ix->ev[1], s));
+    }
+
+    /* 3. top-k + gather (shared kernels) */
+    ix->w_sel_slots.ensure((size_t)nq * limit * 4);
+    ix->w_sel_dists.ensure((size_t)nq * limit * 4);
+    launch_topk(ix->w_cand.as<float>(), ix->w_qoffs.as<int64_t>(), 0, (int)nq,
+                (int)limit, ix->w_sel_slots.as<int32_t>(),
+                ix->w_sel_dists.as<float>(), s);
+    MOANN_KCHECK("pq-cand-topk");
     ix->w_out_ids.ensure((size_t)nq * limit * 8);
     ix->w_out_dists.ensure((size_t)nq * limit * 4);
     launch_gather(ix->w_sel_slots.as<int32_t>(), ix->w_sel_dists.as<float>(),
@@ -552,6 +533,66 @@ void moann_ivf_pq_build(moann_ivf_pq_c h, void* errmsg) {
                          (int64_t)ix->nlist, true, ix->d_btab, ix->stream);
         MOANN_HIP_CHECK(hipStreamSynchronize(ix->stream));
         MOANN_KCHECK("pq-btab");
+        /* fold B into the per-row constant row_b[slot] = sum_m B_l[m][code],
+         * then drop the B table — the search LUT is A_q only */
+        {
+            std::vector<int32_t> g2list(std::max<int64_t>(1, ix->total_groups));
+            std::vector<int64_t> g2slot(std::max<int64_t>(1, ix->total_groups));
+            std::vector<int32_t> g2valid(std::max<int64_t>(1, ix->total_groups));
+            for (uint32_t l = 0; l < ix->nlist; ++l) {
+                const int64_t lg = (ix->list_rows[l] + 63) / 64;
+                for (int64_t gi = 0; gi < lg; ++gi) {
+                    const int64_t g = ix->list_gbase[l] + gi;
+                    g2list[g] = (int32_t)l;
+                    g2slot[g] = ix->list_slot_base[l] + gi * 64;
+                    g2valid[g] = (int32_t)std::min<int64_t>(
+                        64, ix->list_rows[l] - gi * 64);
+                }
+            }
+            MoannDevBuf d_l, d_s2, d_v;
+            d_l.ensure(g2list.size() * 4);
+            d_s2.ensure(g2slot.size() * 8);
+            d_v.ensure(g2valid.size() * 4);
+            MOANN_HIP_CHECK(hipMemcpy(d_l.ptr, g2list.data(),
+                                      g2list.size() * 4,
+                                      hipMemcpyHostToDevice));
+            MOANN_HIP_CHECK(hipMemcpy(d_s2.ptr, g2slot.data(),
+                                      g2slot.size() * 8,
+                                      hipMemcpyHostToDevice));
+            MOANN_HIP_CHECK(hipMemcpy(d_v.ptr, g2valid.data(),
+                                      g2valid.size() * 4,
+                                      hipMemcpyHostToDevice));
+            MOANN_HIP_CHECK(hipMalloc(&ix->d_rowb,
+                                      std::max<uint64_t>(1, ix->count) * 4));
+            launch_pq_rowb((ix->nsub & 15) == 0, ix->d_codes, ix->d_btab,
+                           d_l.as<int32_t>(), d_s2.as<int64_t>(),
+                           d_v.as<int32_t>(), ix->total_groups,
+                           (int)ix->nsub, ix->d_rowb, ix->stream);
+            MOANN_HIP_CHECK(hipStreamSynchronize(ix->stream));
+            MOANN_KCHECK("pq-rowb");
+            MOANN_HIP_CHECK(hipFree(ix->d_btab));
+            ix->d_btab = nullptr;
+        }
+        /* |c_l|^2 + list geometry on device (per-query scan kernel) */
+        {
+            std::vector<float> cn(ix->nlist);
+            for (uint32_t l = 0; l < ix->nlist; ++l) {
+                double a = 0;
+                const float* c = ix->h_centroids.data() + (size_t)l * ix->dim;
+                for (uint32_t t = 0; t < ix->dim; ++t)
+                    a += (double)c[t] * c[t];
+                cn[l] = (float)a;
+            }
+            MOANN_HIP_CHECK(hipMalloc(&ix->d_cnorm2, ix->nlist * 4));
+            MOANN_HIP_CHECK(hipMemcpy(ix->d_cnorm2, cn.data(), ix->nlist * 4,
+                                      hipMemcpyHostToDevice));
+            MOANN_HIP_CHECK(hipMalloc(&ix->d_gbase_dev, ix->nlist * 8));
+            MOANN_HIP_CHECK(hipMemcpy(ix->d_gbase_dev, ix->list_gbase.data(),
+                                      ix->nlist * 8, hipMemcpyHostToDevice));
+            MOANN_HIP_CHECK(hipMalloc(&ix->d_rows_dev, ix->nlist * 4));
+            MOANN_HIP_CHECK(hipMemcpy(ix->d_rows_dev, ix->list_rows.data(),
+                                      ix->nlist * 4, hipMemcpyHostToDevice));
+        }
         std::vector<int64_t> id_by_slot(std::max<uint64_t>(1, ix->count));
         for (uint64_t sidx = 0; sidx < ix->count; ++sidx)
             id_by_slot[sidx] = ix->h_ids[slot_rows[sidx]];

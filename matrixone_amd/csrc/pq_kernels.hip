@@ -70,129 +70,157 @@ __global__ __launch_bounds__(256) void pq_atab_kernel(
     atab[idx] = a;
 }
 
-/* One workgroup per (probed list, query): LUT[m][c] = A_q + B_l summed in
- * LDS, the per-pair constant -2*(q . c_l) block-reduced once, then the
- * list's packed codes streamed — W=16 (dwordx4, layout16) when
- * nsub%16==0, else the uchar4 layout. */
+/* Per-row ADC constant, folded at BUILD: row_b[slot] = sum_m B_l[m][code_m]
+ * — query-independent, so the search LUT is A_q ONLY and one workgroup
+ * serves a whole QUERY (every probed list) with a single LDS table.
+ * dist(q,row) = sum_m A_q[m][code_m] + row_b[slot] - 2*(q . c_l), and
+ * -2(q.c_l) = cdist(q,l) - |q|^2 - |c_l|^2 comes free from the coarse-rank
+ * stage's exact L2 distances. */
 template <int W>
-__global__ __launch_bounds__(256) void pq_scan_kernel(
-    const uint8_t* __restrict__ codes,
-    const float* __restrict__ queries, /* [nq][dpad] */
-    const float* __restrict__ cents,   /* [nlist][dpad] row-major */
-    const float* __restrict__ atab,    /* [nq][nsub*256] */
-    const float* __restrict__ btab,    /* [nlist][nsub*256] */
-    int nsub, int dpad,
-    const int64_t* __restrict__ j_databaseg,
-    const int32_t* __restrict__ j_gcount, const int32_t* __restrict__ j_rows,
-    const int32_t* __restrict__ j_query, const int32_t* __restrict__ j_list,
-    const int64_t* __restrict__ j_outbase,
-    float* __restrict__ dists_out) {
-    extern __shared__ float lut[]; /* [nsub*256] + red[64] + dconst */
-    float* red = lut + nsub * 256;
-    const int j = blockIdx.x;
-    const int q = j_query[j], l = j_list[j];
-
-    const float* aq = atab + (int64_t)q * nsub * 256;
-    const float* bl = btab + (int64_t)l * nsub * 256;
-    for (int e = threadIdx.x; e < nsub * 256; e += blockDim.x)
-        lut[e] = aq[e] + bl[e];
-
-    /* block dot q . c_l */
-    {
-        const float* qv = queries + (int64_t)q * dpad;
-        const float* cv = cents + (int64_t)l * dpad;
-        float part = 0.f;
-        for (int t = threadIdx.x; t < dpad; t += blockDim.x)
-            part = fmaf(qv[t], cv[t], part);
-        for (int off = 32; off; off >>= 1)
-            part += __shfl_down(part, off, 64);
-        if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = part;
-        __syncthreads();
-        if (threadIdx.x == 0)
-            red[32] = -2.f * (red[0] + red[1] + red[2] + red[3]);
-    }
-    __syncthreads();
-    const float dconst = red[32];
-
+__global__ __launch_bounds__(256) void pq_rowb_kernel(
+    const uint8_t* __restrict__ codes, const float* __restrict__ btab,
+    const int32_t* __restrict__ g2list, const int64_t* __restrict__ g2slot,
+    const int32_t* __restrict__ g2valid, int64_t ngroups, int nsub,
+    float* __restrict__ row_b) {
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-    const int gcount = j_gcount[j];
-    const int rows = j_rows[j];
-    const int64_t baseg = j_databaseg[j];
-    const int64_t ob = j_outbase[j];
-
-    for (int g = wave; g < gcount; g += 4) {
-        const uint8_t* gp = codes + (baseg + g) * (int64_t)nsub * 64;
-        float acc = 0.f;
-        if (W == 16) {
-            const uint4* cptr = (const uint4*)gp + lane;
-            const int m16 = nsub >> 4;
-#pragma unroll 2
-            for (int mq = 0; mq < m16; ++mq) {
-                const uint4 cd = cptr[mq * 64];
-                const uint32_t w[4] = {cd.x, cd.y, cd.z, cd.w};
-#pragma unroll
-                for (int b = 0; b < 4; ++b) {
-                    const int mbase = (16 * mq + 4 * b) * 256;
-                    acc += lut[mbase + (w[b] & 255)];
-                    acc += lut[mbase + 256 + ((w[b] >> 8) & 255)];
-                    acc += lut[mbase + 512 + ((w[b] >> 16) & 255)];
-                    acc += lut[mbase + 768 + (w[b] >> 24)];
-                }
-            }
-        } else {
-            const uchar4* cptr = (const uchar4*)gp + lane;
-            const int mq4 = nsub >> 2;
-#pragma unroll 2
-            for (int mq = 0; mq < mq4; ++mq) {
-                const uchar4 cd = cptr[mq * 64];
-                acc += lut[(4 * mq + 0) * 256 + cd.x];
-                acc += lut[(4 * mq + 1) * 256 + cd.y];
-                acc += lut[(4 * mq + 2) * 256 + cd.z];
-                acc += lut[(4 * mq + 3) * 256 + cd.w];
+    const int64_t g = (int64_t)blockIdx.x * 4 + wave;
+    if (g >= ngroups) return;
+    const float* bl = btab + (int64_t)g2list[g] * nsub * 256;
+    const uint8_t* gp = codes + g * (int64_t)nsub * 64;
+    float acc = 0.f;
+    if (W == 16) {
+        const uint4* cptr = (const uint4*)gp + lane;
+        for (int mq = 0; mq < (nsub >> 4); ++mq) {
+            const uint4 cd = cptr[mq * 64];
+            const uint32_t w[4] = {cd.x, cd.y, cd.z, cd.w};
+            for (int b = 0; b < 4; ++b) {
+                const int mbase = (16 * mq + 4 * b) * 256;
+                acc += bl[mbase + (w[b] & 255)];
+                acc += bl[mbase + 256 + ((w[b] >> 8) & 255)];
+                acc += bl[mbase + 512 + ((w[b] >> 16) & 255)];
+                acc += bl[mbase + 768 + (w[b] >> 24)];
             }
         }
-        const int row = g * 64 + lane;
-        if (row < rows) dists_out[ob + row] = acc + dconst;
+    } else {
+        const uchar4* cptr = (const uchar4*)gp + lane;
+        for (int mq = 0; mq < (nsub >> 2); ++mq) {
+            const uchar4 cd = cptr[mq * 64];
+            acc += bl[(4 * mq + 0) * 256 + cd.x];
+            acc += bl[(4 * mq + 1) * 256 + cd.y];
+            acc += bl[(4 * mq + 2) * 256 + cd.z];
+            acc += bl[(4 * mq + 3) * 256 + cd.w];
+        }
+    }
+    if (lane < g2valid[g]) row_b[g2slot[g] + lane] = acc;
+}
+
+void launch_pq_rowb(bool w16, const uint8_t* codes, const float* btab,
+                    const int32_t* g2list, const int64_t* g2slot,
+                    const int32_t* g2valid, int64_t ngroups, int nsub,
+                    float* row_b, hipStream_t stream) {
+    if (!ngroups) return;
+    const dim3 grid((uint32_t)((ngroups + 3) / 4)), block(256);
+    if (w16)
+        hipLaunchKernelGGL(pq_rowb_kernel<16>, grid, block, 0, stream, codes,
+                           btab, g2list, g2slot, g2valid, ngroups, nsub,
+                           row_b);
+    else
+        hipLaunchKernelGGL(pq_rowb_kernel<4>, grid, block, 0, stream, codes,
+                           btab, g2list, g2slot, g2valid, ngroups, nsub,
+                           row_b);
+}
+
+/* One workgroup per QUERY: A_q LUT once in LDS, then every probed list's
+ * codes streamed (W=16: dwordx4 over the 16-byte-unit layout). */
+template <int W>
+__global__ __launch_bounds__(256) void pq_scan_q_kernel(
+    const uint8_t* __restrict__ codes, const float* __restrict__ atab,
+    const float* __restrict__ row_b, const float* __restrict__ qnorms,
+    const float* __restrict__ cnorm2,
+    const int32_t* __restrict__ csel_slots,
+    const float* __restrict__ csel_dists, int probe,
+    const int64_t* __restrict__ qoffs, const int64_t* __restrict__ probe_offs,
+    const int64_t* __restrict__ list_gbase,
+    const int32_t* __restrict__ list_rows,
+    const int64_t* __restrict__ list_slot_base,
+    int nsub, float* __restrict__ dists_out) {
+    extern __shared__ float lut[]; /* [nsub*256] */
+    const int q = blockIdx.x;
+    const float* aq = atab + (int64_t)q * nsub * 256;
+    for (int e = threadIdx.x; e < nsub * 256; e += blockDim.x)
+        lut[e] = aq[e];
+    __syncthreads();
+    const float qn2 = qnorms[q];
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+
+    for (int r = 0; r < probe; ++r) {
+        const int l = csel_slots[q * probe + r];
+        if (l < 0) continue;
+        const int rows = list_rows[l];
+        if (!rows) continue;
+        const float dconst = csel_dists[q * probe + r] - qn2 - cnorm2[l];
+        const int64_t ob = qoffs[q] + probe_offs[q * (probe + 1) + r];
+        const int64_t baseg = list_gbase[l];
+        const int64_t sbase = list_slot_base[l];
+        const int lg = (rows + 63) >> 6;
+        for (int g = wave; g < lg; g += 4) {
+            const uint8_t* gp = codes + (baseg + g) * (int64_t)nsub * 64;
+            float acc = 0.f;
+            if (W == 16) {
+                const uint4* cptr = (const uint4*)gp + lane;
+#pragma unroll 2
+                for (int mq = 0; mq < (nsub >> 4); ++mq) {
+                    const uint4 cd = cptr[mq * 64];
+                    const uint32_t w[4] = {cd.x, cd.y, cd.z, cd.w};
+#pragma unroll
+                    for (int b = 0; b < 4; ++b) {
+                        const int mbase = (16 * mq + 4 * b) * 256;
+                        acc += lut[mbase + (w[b] & 255)];
+                        acc += lut[mbase + 256 + ((w[b] >> 8) & 255)];
+                        acc += lut[mbase + 512 + ((w[b] >> 16) & 255)];
+                        acc += lut[mbase + 768 + (w[b] >> 24)];
+                    }
+                }
+            } else {
+                const uchar4* cptr = (const uchar4*)gp + lane;
+#pragma unroll 2
+                for (int mq = 0; mq < (nsub >> 2); ++mq) {
+                    const uchar4 cd = cptr[mq * 64];
+                    acc += lut[(4 * mq + 0) * 256 + cd.x];
+                    acc += lut[(4 * mq + 1) * 256 + cd.y];
+                    acc += lut[(4 * mq + 2) * 256 + cd.z];
+                    acc += lut[(4 * mq + 3) * 256 + cd.w];
+                }
+            }
+            const int row = g * 64 + lane;
+            if (row < rows)
+                dists_out[ob + row] = acc + row_b[sbase + row] + dconst;
+        }
     }
 }
 
-void launch_pq_tables(const float* cents, const float* queries,
-                      const float* cb, int nsub, int sd, int dpad,
-                      int64_t nlist_or_nq, bool btab_mode, float* out,
-                      hipStream_t stream) {
-    const int64_t total = nlist_or_nq * nsub * 256;
-    if (!total) return;
-    const int64_t grid = (total + 255) / 256;
-    if (btab_mode)
-        hipLaunchKernelGGL(pq_btab_kernel, dim3((uint32_t)grid), dim3(256), 0,
-                           stream, cents, cb, nsub, sd, dpad, nlist_or_nq,
-                           out);
-    else
-        hipLaunchKernelGGL(pq_atab_kernel, dim3((uint32_t)grid), dim3(256), 0,
-                           stream, queries, cb, nsub, sd, dpad, nlist_or_nq,
-                           out);
-}
-
-void launch_pq_scan(const uint8_t* codes, const float* queries,
-                    const float* cents, const float* atab, const float* btab,
-                    int nsub, int dpad, int njobs,
-                    const int64_t* j_databaseg,
-                    const int32_t* j_gcount, const int32_t* j_rows,
-                    const int32_t* j_query, const int32_t* j_list,
-                    const int64_t* j_outbase, float* dists_out,
-                    hipStream_t stream) {
-    if (!njobs) return;
-    const size_t shmem = ((size_t)nsub * 256 + 40) * 4;
+void launch_pq_scan_q(const uint8_t* codes, const float* atab,
+                      const float* row_b, const float* qnorms,
+                      const float* cnorm2, const int32_t* csel_slots,
+                      const float* csel_dists, int probe,
+                      const int64_t* qoffs, const int64_t* probe_offs,
+                      const int64_t* list_gbase, const int32_t* list_rows,
+                      const int64_t* list_slot_base, int nsub, int nq,
+                      float* dists_out, hipStream_t stream) {
+    if (!nq) return;
+    const size_t shmem = (size_t)nsub * 256 * 4;
     if ((nsub & 15) == 0)
-        hipLaunchKernelGGL(pq_scan_kernel<16>, dim3(njobs), dim3(256), shmem,
-                           stream, codes, queries, cents, atab, btab, nsub,
-                           dpad, j_databaseg, j_gcount, j_rows, j_query,
-                           j_list, j_outbase, dists_out);
+        hipLaunchKernelGGL(pq_scan_q_kernel<16>, dim3(nq), dim3(256), shmem,
+                           stream, codes, atab, row_b, qnorms, cnorm2,
+                           csel_slots, csel_dists, probe, qoffs, probe_offs,
+                           list_gbase, list_rows, list_slot_base, nsub,
+                           dists_out);
     else
-        hipLaunchKernelGGL(pq_scan_kernel<4>, dim3(njobs), dim3(256), shmem,
-                           stream, codes, queries, cents, atab, btab, nsub,
-                           dpad, j_databaseg, j_gcount, j_rows, j_query,
-                           j_list, j_outbase, dists_out);
+        hipLaunchKernelGGL(pq_scan_q_kernel<4>, dim3(nq), dim3(256), shmem,
+                           stream, codes, atab, row_b, qnorms, cnorm2,
+                           csel_slots, csel_dists, probe, qoffs, probe_offs,
+                           list_gbase, list_rows, list_slot_base, nsub,
+                           dists_out);
 }
 
 /* pack codes: thread per output uchar4 */
