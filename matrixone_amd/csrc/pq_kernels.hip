@@ -70,6 +70,23 @@ __global__ __launch_bounds__(256) void pq_atab_kernel(
     atab[idx] = a;
 }
 
+void launch_pq_tables(const float* cents, const float* queries,
+                      const float* cb, int nsub, int sd, int dpad,
+                      int64_t nlist_or_nq, bool btab_mode, float* out,
+                      hipStream_t stream) {
+    const int64_t total = nlist_or_nq * nsub * 256;
+    if (!total) return;
+    const int64_t grid = (total + 255) / 256;
+    if (btab_mode)
+        hipLaunchKernelGGL(pq_btab_kernel, dim3((uint32_t)grid), dim3(256), 0,
+                           stream, cents, cb, nsub, sd, dpad, nlist_or_nq,
+                           out);
+    else
+        hipLaunchKernelGGL(pq_atab_kernel, dim3((uint32_t)grid), dim3(256), 0,
+                           stream, queries, cb, nsub, sd, dpad, nlist_or_nq,
+                           out);
+}
+
 /* Per-row ADC constant, folded at BUILD: row_b[slot] = sum_m B_l[m][code_m]
  * — query-independent, so the search LUT is A_q ONLY and one workgroup
  * serves a whole QUERY (every probed list) with a single LDS table.
