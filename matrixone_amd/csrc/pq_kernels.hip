@@ -150,7 +150,7 @@ void launch_pq_rowb(bool w16, const uint8_t* codes, const float* btab,
 /* One workgroup per QUERY: A_q LUT once in LDS, then every probed list's
  * codes streamed (W=16: dwordx4 over the 16-byte-unit layout). */
 template <int W>
-__global__ __launch_bounds__(256) void pq_scan_q_kernel(
+__global__ __launch_bounds__(1024) void pq_scan_q_kernel(
     const uint8_t* __restrict__ codes, const float* __restrict__ atab,
     const float* __restrict__ row_b, const float* __restrict__ qnorms,
     const float* __restrict__ cnorm2,
@@ -168,7 +168,10 @@ __global__ __launch_bounds__(256) void pq_scan_q_kernel(
         lut[e] = aq[e];
     __syncthreads();
     const float qn2 = qnorms[q];
+    /* 1024 threads = 16 waves: the LUT caps residency at one workgroup
+     * per CU, so the waves come from block WIDTH, not block count */
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int nwave = blockDim.x >> 6;
 
     for (int r = 0; r < probe; ++r) {
         const int l = csel_slots[q * probe + r];
@@ -180,7 +183,7 @@ __global__ __launch_bounds__(256) void pq_scan_q_kernel(
         const int64_t baseg = list_gbase[l];
         const int64_t sbase = list_slot_base[l];
         const int lg = (rows + 63) >> 6;
-        for (int g = wave; g < lg; g += 4) {
+        for (int g = wave; g < lg; g += nwave) {
             const uint8_t* gp = codes + (baseg + g) * (int64_t)nsub * 64;
             float acc = 0.f;
             if (W == 16) {
@@ -227,13 +230,13 @@ void launch_pq_scan_q(const uint8_t* codes, const float* atab,
     if (!nq) return;
     const size_t shmem = (size_t)nsub * 256 * 4;
     if ((nsub & 15) == 0)
-        hipLaunchKernelGGL(pq_scan_q_kernel<16>, dim3(nq), dim3(256), shmem,
+        hipLaunchKernelGGL(pq_scan_q_kernel<16>, dim3(nq), dim3(1024), shmem,
                            stream, codes, atab, row_b, qnorms, cnorm2,
                            csel_slots, csel_dists, probe, qoffs, probe_offs,
                            list_gbase, list_rows, list_slot_base, nsub,
                            dists_out);
     else
-        hipLaunchKernelGGL(pq_scan_q_kernel<4>, dim3(nq), dim3(256), shmem,
+        hipLaunchKernelGGL(pq_scan_q_kernel<4>, dim3(nq), dim3(1024), shmem,
                            stream, codes, atab, row_b, qnorms, cnorm2,
                            csel_slots, csel_dists, probe, qoffs, probe_offs,
                            list_gbase, list_rows, list_slot_base, nsub,
