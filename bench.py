@@ -180,6 +180,10 @@ def main():
     ap.add_argument("--recall-queries", type=int, default=256)
     ap.add_argument("--cpu-baseline-queries", type=int, default=48)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--refine", type=int, default=0,
+                    help="two-stage scan: byte first pass + exact f32 "
+                         "re-rank of the top-R (0 = exact one-pass scan). "
+                         "recall@10 is still gated at 0.95 either way")
     ap.add_argument("--qtype", default="f32", choices=["f32", "int8", "uint8"],
                     help="storage type; the BASELINE metric is f32 — narrow "
                          "runs are secondary datapoints and report their own "
@@ -238,6 +242,8 @@ def main():
         hi = float(torch.quantile(sample, 0.999))
         ix.set_quantizer(lo, hi)
     ix.build()
+    if args.refine and args.qtype == "f32":
+        ix.enable_refine(args.refine)
     log(f"[bench] build done in {time.perf_counter() - t_build0:.1f}s")
 
     # ---- recall check ------------------------------------------------------

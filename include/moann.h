@@ -159,6 +159,15 @@ gpu_ivf_flat_search_res_t gpu_ivf_flat_search(gpu_ivf_flat_c index_c,
                                               ivf_flat_search_params_t search_params,
                                               void* errmsg);
 
+/* moann extension — two-stage scan: byte-quantized first pass over an
+ * INTERNAL clip-quantized image (0.1%/99.9% sampled quantiles; independent
+ * of the storage quantizer) + exact f32 re-rank of the per-query top-R
+ * (SQ8-with-refine). Cuts first-pass HBM traffic 4x at a recall cost
+ * bounded by R; opt-in — depth 0 restores the exact one-pass scan.
+ * f32 storage only; depth <= 4096. */
+void moann_ivf_flat_enable_refine(gpu_ivf_flat_c index_c, uint32_t depth,
+                                  void* errmsg);
+
 /* moann extension — pipelined batch submit/wait. Submit runs the centroid
  * rank and the host job build immediately and enqueues the list scan and
  * result readback without waiting; a second submit overlaps its rank/build

@@ -118,6 +118,16 @@ void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
                          const int32_t* group_valid, int64_t ngroups,
                          int64_t slot_lo, int64_t slot_hi, uint8_t* out,
                          hipStream_t stream);
+/* exact f32 re-rank of byte-stage top-R (two-stage scan; SQ8 + refine) */
+void launch_refine(int metric, const float* rows_f32, const float* queries,
+                   const float* qnorms, int dim, int dpad, int R, int nq,
+                   const int32_t* rsel_slots, const float* rsel_dists,
+                   const int32_t* probe_lists, const int64_t* probe_offs,
+                   const int64_t* list_slot_base, int nprobe, float* refined,
+                   hipStream_t stream);
+void launch_compose_select(const int32_t* sel2, const int32_t* rsel_slots,
+                           int R, int k, int nq, int32_t* out,
+                           hipStream_t stream);
 void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
                           int in_stride, int dim, int dpad, float fmul,
                           float fadd, uint8_t* out, hipStream_t stream);

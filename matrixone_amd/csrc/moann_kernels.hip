@@ -733,6 +733,125 @@ void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
                            slot_hi, out);
 }
 
+/* ---------------- exact f32 refine (two-stage scan) ---------------------
+ * Re-rank the byte-stage top-R candidates with exact f32 distances over a
+ * row-major copy of the entries (SQ8-with-refine, the standard recall
+ * recovery; the byte stage reads 1/4 the bytes of the f32 scan). One
+ * workgroup per query: query row cached in LDS, one WAVE per candidate,
+ * lane-parallel over dims, shfl reduce. Distance semantics identical to
+ * the f32 scan kernel (MO conventions incl. cosine's double denom). */
+template <int METRIC>
+__global__ __launch_bounds__(256) void refine_kernel(
+    const float* __restrict__ rows_f32, /* [count][dim] slot-major */
+    const float* __restrict__ queries,  /* [nq][dpad] */
+    const float* __restrict__ qnorms,   /* [nq] |q|^2 (cos) */
+    int dim, int dpad, int R,
+    const int32_t* __restrict__ rsel_slots, /* [nq][R] candidate positions */
+    const float* __restrict__ rsel_dists,   /* [nq][R] byte dists (padding) */
+    const int32_t* __restrict__ probe_lists,
+    const int64_t* __restrict__ probe_offs,
+    const int64_t* __restrict__ list_slot_base, int nprobe,
+    float* __restrict__ refined /* [nq][R] */) {
+    extern __shared__ float ldsq[]; /* [dpad] */
+    const int q = blockIdx.x;
+    const float* qv = queries + (int64_t)q * dpad;
+    for (int e = threadIdx.x; e < dpad; e += blockDim.x) ldsq[e] = qv[e];
+    __syncthreads();
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int64_t* offs = probe_offs + (int64_t)q * (nprobe + 1);
+
+    for (int c = wave; c < R; c += 4) {
+        const int64_t idx = (int64_t)q * R + c;
+        const int32_t pos = rsel_slots[idx];
+        if (pos < 0 || rsel_dists[idx] == FLT_MAX) {
+            if (lane == 0) refined[idx] = FLT_MAX;
+            continue;
+        }
+        /* candidate position -> (probed list, local row) -> global slot */
+        int lo = 0, hi = nprobe - 1;
+        while (lo < hi) {
+            const int mid = (lo + hi + 1) >> 1;
+            if (offs[mid] <= pos) lo = mid; else hi = mid - 1;
+        }
+        const int32_t list = probe_lists[(int64_t)q * nprobe + lo];
+        const int64_t slot = list_slot_base[list] + (pos - offs[lo]);
+        const float* row = rows_f32 + slot * (int64_t)dim;
+        float acc = 0.f, rn = 0.f;
+        for (int t = lane; t < dim; t += 64) {
+            const float x = row[t], qq = ldsq[t];
+            if (METRIC == KM_L2SQ) {
+                const float e = x - qq;
+                acc = fmaf(e, e, acc);
+            } else if (METRIC == KM_IP || METRIC == KM_COS) {
+                acc = fmaf(x, qq, acc);
+                if (METRIC == KM_COS) rn = fmaf(x, x, rn);
+            } else {
+                acc += fabsf(x - qq);
+            }
+        }
+        for (int off = 32; off; off >>= 1) {
+            acc += __shfl_down(acc, off, 64);
+            if (METRIC == KM_COS) rn += __shfl_down(rn, off, 64);
+        }
+        if (lane == 0) {
+            float dist;
+            if (METRIC == KM_IP) dist = -acc;
+            else if (METRIC == KM_COS) {
+                const double denom = sqrt((double)rn) * sqrt((double)qnorms[q]);
+                if (denom == 0.0) dist = 1.0f;
+                else {
+                    double sim = (double)acc / denom;
+                    sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                    dist = (float)(1.0 - sim);
+                }
+            } else dist = acc;
+            refined[idx] = dist;
+        }
+    }
+}
+
+void launch_refine(int metric, const float* rows_f32, const float* queries,
+                   const float* qnorms, int dim, int dpad, int R, int nq,
+                   const int32_t* rsel_slots, const float* rsel_dists,
+                   const int32_t* probe_lists, const int64_t* probe_offs,
+                   const int64_t* list_slot_base, int nprobe, float* refined,
+                   hipStream_t stream) {
+    if (!nq) return;
+    const size_t shmem = (size_t)dpad * 4;
+#define RARGS rows_f32, queries, qnorms, dim, dpad, R, rsel_slots, \
+    rsel_dists, probe_lists, probe_offs, list_slot_base, nprobe, refined
+    switch (metric) {
+    case KM_L2SQ: hipLaunchKernelGGL(refine_kernel<KM_L2SQ>, dim3(nq), dim3(256), shmem, stream, RARGS); break;
+    case KM_IP:   hipLaunchKernelGGL(refine_kernel<KM_IP>, dim3(nq), dim3(256), shmem, stream, RARGS); break;
+    case KM_COS:  hipLaunchKernelGGL(refine_kernel<KM_COS>, dim3(nq), dim3(256), shmem, stream, RARGS); break;
+    default:      hipLaunchKernelGGL(refine_kernel<KM_L1>, dim3(nq), dim3(256), shmem, stream, RARGS); break;
+    }
+#undef RARGS
+}
+
+/* final-selection composition: sel2 indexes into the refine set; map back
+ * to candidate positions so the shared gather kernel applies unchanged */
+__global__ void compose_select_kernel(const int32_t* __restrict__ sel2,
+                                      const int32_t* __restrict__ rsel_slots,
+                                      int R, int k, int64_t total,
+                                      int32_t* __restrict__ out) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total) return;
+    const int32_t s2 = sel2[i];
+    const int64_t q = i / k;
+    out[i] = s2 < 0 ? -1 : rsel_slots[q * R + s2];
+}
+
+void launch_compose_select(const int32_t* sel2, const int32_t* rsel_slots,
+                           int R, int k, int nq, int32_t* out,
+                           hipStream_t stream) {
+    const int64_t total = (int64_t)nq * k;
+    if (!total) return;
+    hipLaunchKernelGGL(compose_select_kernel,
+                       dim3((uint32_t)((total + 255) / 256)), dim3(256), 0,
+                       stream, sel2, rsel_slots, R, k, total, out);
+}
+
 /* -------------------------- quantized (int8/uint8) scan ------------------
  * Narrow-storage variant of the list scan (reference: int8/uint8 entries
  * quantized by q(x)=round(x*mul+add), distance computed IN THE QUANTIZED

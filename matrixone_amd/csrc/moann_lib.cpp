@@ -125,6 +125,7 @@ struct SearchCtx {
     DevBuf w_jobs, w_qslots, w_cand, w_sel_slots, w_sel_dists;
     DevBuf w_probe_lists, w_probe_offs, w_qoffs, w_out_ids, w_out_dists;
     DevBuf w_filter, w_queries_q, w_qnorms_i;
+    DevBuf w_rsel_slots, w_rsel_dists, w_refined, w_sel2; /* refine stage */
     void* h_out = nullptr;   /* pinned [ids int64][dists f32] bounce */
     size_t h_out_bytes = 0;
     MoannResult* pending = nullptr; /* submitted, not yet collected */
@@ -165,6 +166,13 @@ struct IvfIndex {
     double quant_min = 0.0, quant_max = 0.0;
     uint8_t* d_packed_q = nullptr;  /* quantized interleaved entries */
     int32_t* d_rownorm_q = nullptr; /* [total_groups*64] byte-row |x|^2 */
+    /* two-stage scan (moann_ivf_flat_enable_refine): byte first pass over
+     * an INTERNAL clip-quantized image, exact f32 re-rank of the top-R */
+    int refine_depth = 0;
+    float* d_rows_f32 = nullptr;    /* [count][dim] slot-major (refine) */
+    uint8_t* d_packed_rq = nullptr; /* clip-quantized interleaved image */
+    int32_t* d_rownorm_rq = nullptr;
+    double rq_mul = 1.0, rq_add = 0.0;
     bool built = false, started = false;
 
     hipStream_t stream = nullptr;
@@ -233,7 +241,8 @@ struct IvfIndex {
         (void)hipSetDevice(device);
         for (auto p : {(void*)d_staging, (void*)d_packed, (void*)d_cent_packed,
                        (void*)d_cent_rows, (void*)d_cnorms, (void*)d_packed_q,
-                       (void*)d_rownorm_q,
+                       (void*)d_rownorm_q, (void*)d_rows_f32,
+                       (void*)d_packed_rq, (void*)d_rownorm_rq,
                        (void*)d_id_by_slot, (void*)d_list_slot_base})
             if (p) (void)hipFree(p);
         if (stream) (void)hipStreamDestroy(stream);
@@ -565,6 +574,8 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     /* 4. list scan into the ragged candidate buffer */
     const bool narrow = ix->qtype != Quantization_F32;
     const bool uns = ix->qtype == Quantization_UINT8;
+    const bool use_refine =
+        !narrow && ix->refine_depth > 0 && ix->d_packed_rq != nullptr;
     const uint32_t* d_filter = nullptr;
     if (filter_words) {
         const size_t fw = (size_t)((ix->count + 31) / 32);
@@ -600,7 +611,26 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         jb.slot_base = (int64_t*)up(jslot_base.data(), njobs * b64);
         jb.njobs = njobs;
         HIP_CHECK(hipEventRecord(cx->ev[2], s));
-        if (narrow) {
+        if (use_refine) {
+            /* byte first pass over the internal clip-quantized image */
+            cx->w_queries_q.ensure((size_t)nq * dpad);
+            launch_quantize_rows(false, d_q, (int64_t)nq, dpad, (int)ix->dim,
+                                 dpad, (float)ix->rq_mul, (float)ix->rq_add,
+                                 cx->w_queries_q.as<uint8_t>(), s);
+            KCHECK("refine-query-quantize");
+            const int32_t* d_qni = nullptr;
+            if (ix->kmetric == KM_COS || ix->kmetric == KM_L2SQ) {
+                cx->w_qnorms_i.ensure(nq * 4);
+                launch_qnorms_i8(false, cx->w_queries_q.as<uint8_t>(),
+                                 (int)nq, dpad,
+                                 cx->w_qnorms_i.as<int32_t>(), s);
+                d_qni = cx->w_qnorms_i.as<int32_t>();
+            }
+            launch_scan_i8(ix->kmetric, false, ix->d_packed_rq,
+                           cx->w_queries_q.as<uint8_t>(), d_qni,
+                           ix->d_rownorm_rq, dpad, jb,
+                           cx->w_cand.as<float>(), s, d_filter);
+        } else if (narrow) {
             /* quantize the padded f32 queries with the entry map (the
              * search_quantize contract: query narrowed the same way as the
              * entries — quantizer.go ApplyInt8/entry SQL) */
@@ -634,20 +664,48 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     cx->w_qoffs.ensure((nq + 1) * 8);
     HIP_CHECK(hipMemcpyAsync(cx->w_qoffs.ptr, qoffs.data(), (nq + 1) * 8,
                              hipMemcpyHostToDevice, s));
-    cx->w_sel_slots.ensure((size_t)nq * limit * 4);
-    cx->w_sel_dists.ensure((size_t)nq * limit * 4);
-    HIP_CHECK(hipEventRecord(cx->ev[4], s));
-    launch_topk(cx->w_cand.as<float>(), cx->w_qoffs.as<int64_t>(), 0, (int)nq,
-                (int)limit, cx->w_sel_slots.as<int32_t>(),
-                cx->w_sel_dists.as<float>(), s);
-    KCHECK("cand-topk");
-    HIP_CHECK(hipEventRecord(cx->ev[5], s));
     cx->w_probe_lists.ensure((size_t)nq * probe * 4);
     cx->w_probe_offs.ensure(probe_offs.size() * 8);
     HIP_CHECK(hipMemcpyAsync(cx->w_probe_lists.ptr, h_probe.data(),
                              h_probe.size() * 4, hipMemcpyHostToDevice, s));
     HIP_CHECK(hipMemcpyAsync(cx->w_probe_offs.ptr, probe_offs.data(),
                              probe_offs.size() * 8, hipMemcpyHostToDevice, s));
+    cx->w_sel_slots.ensure((size_t)nq * limit * 4);
+    cx->w_sel_dists.ensure((size_t)nq * limit * 4);
+    HIP_CHECK(hipEventRecord(cx->ev[4], s));
+    if (use_refine) {
+        /* top-R by byte distance -> exact f32 re-rank -> top-limit */
+        const int R = std::max<int>((int)limit, ix->refine_depth);
+        cx->w_rsel_slots.ensure((size_t)nq * R * 4);
+        cx->w_rsel_dists.ensure((size_t)nq * R * 4);
+        launch_topk(cx->w_cand.as<float>(), cx->w_qoffs.as<int64_t>(), 0,
+                    (int)nq, R, cx->w_rsel_slots.as<int32_t>(),
+                    cx->w_rsel_dists.as<float>(), s);
+        KCHECK("refine-topR");
+        cx->w_refined.ensure((size_t)nq * R * 4);
+        launch_refine(ix->kmetric, ix->d_rows_f32, d_q, d_qn, (int)ix->dim,
+                      dpad, R, (int)nq, cx->w_rsel_slots.as<int32_t>(),
+                      cx->w_rsel_dists.as<float>(),
+                      cx->w_probe_lists.as<int32_t>(),
+                      cx->w_probe_offs.as<int64_t>(), ix->d_list_slot_base,
+                      (int)probe, cx->w_refined.as<float>(), s);
+        KCHECK("refine");
+        cx->w_sel2.ensure((size_t)nq * limit * 4);
+        launch_topk(cx->w_refined.as<float>(), nullptr, R, (int)nq,
+                    (int)limit, cx->w_sel2.as<int32_t>(),
+                    cx->w_sel_dists.as<float>(), s);
+        KCHECK("refine-topk");
+        launch_compose_select(cx->w_sel2.as<int32_t>(),
+                              cx->w_rsel_slots.as<int32_t>(), R, (int)limit,
+                              (int)nq, cx->w_sel_slots.as<int32_t>(), s);
+        KCHECK("refine-compose");
+    } else {
+        launch_topk(cx->w_cand.as<float>(), cx->w_qoffs.as<int64_t>(), 0,
+                    (int)nq, (int)limit, cx->w_sel_slots.as<int32_t>(),
+                    cx->w_sel_dists.as<float>(), s);
+        KCHECK("cand-topk");
+    }
+    HIP_CHECK(hipEventRecord(cx->ev[5], s));
     cx->w_out_ids.ensure((size_t)nq * limit * 8);
     cx->w_out_dists.ensure((size_t)nq * limit * 4);
     launch_gather(cx->w_sel_slots.as<int32_t>(), cx->w_sel_dists.as<float>(),
@@ -1595,6 +1653,96 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
  * mul=255/rng, add=-min*mul-128; :232-240 Uint8Params without the -128
  * shift). Call before build; search rescales raw scores by 1/mul^2
  * (search.go:1062-1077). */
+/* Two-stage scan: byte-quantized first pass (internal CLIP quantizer over
+ * sampled 0.1%/99.9% quantiles — independent of the MO storage quantizer)
+ * + exact f32 re-rank of the per-query top-R. Standard SQ8-with-refine;
+ * cuts first-pass HBM traffic 4x. f32 storage only; opt-in (the default
+ * path stays the exact one-pass scan). */
+void moann_ivf_flat_enable_refine(gpu_ivf_flat_c h, uint32_t depth,
+                                  void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (!ix->built) throw std::runtime_error("index not built");
+        if (ix->qtype != Quantization_F32)
+            throw std::runtime_error("refine applies to f32 storage");
+        if (depth == 0) { ix->refine_depth = 0; return; }
+        if (depth > 4096) throw std::runtime_error("refine depth > 4096");
+        HIP_CHECK(hipSetDevice(ix->device));
+        if (!ix->d_rows_f32) {
+            DevBuf d_gsb, d_gv;
+            d_gsb.ensure(ix->total_groups * 8);
+            d_gv.ensure(ix->total_groups * 4);
+            HIP_CHECK(hipMemcpy(d_gsb.ptr, ix->group_slotbase.data(),
+                                ix->total_groups * 8, hipMemcpyHostToDevice));
+            HIP_CHECK(hipMemcpy(d_gv.ptr, ix->group_valid.data(),
+                                ix->total_groups * 4, hipMemcpyHostToDevice));
+            /* row-major f32 copy for the exact re-rank */
+            HIP_CHECK(hipMalloc(&ix->d_rows_f32,
+                                std::max<uint64_t>(1, ix->count) *
+                                    ix->dim * 4));
+            launch_unpack(ix->d_packed, (int)ix->dim, (int)ix->dpad,
+                          d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
+                          ix->total_groups, 0, (int64_t)ix->count,
+                          ix->d_rows_f32, ix->stream);
+            HIP_CHECK(hipStreamSynchronize(ix->stream));
+            KCHECK("refine-unpack");
+            /* clip quantizer from sampled quantiles */
+            const uint64_t ns = std::min<uint64_t>(ix->count, 131072);
+            std::vector<float> samp((size_t)ns * ix->dim);
+            HIP_CHECK(hipMemcpy(samp.data(), ix->d_rows_f32, samp.size() * 4,
+                                hipMemcpyDeviceToHost));
+            const size_t nlo = samp.size() / 1000;
+            std::nth_element(samp.begin(), samp.begin() + nlo, samp.end());
+            const double lo = samp[nlo];
+            std::nth_element(samp.begin(), samp.end() - 1 - nlo, samp.end());
+            const double hi = samp[samp.size() - 1 - nlo];
+            const double rng = hi - lo;
+            ix->rq_mul = rng > 0 ? 255.0 / rng : 1.0;
+            ix->rq_add = rng > 0 ? -lo * ix->rq_mul - 128.0 : 0.0;
+            /* quantize rows -> pack (16-byte units when possible) */
+            DevBuf d_qrows, d_ident;
+            d_qrows.ensure((size_t)std::max<uint64_t>(1, ix->count) *
+                           ix->dpad);
+            launch_quantize_rows(false, ix->d_rows_f32, (int64_t)ix->count,
+                                 (int)ix->dim, (int)ix->dim, (int)ix->dpad,
+                                 (float)ix->rq_mul, (float)ix->rq_add,
+                                 d_qrows.as<uint8_t>(), ix->stream);
+            KCHECK("refine-quantize");
+            std::vector<int64_t> ident(std::max<uint64_t>(1, ix->count));
+            for (uint64_t i = 0; i < ix->count; ++i) ident[i] = (int64_t)i;
+            d_ident.ensure(ident.size() * 8);
+            HIP_CHECK(hipMemcpy(d_ident.ptr, ident.data(), ident.size() * 8,
+                                hipMemcpyHostToDevice));
+            HIP_CHECK(hipMalloc(&ix->d_packed_rq,
+                                (size_t)ix->total_groups * 64 * ix->dpad));
+            if ((ix->dpad & 15) == 0)
+                launch_bytes_pack16(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                                    d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
+                                    d_ident.as<int64_t>(), ix->total_groups,
+                                    ix->d_packed_rq, ix->stream);
+            else
+                launch_pq_pack(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                               d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
+                               d_ident.as<int64_t>(), ix->total_groups,
+                               ix->d_packed_rq, ix->stream);
+            KCHECK("refine-pack");
+            if ((ix->dpad & 15) == 0) {
+                HIP_CHECK(hipMalloc(&ix->d_rownorm_rq,
+                                    (size_t)ix->total_groups * 64 * 4));
+                launch_rownorms_i8(false, ix->d_packed_rq, ix->total_groups,
+                                   (int)ix->dpad, ix->d_rownorm_rq,
+                                   ix->stream);
+            }
+            HIP_CHECK(hipStreamSynchronize(ix->stream));
+            KCHECK("refine-build");
+        }
+        ix->refine_depth = (int)depth;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_enable_refine", e.what());
+    }
+}
+
 void gpu_ivf_flat_set_quantizer(gpu_ivf_flat_c h, float min, float max,
                                 void* errmsg) {
     try {

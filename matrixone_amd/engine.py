@@ -114,6 +114,8 @@ def lib() -> ct.CDLL:
         ct.c_uint32, SearchParams, ct.c_void_p]
     L.moann_ivf_flat_result_wait.restype = ct.c_int
     L.moann_ivf_flat_result_wait.argtypes = [ct.c_void_p, ct.c_void_p]
+    L.moann_ivf_flat_enable_refine.argtypes = [ct.c_void_p, ct.c_uint32,
+                                               ct.c_void_p]
     L.moann_docfilter_eval.restype = ct.c_int
     L.moann_docfilter_eval.argtypes = [
         ct.c_int, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_uint64,
@@ -380,6 +382,13 @@ class IvfFlatIndex:
         class _R:
             result_ptr = ptr
         return self._read_result(_R, nq, k, "search_collect")
+
+    def enable_refine(self, depth: int):
+        """Two-stage scan (moann_ivf_flat_enable_refine): byte first pass +
+        exact f32 re-rank of the top-`depth`. 0 disables."""
+        err = _Err()
+        lib().moann_ivf_flat_enable_refine(self._h, depth, err.ref)
+        err.check("enable_refine")
 
     def docfilter_bitset(self, kind: int, payload: bytes) -> np.ndarray:
         """Slot bitset from a serialized docfilter: membership of every
