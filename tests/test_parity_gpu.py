@@ -542,3 +542,51 @@ def test_save_load_dir_quantized(tmp_path):
         ix3.load_dir(str(tmp_path))
     ix2.close()
     ix3.close()
+
+
+def test_refine_exactness():
+    """Two-stage scan (enable_refine): with R >= every candidate count the
+    result must EQUAL the exact one-pass search (every candidate re-ranked
+    with exact f32), and at small R the returned distances are still the
+    EXACT f32 distances of the returned ids."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(21))
+    n, d, nlist, nq, probe, k = 20000, 768, 64, 16, 4, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = np.empty(n, dtype=np.int64)
+    for c0 in range(0, n, 4096):
+        blk = vecs[c0:c0 + 4096].astype(np.float64)
+        d2 = ((blk[:, None, :] - cents[None].astype(np.float64)) ** 2).sum(-1)
+        assign[c0:c0 + 4096] = d2.argmin(1)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    exact_ids, exact_d = ix.search(queries, k, probe)
+
+    # R = 4096 > max candidates per query (~20000*4/64 ~ 1250): identical
+    ix.enable_refine(4096)
+    rid, rd = ix.search(queries, k, probe)
+    np.testing.assert_array_equal(rid, exact_ids)
+    np.testing.assert_allclose(rd, exact_d, rtol=2e-5, atol=2e-5)
+
+    # small R: whatever ids come back carry their EXACT f32 distances
+    ix.enable_refine(32)
+    sid, sd = ix.search(queries, k, probe)
+    id2d = {}
+    big_ids, big_d = ix.search(queries, 1200, probe) if False else (None, None)
+    ix.enable_refine(0)  # exact reference for distance lookup
+    ref_all, ref_alld = ix.search(queries, 1200, probe)
+    for i in range(nq):
+        lut = {int(a): float(b) for a, b in zip(ref_all[i], ref_alld[i])
+               if a >= 0}
+        for j in range(k):
+            if sid[i, j] < 0:
+                continue
+            assert int(sid[i, j]) in lut
+            np.testing.assert_allclose(sd[i, j], lut[int(sid[i, j])],
+                                       rtol=2e-5, atol=2e-5)
+    ix.close()
