@@ -88,7 +88,7 @@ __device__ float wave_dist(const float* __restrict__ vecs, int64_t slot,
 }
 
 template <int METRIC, bool FILTERED>
-__global__ __launch_bounds__(256) void hnsw_search_kernel(
+__global__ __launch_bounds__(1024) void hnsw_search_kernel(
     const float* __restrict__ vecs, int dpad, int64_t n,
     const int64_t* __restrict__ l0_offs, const uint32_t* __restrict__ l0_nbrs,
     const int32_t* __restrict__ u_idx, int64_t u_count,
@@ -106,18 +106,21 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
     const int q = blockIdx.x;
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
+    const int nwave = (int)blockDim.x >> 6; /* 16 at the 1024-thread launch:
+        neighbor evaluations are the latency bottleneck (random 3 KB row
+        reads), so width = parallel evals per expansion round */
 
     float* ldsq = (float*)smem;                               /* [dpad]    */
     uint64_t* list = (uint64_t*)(smem + ((dpad * 4 + 7) & ~7));/* [efc]    */
     uint64_t* merged = list + efc;                            /* [efc]     */
     uint64_t* newk = merged + efc;                            /* [maxdegP2]*/
     uint32_t* stage = (uint32_t*)(newk + maxdeg_p2);          /* [maxdegP2]*/
-    float* wdist = (float*)(stage + maxdeg_p2);               /* [8]       */
+    float* wdist = (float*)(stage + maxdeg_p2);               /* [16]      */
     /* scalars */
-    int* sc = (int*)(wdist + 8); /* 0:list_sz 1:nbr_idx 2:kept 3:n_new
+    int* sc = (int*)(wdist + 16); /* 0:list_sz 1:nbr_idx 2:kept 3:n_new
                                     4:best_idx 5:evals 6:best_d/radius
                                     7:pass_cnt */
-    int* cnt_scratch = sc + 8;                            /* [256] */
+    int* cnt_scratch = sc + 8;                            /* [1024] */
 
     const float* qsrc = queries + (int64_t)q * dpad;
     for (int e = tid; e < dpad; e += blockDim.x) ldsq[e] = qsrc[e];
@@ -145,7 +148,7 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
             const int m = o1 - o0;
             if (tid == 0) sc[2] = 0;
             __syncthreads();
-            for (int r = 0; r < m; r += 4) {
+            for (int r = 0; r < m; r += nwave) {
                 const int i = r + wave;
                 float d = 0.f;
                 if (i < m)
@@ -154,7 +157,7 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
                 if (i < m && lane == 0) wdist[wave] = d;
                 __syncthreads();
                 if (tid == 0) {
-                    const int lim = min(4, m - r);
+                    const int lim = min(nwave, m - r);
                     for (int j = 0; j < lim; ++j) {
                         if (wdist[j] < *best_d) {
                             *best_d = wdist[j];
@@ -268,10 +271,10 @@ __global__ __launch_bounds__(256) void hnsw_search_kernel(
         const int mp = sc[3];
         if (tid == 0) atomicAdd(&sc[5], mp);
 
-        /* distances: 4 waves round-robin; FILTERED drops candidates that
-         * fail usearch's push condition (top full && d >= radius) and tags
-         * the rest with the membership bit */
-        for (int r = 0; r < mp; r += 4) {
+        /* distances: all waves round-robin; FILTERED drops candidates
+         * that fail usearch's push condition (top full && d >= radius) and
+         * tags the rest with the membership bit */
+        for (int r = 0; r < mp; r += nwave) {
             const int i = r + wave;
             float d = 0.f;
             if (i < mp)
@@ -385,14 +388,14 @@ void launch_hnsw_search(int metric, const float* vecs, int dpad, int64_t n,
     const int efc = filtered ? 2 * ef : ef;
     const size_t shmem = ((dpad * 4 + 7) & ~7) + (size_t)efc * 8 * 2 +
                          (size_t)maxdeg_p2 * 8 + (size_t)maxdeg_p2 * 4 +
-                         8 * 4 + 8 * 4 + 256 * 4 + 16;
+                         16 * 4 + 8 * 4 + 1024 * 4 + 16;
     if (shmem_out) *shmem_out = shmem;
     if (!nq) return;
 #define HARGS vecs, dpad, n, l0_offs, l0_nbrs, u_idx, u_count, up_offs,     \
     up_nbrs, max_level, entry_slot, queries, qnorms, ef, efc, k, maxdeg_p2, \
     slot_filter, visited, vwords, key_of_slot, out_ids, out_dists, eval_count
 #define HLAUNCH(M, F)                                                       \
-    hipLaunchKernelGGL((hnsw_search_kernel<M, F>), dim3(nq), dim3(256),     \
+    hipLaunchKernelGGL((hnsw_search_kernel<M, F>), dim3(nq), dim3(1024),    \
                        shmem, stream, HARGS)
     if (filtered) {
         switch (metric) {
