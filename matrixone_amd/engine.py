@@ -474,6 +474,40 @@ def pairwise_distance(metric: str, query, rows, device: int = 0):
     return out
 
 
+def search_with_tail(index, queries, k: int, nprobe: int,
+                     tail_vecs, tail_ids, metric: str = None):
+    """CDC-tail overflow search (ivfpq/search_gpu.go:222-428 shape): the
+    built index covers the snapshot; rows appended since (the CDC tail,
+    not yet merged into the index) are brute-forced and the two top-k
+    blocks are merged with cpu_topk_merge_sharded semantics (skip -1,
+    dense ascending, (-1, FLT_MAX) padding)."""
+    metric = metric or getattr(index, "metric", "l2sq")
+    main_ids, main_d = index.search(queries, k, nprobe)
+    if tail_vecs is None or len(tail_vecs) == 0:
+        return main_ids, main_d
+    t_ids, t_d = brute_force_search(tail_vecs, queries, min(k, len(tail_vecs)),
+                                    metric=metric, ids=tail_ids)
+    nq = main_ids.shape[0]
+    kk = t_ids.shape[1]
+    if kk < k:  # pad the tail block to k for a uniform merge
+        pad_i = np.full((nq, k - kk), -1, dtype=np.int64)
+        pad_d = np.full((nq, k - kk), np.finfo(np.float32).max,
+                        dtype=np.float32)
+        t_ids = np.concatenate([t_ids, pad_i], 1)
+        t_d = np.concatenate([t_d, pad_d], 1)
+    both_i = np.stack([main_ids, t_ids])
+    both_d = np.stack([main_d.astype(np.float32), t_d.astype(np.float32)])
+    dd = both_d.copy()
+    dd[both_i < 0] = np.finfo(np.float32).max
+    flat_i = both_i.transpose(1, 0, 2).reshape(nq, -1)
+    flat_d = dd.transpose(1, 0, 2).reshape(nq, -1)
+    order = np.argsort(flat_d, axis=1, kind="stable")[:, :k]
+    out_d = np.take_along_axis(flat_d, order, 1)
+    out_i = np.take_along_axis(flat_i, order, 1)
+    out_i[out_d >= np.finfo(np.float32).max] = -1
+    return out_i, out_d
+
+
 def brute_force_search(dataset, queries, k: int, metric: str = "l2sq",
                        ids=None, device: int = 0):
     """Exact GPU top-k (pkg/vectorindex/brute_force counterpart)."""

@@ -590,3 +590,37 @@ def test_refine_exactness():
             np.testing.assert_allclose(sd[i, j], lut[int(sid[i, j])],
                                        rtol=2e-5, atol=2e-5)
     ix.close()
+
+
+def test_search_with_tail():
+    """CDC-tail overflow merge (§8f4, ivfpq/search_gpu.go:222-428 shape):
+    index search + brute force over the unindexed tail, merged == oracle
+    over the UNION corpus at full probe."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(31))
+    n, ntail, d, nlist, nq, k = 6000, 500, 48, 8, 10, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    tail = rng.standard_normal((ntail, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    ids = np.arange(n, dtype=np.int64)
+    tail_ids = np.arange(n, n + ntail, dtype=np.int64)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs, ids=ids)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    got_i, got_d = engine.search_with_tail(ix, queries, k, nlist, tail,
+                                           tail_ids)
+    union = np.concatenate([vecs, tail])
+    union_ids = np.concatenate([ids, tail_ids])
+    exp_i, exp_d = orc.bruteforce(orc.METRIC_L2SQ, union, queries, k,
+                                  ids=union_ids)
+    _assert_parity(got_i, got_d, exp_i, exp_d, ctx="tail-merge")
+    # empty tail degenerates to the plain index search
+    a_i, a_d = engine.search_with_tail(ix, queries, k, nlist, None, None)
+    b_i, b_d = ix.search(queries, k, nlist)
+    np.testing.assert_array_equal(a_i, b_i)
+    ix.close()
