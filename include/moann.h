@@ -56,7 +56,8 @@ typedef enum {
     Quantization_F32 = 0,
     Quantization_F16 = 1,
     Quantization_INT8 = 2,
-    Quantization_UINT8 = 3
+    Quantization_UINT8 = 3,
+    Quantization_BF16 = 4  /* moann extension: MO vecbf16 plain cast */
 } quantization_t;
 
 typedef enum {

@@ -135,6 +135,16 @@ void launch_pq_pack(const uint8_t* codes_rowmajor, int nsub,
                     const int64_t* group_rowbase, const int32_t* group_valid,
                     const int64_t* slot_rows, int64_t ngroups, uint8_t* packed,
                     hipStream_t stream);
+/* half (f16/bf16) storage: plain-cast quantization + decoded-f32 scan */
+void launch_quantize_half_rows(bool bf, const float* in, int64_t nrows,
+                               int in_stride, int dim, int dpad,
+                               uint16_t* out, hipStream_t stream);
+void launch_qnorms_h(bool bf, const uint16_t* q, int nq, int dpad,
+                     float* out, hipStream_t stream);
+void launch_scan_h(int metric, bool bf, const uint16_t* packed,
+                   const uint16_t* queries_h, const float* qnorms, int dpad,
+                   const ScanJobs& jb, float* dists_out, hipStream_t stream,
+                   const uint32_t* filter_bitset = nullptr);
 void launch_qnorms_i8(bool uns, const uint8_t* q, int nq, int dpad,
                       int32_t* out, hipStream_t stream);
 

@@ -18,6 +18,7 @@
  */
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
 #include <cfloat>
 #include <cstdint>
 #include <cstdlib>
@@ -731,6 +732,291 @@ void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
                            dim3(block), 0, stream, packed, dim, dpad,
                            group_slotbase, group_valid, ngroups, slot_lo,
                            slot_hi, out);
+}
+
+/* ---------------- half-precision (f16 / bf16) storage -------------------
+ * MO's float16/bf16 quantization is a PLAIN CAST (quantizer.go:50-58,
+ * CastSQL:179-183): entries and the query are narrowed with
+ * round-to-nearest-even (float16.go:53-63 BF16FromFloat32 incl. its NaN
+ * rule; f32bitsToF16bits:114-160 == IEEE RTNE, which v_cvt_f16_f32
+ * implements) and distances decode back to f32 and accumulate in f32
+ * (distance_func_narrow.go:27-52). Layout: 16-byte units of u16 elements
+ * ([g][dpad/8][64][8 halves]); dpad is padded to a multiple of 8 for
+ * half-typed indexes. */
+
+__device__ __forceinline__ float moann_h2f(uint16_t b) {
+    __half_raw r;
+    r.x = b;
+    return __half2float((__half)r);
+}
+__device__ __forceinline__ float moann_bf2f(uint16_t b) {
+    return __uint_as_float((uint32_t)b << 16);
+}
+__device__ __forceinline__ uint16_t moann_f2h(float f) {
+    __half_raw r = (__half_raw)__float2half(f); /* RTNE */
+    return r.x;
+}
+__device__ __forceinline__ uint16_t moann_f2bf(float f) {
+    const uint32_t x = __float_as_uint(f);
+    if (((x >> 23) & 0xff) == 0xff && (x & 0x7fffff))
+        return (uint16_t)((x >> 16) | 0x0040); /* quiet the NaN */
+    return (uint16_t)((x + 0x7fffu + ((x >> 16) & 1)) >> 16);
+}
+
+/* f32 rows -> u16 rows (pad columns 0); mirrors quantize_rows_kernel */
+template <bool BF>
+__global__ void quantize_half_rows_kernel(const float* __restrict__ in,
+                                          int64_t nrows, int in_stride,
+                                          int dim, int dpad,
+                                          uint16_t* __restrict__ out) {
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= nrows * dpad) return;
+    const int64_t r = idx / dpad;
+    const int c = (int)(idx % dpad);
+    uint16_t v = 0;
+    if (c < dim) {
+        const float x = in[r * in_stride + c];
+        v = BF ? moann_f2bf(x) : moann_f2h(x);
+    }
+    out[idx] = v;
+}
+
+void launch_quantize_half_rows(bool bf, const float* in, int64_t nrows,
+                               int in_stride, int dim, int dpad,
+                               uint16_t* out, hipStream_t stream) {
+    const int64_t total = nrows * dpad;
+    if (!total) return;
+    const int64_t grid = (total + 255) / 256;
+    if (bf)
+        hipLaunchKernelGGL(quantize_half_rows_kernel<true>,
+                           dim3((uint32_t)grid), dim3(256), 0, stream, in,
+                           nrows, in_stride, dim, dpad, out);
+    else
+        hipLaunchKernelGGL(quantize_half_rows_kernel<false>,
+                           dim3((uint32_t)grid), dim3(256), 0, stream, in,
+                           nrows, in_stride, dim, dpad, out);
+}
+
+/* |q|^2 over DECODED query halves (cosine denominators) */
+template <bool BF>
+__global__ void qnorms_h_kernel(const uint16_t* __restrict__ q, int nq,
+                                int dpad, float* __restrict__ out) {
+    const int i = blockIdx.x;
+    const int lane = threadIdx.x & 63;
+    if (threadIdx.x >= 64 || i >= nq) return;
+    float acc = 0.f;
+    for (int t = lane; t < dpad; t += 64) {
+        const float x = BF ? moann_bf2f(q[(int64_t)i * dpad + t])
+                           : moann_h2f(q[(int64_t)i * dpad + t]);
+        acc = fmaf(x, x, acc);
+    }
+    for (int off = 32; off; off >>= 1) acc += __shfl_down(acc, off, 64);
+    if (lane == 0) out[i] = acc;
+}
+
+void launch_qnorms_h(bool bf, const uint16_t* q, int nq, int dpad, float* out,
+                     hipStream_t stream) {
+    if (!nq) return;
+    if (bf)
+        hipLaunchKernelGGL(qnorms_h_kernel<true>, dim3(nq), dim3(64), 0,
+                           stream, q, nq, dpad, out);
+    else
+        hipLaunchKernelGGL(qnorms_h_kernel<false>, dim3(nq), dim3(64), 0,
+                           stream, q, nq, dpad, out);
+}
+
+/* half-storage list scan: dwordx4 = 8 halves/lane per step, decoded to
+ * f32, same job/filter/output semantics as the other scans. QT=8, GP=2. */
+template <int METRIC, bool BF>
+__global__ __launch_bounds__(256) void scan_h_kernel(
+    const uint16_t* __restrict__ packed, const uint16_t* __restrict__ queries_h,
+    const float* __restrict__ qnorms, int dpad,
+    const int64_t* __restrict__ j_databaseg,
+    const int32_t* __restrict__ j_gstart, const int32_t* __restrict__ j_gcount,
+    const int32_t* __restrict__ j_rows, const int32_t* __restrict__ j_nq,
+    const int32_t* __restrict__ j_qbase,
+    const int32_t* __restrict__ qslot_query,
+    const int64_t* __restrict__ qslot_outbase,
+    const int64_t* __restrict__ j_slot_base,
+    const uint32_t* __restrict__ filter_bitset,
+    float* __restrict__ dists_out) {
+    constexpr int QT = 8;
+    extern __shared__ float ldsf[];
+    uint16_t* ldsq = (uint16_t*)ldsf;           /* [QT][dpad] halves */
+    float* ldsn = (float*)(ldsq + QT * dpad);
+
+    int j;
+    {
+        const int nwg = gridDim.x, bid = blockIdx.x;
+        const int q8 = nwg >> 3, r8 = nwg & 7;
+        const int xcd = bid & 7, idx = bid >> 3;
+        j = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+        if (j >= nwg) j = bid;
+    }
+    const int nq = j_nq[j];
+    const int qbase = j_qbase[j];
+    for (int t = 0; t < QT; ++t) {
+        if (t < nq) {
+            const int q = qslot_query[qbase + t];
+            const uint16_t* src = queries_h + (int64_t)q * dpad;
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = src[e];
+            if (METRIC == KM_COS && threadIdx.x == 0) ldsn[t] = qnorms[q];
+        } else {
+            for (int e = threadIdx.x; e < dpad; e += blockDim.x)
+                ldsq[t * dpad + e] = 0;
+            if (METRIC == KM_COS && threadIdx.x == 0) ldsn[t] = 0.f;
+        }
+    }
+    __syncthreads();
+
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int gstart = j_gstart[j], gcount = j_gcount[j];
+    const int rows = j_rows[j];
+    const int64_t baseg = j_databaseg[j];
+    const int d8 = dpad >> 3;
+
+#define MOANN_H8(w, out0, out1)                                                  {                                                                                const float va = BF ? moann_bf2f((uint16_t)((w) & 0xFFFF))                                       : moann_h2f((uint16_t)((w) & 0xFFFF));                   const float vb = BF ? moann_bf2f((uint16_t)((w) >> 16))                                          : moann_h2f((uint16_t)((w) >> 16));                      out0 = va;                                                                   out1 = vb;                                                               }
+
+    for (int gp = gstart + 2 * wave; gp < gstart + gcount; gp += 8) {
+        const int g0 = gp;
+        const bool has1 = (gp + 1) < (gstart + gcount);
+        const int g1 = has1 ? gp + 1 : gp;
+        const uint4* __restrict__ d0 =
+            (const uint4*)(packed + (baseg + g0) * (int64_t)64 * dpad) + lane;
+        const uint4* __restrict__ d1 =
+            (const uint4*)(packed + (baseg + g1) * (int64_t)64 * dpad) + lane;
+
+        float acc0[QT], acc1[QT];
+#pragma unroll
+        for (int t = 0; t < QT; ++t) acc0[t] = acc1[t] = 0.f;
+        float rn0 = 0.f, rn1 = 0.f;
+
+#pragma unroll 2
+        for (int j8 = 0; j8 < d8; ++j8) {
+            const uint4 c0 = d0[j8 * 64];
+            const uint4 c1 = d1[j8 * 64];
+            float x0[8], x1[8];
+            MOANN_H8(c0.x, x0[0], x0[1]) MOANN_H8(c0.y, x0[2], x0[3])
+            MOANN_H8(c0.z, x0[4], x0[5]) MOANN_H8(c0.w, x0[6], x0[7])
+            MOANN_H8(c1.x, x1[0], x1[1]) MOANN_H8(c1.y, x1[2], x1[3])
+            MOANN_H8(c1.z, x1[4], x1[5]) MOANN_H8(c1.w, x1[6], x1[7])
+            if (METRIC == KM_COS) {
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    rn0 = fmaf(x0[u], x0[u], rn0);
+                    rn1 = fmaf(x1[u], x1[u], rn1);
+                }
+            }
+#pragma unroll
+            for (int t = 0; t < QT; ++t) {
+                const uint4 qw =
+                    *(const uint4*)(ldsq + t * dpad + j8 * 8);
+                float qv[8];
+                MOANN_H8(qw.x, qv[0], qv[1]) MOANN_H8(qw.y, qv[2], qv[3])
+                MOANN_H8(qw.z, qv[4], qv[5]) MOANN_H8(qw.w, qv[6], qv[7])
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    if (METRIC == KM_L2SQ) {
+                        const float e0 = x0[u] - qv[u];
+                        acc0[t] = fmaf(e0, e0, acc0[t]);
+                        const float e1 = x1[u] - qv[u];
+                        acc1[t] = fmaf(e1, e1, acc1[t]);
+                    } else if (METRIC == KM_IP || METRIC == KM_COS) {
+                        acc0[t] = fmaf(x0[u], qv[u], acc0[t]);
+                        acc1[t] = fmaf(x1[u], qv[u], acc1[t]);
+                    } else {
+                        acc0[t] += fabsf(x0[u] - qv[u]);
+                        acc1[t] += fabsf(x1[u] - qv[u]);
+                    }
+                }
+            }
+        }
+#undef MOANN_H8
+
+        const int row0 = g0 * 64 + lane;
+        const int row1 = g1 * 64 + lane;
+        bool pass0 = true, pass1 = true;
+        if (filter_bitset) {
+            const int64_t sb = j_slot_base[j];
+            if (row0 < rows) {
+                const int64_t gs = sb + row0;
+                pass0 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+            if (has1 && row1 < rows) {
+                const int64_t gs = sb + row1;
+                pass1 = (filter_bitset[gs >> 5] >> (gs & 31)) & 1u;
+            }
+        }
+#pragma unroll
+        for (int t = 0; t < QT; ++t) {
+            if (t >= nq) break;
+            const int64_t ob = qslot_outbase[qbase + t];
+            if (row0 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = -acc0[t];
+                else if (METRIC == KM_COS) {
+                    const double denom =
+                        sqrt((double)rn0) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc0[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else dist = acc0[t];
+                dists_out[ob + row0] = pass0 ? dist : FLT_MAX;
+            }
+            if (has1 && row1 < rows) {
+                float dist;
+                if (METRIC == KM_IP) dist = -acc1[t];
+                else if (METRIC == KM_COS) {
+                    const double denom =
+                        sqrt((double)rn1) * sqrt((double)ldsn[t]);
+                    if (denom == 0.0) dist = 1.0f;
+                    else {
+                        double sim = (double)acc1[t] / denom;
+                        sim = sim > 1.0 ? 1.0 : (sim < -1.0 ? -1.0 : sim);
+                        dist = (float)(1.0 - sim);
+                    }
+                } else dist = acc1[t];
+                dists_out[ob + row1] = pass1 ? dist : FLT_MAX;
+            }
+        }
+    }
+}
+
+template <int METRIC>
+static void launch_scan_h_m(bool bf, const uint16_t* packed,
+                            const uint16_t* queries_h, const float* qnorms,
+                            int dpad, const ScanJobs& jb, float* dists_out,
+                            hipStream_t stream,
+                            const uint32_t* filter_bitset) {
+    const dim3 grid(jb.njobs), block(256);
+    const size_t shmem = 8 * dpad * 2 + 8 * 4 + 16;
+#define HFARGS packed, queries_h, qnorms, dpad, jb.databaseg, jb.gstart, \
+    jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase, \
+    jb.slot_base, filter_bitset, dists_out
+    if (bf)
+        hipLaunchKernelGGL((scan_h_kernel<METRIC, true>), grid, block, shmem,
+                           stream, HFARGS);
+    else
+        hipLaunchKernelGGL((scan_h_kernel<METRIC, false>), grid, block,
+                           shmem, stream, HFARGS);
+#undef HFARGS
+}
+
+void launch_scan_h(int metric, bool bf, const uint16_t* packed,
+                   const uint16_t* queries_h, const float* qnorms, int dpad,
+                   const ScanJobs& jb, float* dists_out, hipStream_t stream,
+                   const uint32_t* filter_bitset) {
+    if (!jb.njobs) return;
+    switch (metric) {
+    case KM_L2SQ: launch_scan_h_m<KM_L2SQ>(bf, packed, queries_h, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_IP:   launch_scan_h_m<KM_IP>(bf, packed, queries_h, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_COS:  launch_scan_h_m<KM_COS>(bf, packed, queries_h, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    default:      launch_scan_h_m<KM_L1>(bf, packed, queries_h, qnorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    }
 }
 
 /* ---------------- exact f32 refine (two-stage scan) ---------------------

@@ -534,10 +534,14 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
         const int v = e ? atoi(e) : 0;
         return (int64_t)(v > 0 ? v : 8);
     }();
-    /* the chunk target is BYTES (~1.5 MB of list data); byte storage packs
-     * 4x the rows per byte, so scale the group count accordingly */
-    const int64_t CHUNK_G = ix->qtype != Quantization_F32 ? CHUNK_G_BASE * 4
-                                                          : CHUNK_G_BASE;
+    /* the chunk target is BYTES (~1.5 MB of list data); narrow storage
+     * packs more rows per byte, so scale the group count accordingly */
+    const int64_t chunk_scale =
+        ix->qtype == Quantization_F32 ? 1
+        : (ix->qtype == Quantization_F16 || ix->qtype == Quantization_BF16)
+            ? 2
+            : 4;
+    const int64_t CHUNK_G = CHUNK_G_BASE * chunk_scale;
 
     std::vector<int64_t> databaseg, qslot_outbase, jslot_base;
     std::vector<int32_t> gstart, gcount, rows, jnq, qbase, qslot_query;
@@ -574,6 +578,8 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     /* 4. list scan into the ragged candidate buffer */
     const bool narrow = ix->qtype != Quantization_F32;
     const bool uns = ix->qtype == Quantization_UINT8;
+    const bool half = ix->qtype == Quantization_F16 ||
+                      ix->qtype == Quantization_BF16;
     const bool use_refine =
         !narrow && ix->refine_depth > 0 && ix->d_packed_rq != nullptr;
     const uint32_t* d_filter = nullptr;
@@ -630,6 +636,26 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                            cx->w_queries_q.as<uint8_t>(), d_qni,
                            ix->d_rownorm_rq, dpad, jb,
                            cx->w_cand.as<float>(), s, d_filter);
+        } else if (half) {
+            /* plain-cast narrow search: query cast RTNE like the entries
+             * (relation_search.go:56-63), decoded-f32 distance */
+            cx->w_queries_q.ensure((size_t)nq * dpad * 2);
+            launch_quantize_half_rows(ix->qtype == Quantization_BF16, d_q,
+                                      (int64_t)nq, dpad, (int)ix->dim, dpad,
+                                      cx->w_queries_q.as<uint16_t>(), s);
+            KCHECK("query-half-cast");
+            const float* d_qnh = nullptr;
+            if (ix->kmetric == KM_COS) {
+                cx->w_qnorms_i.ensure(nq * 4);
+                launch_qnorms_h(ix->qtype == Quantization_BF16,
+                                cx->w_queries_q.as<uint16_t>(), (int)nq,
+                                dpad, cx->w_qnorms_i.as<float>(), s);
+                d_qnh = cx->w_qnorms_i.as<float>();
+            }
+            launch_scan_h(ix->kmetric, ix->qtype == Quantization_BF16,
+                          (const uint16_t*)ix->d_packed_q,
+                          cx->w_queries_q.as<uint16_t>(), d_qnh, dpad, jb,
+                          cx->w_cand.as<float>(), s, d_filter);
         } else if (narrow) {
             /* quantize the padded f32 queries with the entry map (the
              * search_quantize contract: query narrowed the same way as the
@@ -858,15 +884,20 @@ gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
     try {
         if (btype != Quantization_F32 ||
             (qtype != Quantization_F32 && qtype != Quantization_INT8 &&
-             qtype != Quantization_UINT8))
+             qtype != Quantization_UINT8 && qtype != Quantization_F16 &&
+             qtype != Quantization_BF16))
             throw std::runtime_error(
-                "wired combos: F32 base with F32/INT8/UINT8 storage");
+                "wired combos: F32 base with F32/F16/BF16/INT8/UINT8 "
+                "storage");
         auto ix = std::make_unique<IvfIndex>();
         ix->device = (devices && device_count > 0) ? devices[0] : 0;
         HIP_CHECK(hipSetDevice(ix->device));
         HIP_CHECK(hipStreamCreate(&ix->stream));
         ix->dim = dimension;
-        ix->dpad = (dimension + 3) & ~3u;
+        /* halves stream 8 elements (16 B) per lane step -> pad to 8 */
+        ix->dpad = (qtype == Quantization_F16 || qtype == Quantization_BF16)
+                       ? ((dimension + 7) & ~7u)
+                       : ((dimension + 3) & ~3u);
         ix->metric = metric;
         ix->kmetric = metric_kind(metric);
         ix->nlist = build_params.n_lists ? build_params.n_lists : 1024;
@@ -983,16 +1014,27 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
             }
         if (ix->qtype != Quantization_F32) {
             /* quantize the staged rows (pad columns emit 0 so padded dims
-             * stay identity), byte-pack into [g][dpad/4][64][4], and drop
-             * the f32 copy — storage stays narrow (search.go:1062 quantized
-             * domain). */
+             * stay identity), byte-pack, and drop the f32 copy — storage
+             * stays narrow (search.go:1062 quantized domain). f16/bf16 are
+             * MO's PLAIN-CAST narrowing (quantizer.go:50-58): u16 elements,
+             * no affine scaling. */
             const bool uns = ix->qtype == Quantization_UINT8;
+            const bool half = ix->qtype == Quantization_F16 ||
+                              ix->qtype == Quantization_BF16;
+            const size_t eb = half ? 2 : 1;
             DevBuf d_qrows, d_rb2, d_gv2, d_sr2;
-            d_qrows.ensure((size_t)ix->count * ix->dpad);
-            launch_quantize_rows(uns, ix->d_staging, (int64_t)ix->count,
-                                 (int)ix->dim, (int)ix->dim, (int)ix->dpad,
-                                 (float)ix->quant_mul, (float)ix->quant_add,
-                                 d_qrows.as<uint8_t>(), ix->stream);
+            d_qrows.ensure((size_t)ix->count * ix->dpad * eb);
+            if (half)
+                launch_quantize_half_rows(
+                    ix->qtype == Quantization_BF16, ix->d_staging,
+                    (int64_t)ix->count, (int)ix->dim, (int)ix->dim,
+                    (int)ix->dpad, d_qrows.as<uint16_t>(), ix->stream);
+            else
+                launch_quantize_rows(uns, ix->d_staging, (int64_t)ix->count,
+                                     (int)ix->dim, (int)ix->dim,
+                                     (int)ix->dpad, (float)ix->quant_mul,
+                                     (float)ix->quant_add,
+                                     d_qrows.as<uint8_t>(), ix->stream);
             KCHECK("quantize-rows");
             d_rb2.ensure(ix->total_groups * 8);
             d_gv2.ensure(ix->total_groups * 4);
@@ -1014,19 +1056,21 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
                 ix->d_rownorm_q = nullptr;
             }
             HIP_CHECK(hipMalloc(&ix->d_packed_q,
-                                (size_t)ix->total_groups * 64 * ix->dpad));
-            if ((ix->dpad & 15) == 0)
-                /* 16-byte-unit layout for the dot-form scan */
-                launch_bytes_pack16(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                                (size_t)ix->total_groups * 64 * ix->dpad *
+                                    eb));
+            const int bpad = (int)(ix->dpad * eb); /* row bytes */
+            if ((bpad & 15) == 0)
+                /* 16-byte-unit layout for the dwordx4 scans */
+                launch_bytes_pack16(d_qrows.as<uint8_t>(), bpad,
                                     d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
                                     d_sr2.as<int64_t>(), ix->total_groups,
                                     ix->d_packed_q, ix->stream);
             else
-                launch_pq_pack(d_qrows.as<uint8_t>(), (int)ix->dpad,
+                launch_pq_pack(d_qrows.as<uint8_t>(), bpad,
                                d_rb2.as<int64_t>(), d_gv2.as<int32_t>(),
                                d_sr2.as<int64_t>(), ix->total_groups,
                                ix->d_packed_q, ix->stream);
-            if ((ix->dpad & 15) == 0) {
+            if (!half && (ix->dpad & 15) == 0) {
                 /* row norms for the dot-form byte scan (rn + qn - 2*dot) */
                 HIP_CHECK(hipMalloc(&ix->d_rownorm_q,
                                     (size_t)ix->total_groups * 64 * 4));
@@ -1479,6 +1523,9 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
              * reconstruction error is << 0.5 code steps). */
             FILE* f = fopen((d + "/vectors.u8").c_str(), "wb");
             if (!f) throw std::runtime_error("cannot write vectors.u8");
+            const size_t s_eb = (ix->qtype == Quantization_F16 ||
+                                 ix->qtype == Quantization_BF16)
+                                    ? 2 : 1; /* bytes per element */
             DevBuf d_gsb, d_gv, d_tmp;
             d_gsb.ensure(ix->total_groups * 8);
             d_gv.ensure(ix->total_groups * 4);
@@ -1487,20 +1534,21 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
             HIP_CHECK(hipMemcpy(d_gv.ptr, ix->group_valid.data(),
                                 ix->total_groups * 4, hipMemcpyHostToDevice));
             const int64_t CH = 1 << 21;
-            d_tmp.ensure((size_t)CH * ix->dim);
-            std::vector<uint8_t> hbuf((size_t)CH * ix->dim);
+            d_tmp.ensure((size_t)CH * ix->dim * s_eb);
+            std::vector<uint8_t> hbuf((size_t)CH * ix->dim * s_eb);
             for (int64_t lo = 0; lo < (int64_t)ix->count; lo += CH) {
                 const int64_t hi = std::min<int64_t>(lo + CH, ix->count);
-                launch_unpack_bytes(ix->d_packed_q, (int)ix->dim,
-                                    (int)ix->dpad, d_gsb.as<int64_t>(),
+                launch_unpack_bytes(ix->d_packed_q, (int)(ix->dim * s_eb),
+                                    (int)(ix->dpad * s_eb),
+                                    d_gsb.as<int64_t>(),
                                     d_gv.as<int32_t>(), ix->total_groups, lo,
                                     hi, d_tmp.as<uint8_t>(), ix->stream);
                 HIP_CHECK(hipMemcpyAsync(hbuf.data(), d_tmp.ptr,
-                                         (size_t)(hi - lo) * ix->dim,
+                                         (size_t)(hi - lo) * ix->dim * s_eb,
                                          hipMemcpyDeviceToHost, ix->stream));
                 HIP_CHECK(hipStreamSynchronize(ix->stream));
                 HIP_CHECK(hipGetLastError());
-                const size_t n = (size_t)(hi - lo) * ix->dim;
+                const size_t n = (size_t)(hi - lo) * ix->dim * s_eb;
                 if (fwrite(hbuf.data(), 1, n, f) != n) {
                     fclose(f);
                     throw std::runtime_error("short write vectors.u8");
@@ -1602,34 +1650,76 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
             rfile("assign.i32", assign.data(), count * 4);
             /* vectors streamed into add_chunk */
             const bool narrow = qtype != (int)Quantization_F32;
+            const bool l_half_pre = qtype == (int)Quantization_F16 ||
+                                    qtype == (int)Quantization_BF16;
             const char* vname = narrow ? "/vectors.u8" : "/vectors.f32";
             FILE* vf = fopen((d + vname).c_str(), "rb");
             if (!vf)
                 throw std::runtime_error(std::string("cannot read ") + vname);
-            if (narrow) {
+            if (narrow && !l_half_pre) {
                 /* pin the quantizer BEFORE build so re-quantization uses
-                 * the saved parameters */
+                 * the saved parameters (halves are plain casts) */
                 gpu_ivf_flat_set_quantizer(h, (float)qmin, (float)qmax,
                                            errmsg);
             }
             const double mul = ix->quant_mul, add = ix->quant_add;
+            const bool l_half = qtype == (int)Quantization_F16 ||
+                                qtype == (int)Quantization_BF16;
+            const size_t l_eb = l_half ? 2 : 1;
             const size_t CH = 1 << 20;
             std::vector<float> vbuf(CH * dim);
-            std::vector<uint8_t> qbuf(narrow ? CH * dim : 0);
+            std::vector<uint8_t> qbuf(narrow ? CH * dim * l_eb : 0);
             unsigned long long done = 0;
             while (done < count) {
                 const size_t n = (size_t)std::min<unsigned long long>(
                     CH, count - done);
                 if (narrow) {
-                    if (fread(qbuf.data(), 1, n * dim, vf) != n * dim) {
+                    if (fread(qbuf.data(), 1, n * dim * l_eb, vf) !=
+                        n * dim * l_eb) {
                         fclose(vf);
                         throw std::runtime_error("short read vectors.u8");
                     }
-                    const bool uns = qtype == (int)Quantization_UINT8;
-                    for (size_t i = 0; i < n * dim; ++i) {
-                        const double q = uns ? (double)qbuf[i]
-                                             : (double)(int8_t)qbuf[i];
-                        vbuf[i] = (float)((q - add) / mul);
+                    if (l_half) {
+                        const uint16_t* hp = (const uint16_t*)qbuf.data();
+                        const bool bf = qtype == (int)Quantization_BF16;
+                        for (size_t i = 0; i < n * dim; ++i) {
+                            const uint16_t b = hp[i];
+                            uint32_t u;
+                            if (bf) {
+                                u = (uint32_t)b << 16;
+                            } else { /* IEEE binary16 -> binary32 widen */
+                                const uint32_t sign = (uint32_t)(b & 0x8000)
+                                                      << 16;
+                                uint32_t exp = (b >> 10) & 0x1f;
+                                uint32_t coef = b & 0x3ff;
+                                if (exp == 0x1f) {
+                                    u = sign | 0x7f800000u | (coef << 13);
+                                } else if (exp == 0 && coef == 0) {
+                                    u = sign;
+                                } else {
+                                    if (exp == 0) { /* subnormal */
+                                        exp = 1;
+                                        while (!(coef & 0x400)) {
+                                            coef <<= 1;
+                                            exp--;
+                                        }
+                                        coef &= 0x3ff;
+                                    }
+                                    u = sign | ((exp + 112) << 23) |
+                                        (coef << 13);
+                                }
+                            }
+                            float fv;
+                            memcpy(&fv, &u, 4);
+                            vbuf[i] = fv;
+                        }
+                    } else {
+                        const bool uns = qtype == (int)Quantization_UINT8;
+                        for (size_t i = 0; i < n * dim; ++i) {
+                            const double q = uns ? (double)qbuf[i]
+                                                 : (double)(int8_t)qbuf[i];
+                            vbuf[i] = (float)((q - add) / mul);
+                        }
                     }
                 } else if (fread(vbuf.data(), 4, n * dim, vf) != n * dim) {
                     fclose(vf);
@@ -1747,8 +1837,10 @@ void gpu_ivf_flat_set_quantizer(gpu_ivf_flat_c h, float min, float max,
                                 void* errmsg) {
     try {
         auto ix = IX(h);
-        if (ix->qtype == Quantization_F32)
-            throw std::runtime_error("index storage is f32 (no quantizer)");
+        if (ix->qtype != Quantization_INT8 &&
+            ix->qtype != Quantization_UINT8)
+            throw std::runtime_error(
+                "storage has no affine quantizer (int8/uint8 only)");
         const double rng = (double)max - (double)min;
         ix->quant_min = min;
         ix->quant_max = max;

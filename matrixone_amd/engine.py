@@ -222,7 +222,7 @@ class IvfFlatIndex:
     SearchAsync / SearchWait), with the MatrixOne-CPU-path ingestion shape
     (explicit centroids + entry assignments from the hidden tables)."""
 
-    QTYPES = {"f32": 0, "int8": 2, "uint8": 3}  # quantization_t
+    QTYPES = {"f32": 0, "f16": 1, "int8": 2, "uint8": 3, "bf16": 4}  # quantization_t
 
     def __init__(self, dim: int, nlist: int, metric: str = "l2sq",
                  capacity: int = 0, device: int = 0, qtype: str = "f32"):

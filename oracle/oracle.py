@@ -313,3 +313,41 @@ def ivf_search_quantized(idx: "IvfIndex", metric, queries, probe, k,
                                     else sel)
         out_d[qi, :len(order)] = raw.astype(np.float32)
     return out_ids, out_d
+
+
+# ---------------- half-precision (f16 / bf16) narrowing ---------------------
+# MO's float16/bf16 quantization is a PLAIN CAST (quantizer.go:50-58): entries
+# and the query are narrowed round-to-nearest-even and distances decode back
+# to f32 (distance_func_narrow.go:27-52). numpy's float16 conversion IS IEEE
+# RTNE (== float16.go f32bitsToF16bits); bf16 is restated bit-exactly from
+# BF16FromFloat32 (float16.go:53-63, incl. the quiet-NaN rule).
+
+def encode_decode_f16(x: np.ndarray) -> np.ndarray:
+    """f32 -> binary16 (RTNE) -> f32."""
+    with np.errstate(over="ignore"):  # overflow -> inf is the IEEE (and
+        # f32bitsToF16bits:131-134) behavior, not an error
+        return (np.asarray(x, dtype=np.float32).astype(np.float16)
+                .astype(np.float32))
+
+
+def encode_decode_bf16(x: np.ndarray) -> np.ndarray:
+    """f32 -> bfloat16 (RTNE, NaN quieted) -> f32, bit-exact to
+    types.BF16FromFloat32/ToFloat32."""
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    u = x.view(np.uint32)
+    nan = ((u >> 23) & 0xFF == 0xFF) & (u & 0x7FFFFF != 0)
+    rounded = (u + np.uint32(0x7FFF) + ((u >> 16) & 1)) >> 16
+    b16 = np.where(nan, (u >> 16) | np.uint32(0x0040), rounded)
+    return ((b16.astype(np.uint32) << 16)).view(np.float32).reshape(x.shape)
+
+
+def ivf_search_half(idx: "IvfIndex", metric: int, queries: np.ndarray,
+                    probe: int, k: int, bf16: bool, orig_l2: bool = False):
+    """Half-storage search spec: run the exact IVF oracle over the DECODED
+    entries/centroids-as-stored and the DECODED query (entries narrowed at
+    build, query narrowed at search — relation_search.go:56-63; distance =
+    f32 arithmetic on decoded values)."""
+    enc = encode_decode_bf16 if bf16 else encode_decode_f16
+    dec_vecs = enc(idx.vecs)
+    didx = IvfIndex(idx.centroids, dec_vecs, idx.assign, ids=idx.ids)
+    return didx.search(metric, enc(queries), probe, k, orig_l2=orig_l2)

@@ -624,3 +624,68 @@ def test_search_with_tail():
     b_i, b_d = ix.search(queries, k, nlist)
     np.testing.assert_array_equal(a_i, b_i)
     ix.close()
+
+
+@pytest.mark.parametrize("qtype,metric", [
+    ("f16", "l2sq"), ("f16", "cos"), ("bf16", "l2sq"), ("bf16", "ip")])
+def test_half_storage(qtype, metric):
+    """f16/bf16 plain-cast storage (quantizer.go:50-58): the HIP half scan
+    == the oracle run over RTNE-decoded entries and query
+    (relation_search.go:56-63, distance_func_narrow.go:27-52)."""
+    from matrixone_amd import engine
+    mmap = {"l2sq": orc.METRIC_L2SQ, "ip": orc.METRIC_IP,
+            "cos": orc.METRIC_COS}
+    rng = np.random.Generator(np.random.PCG64(1600 + len(qtype)))
+    n, d, nlist, nq, probe, k = 12000, 96, 32, 12, 6, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    ids64 = rng.permutation(n).astype(np.int64) + 5
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+
+    idx = orc.IvfIndex(cents, vecs, assign, ids=ids64)
+    ref_ids, ref_d = orc.ivf_search_half(idx, mmap[metric], queries, probe,
+                                         k, bf16=(qtype == "bf16"))
+    ix = engine.IvfFlatIndex(d, nlist, metric=metric, capacity=n,
+                             qtype=qtype)
+    ix.add(vecs, ids=ids64)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids, dists = ix.search(queries, k, probe)
+    _assert_parity(ids, dists, ref_ids, ref_d, ctx=f"{qtype}/{metric}")
+    ix.close()
+
+
+def test_half_storage_odd_dim_and_saveload(tmp_path):
+    """dim not a multiple of 8 (dpad rounds to 8 for halves) + persistence
+    round trip for f16 storage."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(77))
+    n, d, nlist, nq, probe, k = 5000, 67, 16, 8, 4, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    idx = orc.IvfIndex(cents, vecs, assign)
+    ref_ids, ref_d = orc.ivf_search_half(idx, orc.METRIC_L2SQ, queries,
+                                         probe, k, bf16=False)
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n,
+                             qtype="f16")
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ids, dists = ix.search(queries, k, probe)
+    _assert_parity(ids, dists, ref_ids, ref_d, ctx="f16-d67")
+    ix.save_dir(str(tmp_path))
+    ix2 = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n,
+                              qtype="f16")
+    ix2.load_dir(str(tmp_path))
+    ids2, d2 = ix2.search(queries, k, probe)
+    np.testing.assert_array_equal(ids2, ids)
+    np.testing.assert_allclose(d2, dists)
+    ix.close()
+    ix2.close()
