@@ -184,7 +184,8 @@ def main():
                     help="two-stage scan: byte first pass + exact f32 "
                          "re-rank of the top-R (0 = exact one-pass scan). "
                          "recall@10 is still gated at 0.95 either way")
-    ap.add_argument("--qtype", default="f32", choices=["f32", "int8", "uint8"],
+    ap.add_argument("--qtype", default="f32",
+                    choices=["f32", "f16", "bf16", "int8", "uint8"],
                     help="storage type; the BASELINE metric is f32 — narrow "
                          "runs are secondary datapoints and report their own "
                          "dtype")
@@ -236,7 +237,7 @@ def main():
             ix.add(data[c0:c0 + CH].cpu().numpy())
     ix.set_centroids(cent_np)
     ix.set_assignments(assign_np)
-    if args.qtype != "f32":
+    if args.qtype in ("int8", "uint8"):
         sample = data[:2048].reshape(-1).float()  # quantile caps at 2^24
         lo = float(torch.quantile(sample, 0.001))
         hi = float(torch.quantile(sample, 0.999))
@@ -357,7 +358,7 @@ def main():
         f"step total {elapsed/args.steps*1e3:.3f} (pipelined={pipeline})")
 
     # ---- roofline (list-scan kernel, HIP events on the index stream) ------
-    elem = 1 if args.qtype != "f32" else 4
+    elem = {"f32": 4, "f16": 2, "bf16": 2}.get(args.qtype, 1)
     scan_gbs = (perf["scan_bytes"] / 4 * elem / 1e9) / max(
         perf["scan_ms"] / 1e3, 1e-9)
     roofline = {
