@@ -19,6 +19,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
+#include <algorithm>
 #include <cfloat>
 #include <cstdint>
 #include <cstdlib>
@@ -608,23 +609,25 @@ __global__ void pack_kernel(const float* __restrict__ vecs, int dim, int dpad,
                             const int64_t* __restrict__ slot_rows,
                             int64_t ngroups, float* __restrict__ packed) {
     const int64_t nquads = ngroups * (dpad >> 2) * 64;
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= nquads) return;
-    const int lane = (int)(idx & 63);
-    const int64_t rest = idx >> 6;
-    const int j4 = (int)(rest % (dpad >> 2));
-    const int64_t g = rest / (dpad >> 2);
-    float4 v = {0.f, 0.f, 0.f, 0.f};
-    if (lane < group_valid[g]) {
-        const int64_t row = slot_rows[group_rowbase[g] + lane];
-        const float* src = vecs + row * (int64_t)dim;
-        const int base = 4 * j4;
-        v.x = base + 0 < dim ? src[base + 0] : 0.f;
-        v.y = base + 1 < dim ? src[base + 1] : 0.f;
-        v.z = base + 2 < dim ? src[base + 2] : 0.f;
-        v.w = base + 3 < dim ? src[base + 3] : 0.f;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < nquads; idx += stride) {
+        const int lane = (int)(idx & 63);
+        const int64_t rest = idx >> 6;
+        const int j4 = (int)(rest % (dpad >> 2));
+        const int64_t g = rest / (dpad >> 2);
+        float4 v = {0.f, 0.f, 0.f, 0.f};
+        if (lane < group_valid[g]) {
+            const int64_t row = slot_rows[group_rowbase[g] + lane];
+            const float* src = vecs + row * (int64_t)dim;
+            const int base = 4 * j4;
+            v.x = base + 0 < dim ? src[base + 0] : 0.f;
+            v.y = base + 1 < dim ? src[base + 1] : 0.f;
+            v.z = base + 2 < dim ? src[base + 2] : 0.f;
+            v.w = base + 3 < dim ? src[base + 3] : 0.f;
+        }
+        ((float4*)packed)[idx] = v;
     }
-    ((float4*)packed)[idx] = v;
 }
 
 void launch_pack(const float* vecs, int dim, int dpad,
@@ -634,7 +637,8 @@ void launch_pack(const float* vecs, int dim, int dpad,
     const int64_t nquads = ngroups * (int64_t)(dpad >> 2) * 64;
     if (!nquads) return;
     const int block = 256;
-    const int64_t grid = (nquads + block - 1) / block;
+    const int64_t grid = std::min<int64_t>((nquads + block - 1) / block,
+                                           1 << 22);
     hipLaunchKernelGGL(pack_kernel, dim3((uint32_t)grid), dim3(block), 0,
                        stream, vecs, dim, dpad, group_rowbase, group_valid,
                        slot_rows, ngroups, packed);
@@ -650,22 +654,24 @@ __global__ void unpack_kernel(const float* __restrict__ packed, int dim,
                               int64_t ngroups, int64_t slot_lo,
                               int64_t slot_hi, float* __restrict__ out) {
     const int64_t nquads = ngroups * (dpad >> 2) * 64;
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= nquads) return;
-    const int lane = (int)(idx & 63);
-    const int64_t rest = idx >> 6;
-    const int j4 = (int)(rest % (dpad >> 2));
-    const int64_t g = rest / (dpad >> 2);
-    if (lane >= group_valid[g]) return;
-    const int64_t slot = group_slotbase[g] + lane;
-    if (slot < slot_lo || slot >= slot_hi) return;
-    const float4 v = ((const float4*)packed)[idx];
-    float* dst = out + (slot - slot_lo) * (int64_t)dim;
-    const int base = 4 * j4;
-    if (base + 0 < dim) dst[base + 0] = v.x;
-    if (base + 1 < dim) dst[base + 1] = v.y;
-    if (base + 2 < dim) dst[base + 2] = v.z;
-    if (base + 3 < dim) dst[base + 3] = v.w;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < nquads; idx += stride) {
+        const int lane = (int)(idx & 63);
+        const int64_t rest = idx >> 6;
+        const int j4 = (int)(rest % (dpad >> 2));
+        const int64_t g = rest / (dpad >> 2);
+        if (lane >= group_valid[g]) continue;
+        const int64_t slot = group_slotbase[g] + lane;
+        if (slot < slot_lo || slot >= slot_hi) continue;
+        const float4 v = ((const float4*)packed)[idx];
+        float* dst = out + (slot - slot_lo) * (int64_t)dim;
+        const int base = 4 * j4;
+        if (base + 0 < dim) dst[base + 0] = v.x;
+        if (base + 1 < dim) dst[base + 1] = v.y;
+        if (base + 2 < dim) dst[base + 2] = v.z;
+        if (base + 3 < dim) dst[base + 3] = v.w;
+    }
 }
 
 void launch_unpack(const float* packed, int dim, int dpad,
@@ -675,7 +681,8 @@ void launch_unpack(const float* packed, int dim, int dpad,
     const int64_t nquads = ngroups * (int64_t)(dpad >> 2) * 64;
     if (!nquads) return;
     const int block = 256;
-    const int64_t grid = (nquads + block - 1) / block;
+    const int64_t grid = std::min<int64_t>((nquads + block - 1) / block,
+                                           1 << 22);
     hipLaunchKernelGGL(unpack_kernel, dim3((uint32_t)grid), dim3(block), 0,
                        stream, packed, dim, dpad, group_slotbase, group_valid,
                        ngroups, slot_lo, slot_hi, out);
@@ -694,21 +701,23 @@ __global__ void unpack_bytes_kernel(const uint8_t* __restrict__ packed,
                                     uint8_t* __restrict__ out) {
     const int U = L16 ? 16 : 4;
     const int64_t nunits = ngroups * (int64_t)(dpad / U) * 64;
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= nunits) return;
-    const int lane = (int)(idx & 63);
-    const int64_t rest = idx >> 6;
-    const int m = (int)(rest % (dpad / U));
-    const int64_t g = rest / (dpad / U);
-    if (lane >= group_valid[g]) return;
-    const int64_t slot = group_slotbase[g] + lane;
-    if (slot < slot_lo || slot >= slot_hi) return;
-    const uint8_t* src = packed + idx * U;
-    uint8_t* dst = out + (slot - slot_lo) * (int64_t)dim;
-    const int base = m * U;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < nunits; idx += stride) {
+        const int lane = (int)(idx & 63);
+        const int64_t rest = idx >> 6;
+        const int m = (int)(rest % (dpad / U));
+        const int64_t g = rest / (dpad / U);
+        if (lane >= group_valid[g]) continue;
+        const int64_t slot = group_slotbase[g] + lane;
+        if (slot < slot_lo || slot >= slot_hi) continue;
+        const uint8_t* src = packed + idx * U;
+        uint8_t* dst = out + (slot - slot_lo) * (int64_t)dim;
+        const int base = m * U;
 #pragma unroll
-    for (int b = 0; b < U; ++b)
-        if (base + b < dim) dst[base + b] = src[b];
+        for (int b = 0; b < U; ++b)
+            if (base + b < dim) dst[base + b] = src[b];
+    }
 }
 
 void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
@@ -721,7 +730,8 @@ void launch_unpack_bytes(const uint8_t* packed, int dim, int dpad,
     const int64_t nunits = ngroups * (int64_t)(dpad / U) * 64;
     if (!nunits) return;
     const int block = 256;
-    const int64_t grid = (nunits + block - 1) / block;
+    const int64_t grid = std::min<int64_t>((nunits + block - 1) / block,
+                                           1 << 22);
     if (l16)
         hipLaunchKernelGGL(unpack_bytes_kernel<true>, dim3((uint32_t)grid),
                            dim3(block), 0, stream, packed, dim, dpad,
@@ -769,24 +779,28 @@ __global__ void quantize_half_rows_kernel(const float* __restrict__ in,
                                           int64_t nrows, int in_stride,
                                           int dim, int dpad,
                                           uint16_t* __restrict__ out) {
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= nrows * dpad) return;
-    const int64_t r = idx / dpad;
-    const int c = (int)(idx % dpad);
-    uint16_t v = 0;
-    if (c < dim) {
-        const float x = in[r * in_stride + c];
-        v = BF ? moann_f2bf(x) : moann_f2h(x);
+    const int64_t total = nrows * (int64_t)dpad;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x; /* see
+        quantize_rows_kernel: 32-bit work-item grids */
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        const int64_t r = idx / dpad;
+        const int c = (int)(idx % dpad);
+        uint16_t v = 0;
+        if (c < dim) {
+            const float x = in[r * (int64_t)in_stride + c];
+            v = BF ? moann_f2bf(x) : moann_f2h(x);
+        }
+        out[idx] = v;
     }
-    out[idx] = v;
 }
 
 void launch_quantize_half_rows(bool bf, const float* in, int64_t nrows,
                                int in_stride, int dim, int dpad,
                                uint16_t* out, hipStream_t stream) {
-    const int64_t total = nrows * dpad;
+    const int64_t total = nrows * (int64_t)dpad;
     if (!total) return;
-    const int64_t grid = (total + 255) / 256;
+    const int64_t grid = std::min<int64_t>((total + 255) / 256, 1 << 22);
     if (bf)
         hipLaunchKernelGGL(quantize_half_rows_kernel<true>,
                            dim3((uint32_t)grid), dim3(256), 0, stream, in,
@@ -1504,18 +1518,20 @@ __global__ void bytes_pack16_kernel(const uint8_t* __restrict__ rows_q,
                                     uint8_t* __restrict__ packed) {
     const int d16 = dpad >> 4;
     const int64_t nunits = ngroups * (int64_t)d16 * 64;
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= nunits) return;
-    const int lane = (int)(idx & 63);
-    const int64_t rest = idx >> 6;
-    const int m16 = (int)(rest % d16);
-    const int64_t g = rest / d16;
-    uint4 v = {0, 0, 0, 0};
-    if (lane < group_valid[g]) {
-        const int64_t row = slot_rows[group_rowbase[g] + lane];
-        v = *(const uint4*)(rows_q + row * (int64_t)dpad + 16 * m16);
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < nunits; idx += stride) {
+        const int lane = (int)(idx & 63);
+        const int64_t rest = idx >> 6;
+        const int m16 = (int)(rest % d16);
+        const int64_t g = rest / d16;
+        uint4 v = {0, 0, 0, 0};
+        if (lane < group_valid[g]) {
+            const int64_t row = slot_rows[group_rowbase[g] + lane];
+            v = *(const uint4*)(rows_q + row * (int64_t)dpad + 16 * m16);
+        }
+        ((uint4*)packed)[idx] = v;
     }
-    ((uint4*)packed)[idx] = v;
 }
 
 void launch_bytes_pack16(const uint8_t* rows_q, int dpad,
@@ -1526,7 +1542,8 @@ void launch_bytes_pack16(const uint8_t* rows_q, int dpad,
     const int64_t nunits = ngroups * (int64_t)(dpad >> 4) * 64;
     if (!nunits) return;
     const int block = 256;
-    const int64_t grid = (nunits + block - 1) / block;
+    const int64_t grid = std::min<int64_t>((nunits + block - 1) / block,
+                                           1 << 22);
     hipLaunchKernelGGL(bytes_pack16_kernel, dim3((uint32_t)grid), dim3(block),
                        0, stream, rows_q, dpad, group_rowbase, group_valid,
                        slot_rows, ngroups, packed);
@@ -1633,24 +1650,32 @@ __global__ void quantize_rows_kernel(const float* __restrict__ in,
                                      int64_t nrows, int in_stride, int dim,
                                      int dpad, float fmul, float fadd,
                                      uint8_t* __restrict__ out) {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= nrows * (int64_t)dpad) return;
-    const int64_t r = i / dpad;
-    const int c = (int)(i % dpad);
-    if (c >= dim) {
-        out[i] = 0;
-        return;
+    /* GRID-STRIDE: the AQL packet's grid size counts WORK-ITEMS in 32
+     * bits, so a flat launch of >= 2^32 threads silently wraps (measured:
+     * exactly total mod 2^32 elements written at 6M x 768 halves). Every
+     * elementwise kernel over rows x dpad strides instead. */
+    const int64_t total = nrows * (int64_t)dpad;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < total; i += stride) {
+        const int64_t r = i / dpad;
+        const int c = (int)(i % dpad);
+        if (c >= dim) {
+            out[i] = 0;
+            continue;
+        }
+        const float x = in[r * (int64_t)in_stride + c];
+        float m = x * fmul;
+        asm volatile("" : "+v"(m)); /* block FMA fusion of the two f32 ops */
+        const float sum = m + fadd;
+        float v;
+        if (sum != sum) v = 0.f;
+        else v = (float)round((double)sum);
+        const float lo = UNSIGNED ? 0.f : -128.f;
+        const float hi = UNSIGNED ? 255.f : 127.f;
+        v = v < lo ? lo : (v > hi ? hi : v);
+        out[i] = UNSIGNED ? (uint8_t)v : (uint8_t)(int8_t)v;
     }
-    const float x = in[r * (int64_t)in_stride + c];
-    float m = x * fmul;
-    asm volatile("" : "+v"(m)); /* block FMA fusion of the two f32 ops */
-    const float sum = m + fadd;
-    float v;
-    if (sum != sum) v = 0.f;
-    else v = (float)round((double)sum);
-    const float lo = UNSIGNED ? 0.f : -128.f, hi = UNSIGNED ? 255.f : 127.f;
-    v = v < lo ? lo : (v > hi ? hi : v);
-    out[i] = UNSIGNED ? (uint8_t)v : (uint8_t)(int8_t)v;
 }
 
 void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
@@ -1659,7 +1684,9 @@ void launch_quantize_rows(bool uns, const float* in, int64_t nrows,
     const int64_t total = nrows * (int64_t)dpad;
     if (!total) return;
     const int block = 256;
-    const int64_t grid = (total + block - 1) / block;
+    /* cap work-items below 2^31: the AQL grid is 32-bit in WORK-ITEMS */
+    const int64_t grid = std::min<int64_t>((total + block - 1) / block,
+                                           1 << 22);
     if (uns)
         hipLaunchKernelGGL((quantize_rows_kernel<true>), dim3((uint32_t)grid),
                            dim3(block), 0, stream, in, nrows, in_stride, dim,

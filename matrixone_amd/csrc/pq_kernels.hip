@@ -13,6 +13,7 @@
  * g*64+lane, subspace 4*mq+c — one coalesced 256 B read per wave step. */
 
 #include <hip/hip_runtime.h>
+#include <algorithm>
 #include <cfloat>
 #include <cstdint>
 
@@ -252,8 +253,9 @@ __global__ void pq_pack_kernel(const uint8_t* __restrict__ codes_rowmajor,
                                int64_t ngroups,
                                uint8_t* __restrict__ packed) {
     const int64_t nquads = ngroups * (nsub >> 2) * 64;
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= nquads) return;
+    const int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < nquads; idx += gstride) {
     const int lane = (int)(idx & 63);
     const int64_t rest = idx >> 6;
     const int mq = (int)(rest % (nsub >> 2));
@@ -265,6 +267,7 @@ __global__ void pq_pack_kernel(const uint8_t* __restrict__ codes_rowmajor,
         v.x = src[0]; v.y = src[1]; v.z = src[2]; v.w = src[3];
     }
     ((uchar4*)packed)[idx] = v;
+    }
 }
 
 void launch_pq_pack(const uint8_t* codes_rowmajor, int nsub,
@@ -274,7 +277,8 @@ void launch_pq_pack(const uint8_t* codes_rowmajor, int nsub,
     const int64_t nquads = ngroups * (int64_t)(nsub >> 2) * 64;
     if (!nquads) return;
     const int block = 256;
-    const int64_t grid = (nquads + block - 1) / block;
+    const int64_t grid = std::min<int64_t>((nquads + block - 1) / block,
+                                           1 << 22);
     hipLaunchKernelGGL(pq_pack_kernel, dim3((uint32_t)grid), dim3(block), 0,
                        stream, codes_rowmajor, nsub, group_rowbase,
                        group_valid, slot_rows, ngroups, packed);
