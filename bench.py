@@ -304,6 +304,8 @@ def main():
         return merge(ids, dists)
 
     def run_steps(n):
+        if n <= 0:
+            return
         if not pipeline:
             for i in range(n):
                 step(i)

@@ -576,8 +576,6 @@ def test_refine_exactness():
     # small R: whatever ids come back carry their EXACT f32 distances
     ix.enable_refine(32)
     sid, sd = ix.search(queries, k, probe)
-    id2d = {}
-    big_ids, big_d = ix.search(queries, 1200, probe) if False else (None, None)
     ix.enable_refine(0)  # exact reference for distance lookup
     ref_all, ref_alld = ix.search(queries, 1200, probe)
     for i in range(nq):
