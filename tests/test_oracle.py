@@ -224,3 +224,31 @@ def test_topk_merge():
     out_ids, out_dists = orc.topk_merge(ids, dists, 3)
     np.testing.assert_array_equal(out_ids[0], [1, 3, 2])
     np.testing.assert_allclose(out_dists[0], [0.5, 1.0, 2.0])
+
+
+def test_half_narrowing_restatements():
+    """f16/bf16 narrowing (float16.go RTNE semantics): numpy float16 IS
+    IEEE RTNE; the bf16 restatement must round-to-nearest-even, tie to
+    even, quiet NaNs, and round-trip every representable bf16."""
+    from oracle.oracle import encode_decode_bf16, encode_decode_f16
+    # every representable bf16 value is a fixed point
+    bits = np.arange(0, 1 << 16, dtype=np.uint32)
+    vals = (bits << 16).view(np.float32)
+    finite = np.isfinite(vals)
+    rt = encode_decode_bf16(vals[finite])
+    np.testing.assert_array_equal(rt, vals[finite])
+    # ties round to even mantissa: 1 + 2^-8 * 1.5 exactly between two bf16
+    x = np.float32(1.0 + 1.5 / 256.0)
+    got = encode_decode_bf16(np.array([x]))[0]
+    assert got in (np.float32(1.0 + 1.0 / 128.0),)  # even neighbor
+    # NaN stays NaN (quieted), infinities preserved
+    out = encode_decode_bf16(np.array([np.nan, np.inf, -np.inf],
+                                      dtype=np.float32))
+    assert np.isnan(out[0]) and np.isposinf(out[1]) and np.isneginf(out[2])
+    # monotone on finite inputs
+    rng = np.random.Generator(np.random.PCG64(2))
+    xs = np.sort(rng.standard_normal(4096).astype(np.float32) * 100)
+    ys = encode_decode_bf16(xs)
+    assert (np.diff(ys) >= 0).all()
+    zs = encode_decode_f16(np.clip(xs, -60000, 60000))
+    assert (np.diff(zs) >= 0).all()
