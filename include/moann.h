@@ -261,6 +261,31 @@ int moann_docfilter_eval(int kind, const void* payload, uint64_t payload_len,
                          const int64_t* keys, uint64_t nkeys,
                          uint32_t* out_words, void* errmsg);
 
+/* SQL residual-predicate filter (cuVS eval_filter_bitmap_cpu shape,
+ * cgo/cuvs/filter.hpp:855-960). preds_json is the planner's conjunctive
+ * predicate list ([{"col":N,"op":"<=","val":5.0}, {"col":-1,"op":"in",
+ * "vals":[...]}, ...]; col -1 = the virtual PK column __mo_pk_host_id);
+ * cols_json = [{"name":"a","type":T}] with T in {0 i32,1 i64,2 f32,
+ * 3 f64,4 u64}. col_data[i]/col_nulls[i] are the caller's columnar
+ * buffers (nulls: u32 words LSB-first, bit=1 null; null pointer = dense).
+ * Empty predicate list passes every row; NULL cells fail every value
+ * comparison; errors fail closed. out_words: (nrows+31)/32 u32, bit i =
+ * row i passes. */
+int moann_preds_eval(const char* cols_json, const char* preds_json,
+                     const void* const* col_data,
+                     const uint32_t* const* col_nulls,
+                     const int64_t* host_ids, uint64_t nrows,
+                     uint32_t* out_words, void* errmsg);
+
+/* Same, over an IVF index's entries: col arrays are in ADD (row) order and
+ * the bitset comes back in GLOBAL SLOT order for search_filtered. */
+int moann_ivf_flat_filter_from_preds(gpu_ivf_flat_c index_c,
+                                     const char* cols_json,
+                                     const char* preds_json,
+                                     const void* const* col_data,
+                                     const uint32_t* const* col_nulls,
+                                     uint32_t* out_words, void* errmsg);
+
 /* Same, over an IVF index's entries in GLOBAL SLOT order — the bitset
  * moann_ivf_flat_search_filtered consumes ((len(index)+31)/32 words). */
 int moann_ivf_flat_filter_from_docfilter(gpu_ivf_flat_c index_c, int kind,

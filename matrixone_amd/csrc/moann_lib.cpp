@@ -203,6 +203,7 @@ struct IvfIndex {
     SearchCtx sctx[2];
     uint64_t sparity = 0;
     std::vector<int64_t> h_id_by_slot;  /* kept for the membership filter */
+    std::vector<int64_t> h_slot_rows;   /* slot -> add-order row (pred filter) */
     std::unordered_map<int64_t, int64_t> id_to_slot; /* lazy */
 
     std::mutex mu;
@@ -1094,6 +1095,7 @@ void gpu_ivf_flat_build(gpu_ivf_flat_c h, void* errmsg) {
         HIP_CHECK(hipMemcpy(ix->d_id_by_slot, id_by_slot.data(),
                             id_by_slot.size() * 8, hipMemcpyHostToDevice));
         ix->h_id_by_slot = std::move(id_by_slot);
+        ix->h_slot_rows = slot_rows;
         HIP_CHECK(hipMalloc(&ix->d_list_slot_base, ix->nlist * 8));
         HIP_CHECK(hipMemcpy(ix->d_list_slot_base, ix->list_slot_base.data(),
                             ix->nlist * 8, hipMemcpyHostToDevice));
@@ -1346,6 +1348,41 @@ void gpu_ivf_flat_free_result(gpu_ivf_flat_result_c r) {
         (void)hipEventDestroy(res->done_ev);
     }
     delete res;
+}
+
+/* SQL residual-predicate filter -> slot bitset (the cuVS
+ * eval_filter_bitmap_cpu shape, filter.hpp:855-960). col_data/col_nulls are
+ * the caller's columnar arrays in ADD (row) order; the result is permuted
+ * to GLOBAL SLOT order for moann_ivf_flat_search_filtered. */
+int moann_ivf_flat_filter_from_preds(gpu_ivf_flat_c h, const char* cols_json,
+                                     const char* preds_json,
+                                     const void* const* col_data,
+                                     const uint32_t* const* col_nulls,
+                                     uint32_t* out_words, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (!ix->built) throw std::runtime_error("index not built");
+        const uint64_t n = ix->count;
+        std::vector<uint32_t> row_bits((n + 31) / 32, 0);
+        /* host ids in ADD (row) order for the virtual PK column */
+        std::vector<int64_t> ids_by_row(n);
+        for (uint64_t s2 = 0; s2 < n; ++s2)
+            ids_by_row[ix->h_slot_rows[s2]] = ix->h_id_by_slot[s2];
+        const int rc = moann_preds_eval(cols_json, preds_json, col_data,
+                                        col_nulls, ids_by_row.data(), n,
+                                        row_bits.data(), errmsg);
+        if (rc != 0) return rc;
+        std::memset(out_words, 0, ((n + 31) / 32) * 4);
+        for (uint64_t s2 = 0; s2 < n; ++s2) {
+            const uint64_t row = (uint64_t)ix->h_slot_rows[s2];
+            if ((row_bits[row >> 5] >> (row & 31)) & 1u)
+                out_words[s2 >> 5] |= 1u << (s2 & 31);
+        }
+        return 0;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "moann_ivf_flat_filter_from_preds", e.what());
+        return -1;
+    }
 }
 
 int moann_ivf_flat_filter_from_docfilter(gpu_ivf_flat_c h, int kind,

@@ -124,6 +124,15 @@ def lib() -> ct.CDLL:
     L.moann_ivf_flat_filter_from_docfilter.argtypes = [
         ct.c_void_p, ct.c_int, ct.c_void_p, ct.c_uint64, ct.c_void_p,
         ct.c_void_p]
+    L.moann_preds_eval.restype = ct.c_int
+    L.moann_preds_eval.argtypes = [
+        ct.c_char_p, ct.c_char_p, ct.POINTER(ct.c_void_p),
+        ct.POINTER(ct.c_void_p), ct.c_void_p, ct.c_uint64, ct.c_void_p,
+        ct.c_void_p]
+    L.moann_ivf_flat_filter_from_preds.restype = ct.c_int
+    L.moann_ivf_flat_filter_from_preds.argtypes = [
+        ct.c_void_p, ct.c_char_p, ct.c_char_p, ct.POINTER(ct.c_void_p),
+        ct.POINTER(ct.c_void_p), ct.c_void_p, ct.c_void_p]
     L.moann_ivf_flat_slots_of_ids.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
     L.moann_brute_force_search.argtypes = [
@@ -187,6 +196,59 @@ def device_count() -> int:
 def _as_f32(a) -> np.ndarray:
     arr = np.ascontiguousarray(a, dtype=np.float32)
     return arr
+
+
+COL_I32, COL_I64, COL_F32, COL_F64, COL_U64 = 0, 1, 2, 3, 4
+_COL_DTYPES = {0: np.int32, 1: np.int64, 2: np.float32, 3: np.float64,
+               4: np.uint64}
+
+
+def _col_arrays(cols_meta, columns, nulls):
+    """-> (cols_json bytes, void** data, void** nulls) keeping refs alive."""
+    import json as _json
+    cols_json = _json.dumps([{"name": n, "type": t}
+                             for n, t in cols_meta]).encode()
+    keep = []
+    data_arr = (ct.c_void_p * max(1, len(columns)))()
+    null_arr = (ct.c_void_p * max(1, len(columns)))()
+    for i, ((name, t), col) in enumerate(zip(cols_meta, columns)):
+        a = np.ascontiguousarray(col, dtype=_COL_DTYPES[t])
+        keep.append(a)
+        data_arr[i] = a.ctypes.data
+        nl = None if nulls is None else nulls[i]
+        if nl is not None:
+            w = np.ascontiguousarray(nl, dtype=np.uint32)
+            keep.append(w)
+            null_arr[i] = w.ctypes.data
+        else:
+            null_arr[i] = None
+    return cols_json, data_arr, null_arr, keep
+
+
+def preds_eval(cols_meta, columns, preds_json: str, nrows: int, nulls=None,
+               host_ids=None) -> np.ndarray:
+    """Evaluate a planner predicate list (cuVS eval_filter_bitmap_cpu
+    shape) over columnar arrays -> bool[nrows]. cols_meta: [(name, type)]
+    with type in COL_*; columns: array per column; nulls: optional u32
+    bitset words per column; host_ids: int64 array for the virtual PK
+    column (col == -1)."""
+    cols_json, data_arr, null_arr, keep = _col_arrays(cols_meta, columns,
+                                                      nulls)
+    out = np.zeros((nrows + 31) // 32, dtype=np.uint32)
+    hid = None
+    if host_ids is not None:
+        hid = np.ascontiguousarray(host_ids, dtype=np.int64)
+        keep.append(hid)
+    err = _Err()
+    rc = lib().moann_preds_eval(
+        cols_json, preds_json.encode(), data_arr, null_arr,
+        None if hid is None else hid.ctypes.data, nrows, out.ctypes.data,
+        err.ref)
+    err.check("preds_eval")
+    if rc != 0:
+        raise MoannError("preds_eval failed")
+    idx = np.arange(nrows)
+    return ((out[idx >> 5] >> (idx & 31)) & 1).astype(bool)
 
 
 FILTER_BLOOM = 0
@@ -389,6 +451,30 @@ class IvfFlatIndex:
         err = _Err()
         lib().moann_ivf_flat_enable_refine(self._h, depth, err.ref)
         err.check("enable_refine")
+
+    def preds_bitset(self, cols_meta, columns, preds_json: str,
+                     nulls=None) -> np.ndarray:
+        """Slot bitset from a planner predicate list over ADD-order column
+        arrays (moann_ivf_flat_filter_from_preds) — feed search_filtered."""
+        cols_json, data_arr, null_arr, keep = _col_arrays(cols_meta, columns,
+                                                          nulls)
+        n = len(self)
+        out = np.zeros(max(1, (n + 31) // 32), dtype=np.uint32)
+        err = _Err()
+        rc = lib().moann_ivf_flat_filter_from_preds(
+            self._h, cols_json, preds_json.encode(), data_arr, null_arr,
+            out.ctypes.data, err.ref)
+        err.check("preds_bitset")
+        if rc != 0:
+            raise MoannError("preds_bitset failed")
+        return out
+
+    def search_pred_filtered(self, queries, k: int, nprobe: int, cols_meta,
+                             columns, preds_json: str, nulls=None):
+        """Search gated by a SQL residual-predicate list (§8f3)."""
+        return self.search_filtered(
+            queries, k, nprobe,
+            self.preds_bitset(cols_meta, columns, preds_json, nulls))
 
     def docfilter_bitset(self, kind: int, payload: bytes) -> np.ndarray:
         """Slot bitset from a serialized docfilter: membership of every

@@ -687,3 +687,45 @@ def test_half_storage_odd_dim_and_saveload(tmp_path):
     np.testing.assert_allclose(d2, dists)
     ix.close()
     ix2.close()
+
+
+def test_pred_filtered_search():
+    """SQL residual-predicate gated search (§8f3): preds -> slot bitset
+    (ADD-order columns permuted to slots) == the same search with an
+    explicitly built id bitset."""
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(91))
+    n, d, nlist, nq, probe, k = 8000, 48, 16, 10, 5, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    ids64 = rng.permutation(n).astype(np.int64) + 11
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    price = rng.integers(0, 1000, n).astype(np.int64)  # ADD-order column
+
+    ix = engine.IvfFlatIndex(d, nlist, metric="l2sq", capacity=n)
+    ix.add(vecs, ids=ids64)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    preds = '[{"col":0,"op":"between","lo":100,"hi":400}]'
+    gids, gdists = ix.search_pred_filtered(
+        queries, k, probe, [("price", engine.COL_I64)], [price], preds)
+    allowed = ids64[(price >= 100) & (price <= 400)]
+    eids, edists = ix.search_filtered(queries, k, probe,
+                                     ix.filter_bitset(allowed))
+    np.testing.assert_array_equal(gids, eids)
+    np.testing.assert_allclose(gdists, edists)
+    # PK predicate composes with the column predicate
+    sel = set(int(x) for x in allowed[:50])
+    preds2 = ('[{"col":0,"op":">=","val":100},'
+              '{"col":0,"op":"<=","val":400},'
+              '{"col":-1,"op":"in","vals":[%s]}]'
+              % ",".join(str(x) for x in sorted(sel)))
+    g2, d2 = ix.search_pred_filtered(
+        queries, k, probe, [("price", engine.COL_I64)], [price], preds2)
+    e2, ed2 = ix.search_filtered(queries, k, probe,
+                                 ix.filter_bitset(np.array(sorted(sel))))
+    np.testing.assert_array_equal(g2, e2)
+    ix.close()
