@@ -90,19 +90,19 @@ struct Rd {
         ws();
         const size_t start = i;
         if (i < s.size() && (s[i] == '-' || s[i] == '+')) ++i;
-        bool dig = false, flt = false;
+        bool dig = false, flt = false, dot = false, ex = false;
         while (i < s.size()) {
             const char c = s[i];
             if (c >= '0' && c <= '9') {
                 dig = true;
                 ++i;
-            } else if (c == '.' || c == 'e' || c == 'E' || c == '-' ||
-                       c == '+') {
-                if (c == '.' || c == 'e' || c == 'E') flt = true;
-                if ((c == '-' || c == '+') &&
-                    !(s[i - 1] == 'e' || s[i - 1] == 'E'))
-                    break;
+            } else if (c == '.' && !dot && !ex) {
+                dot = flt = true;
                 ++i;
+            } else if ((c == 'e' || c == 'E') && !ex && dig) {
+                ex = flt = true;
+                ++i;
+                if (i < s.size() && (s[i] == '-' || s[i] == '+')) ++i;
             } else
                 break;
         }
