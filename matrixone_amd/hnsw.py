@@ -78,6 +78,111 @@ class HnswGraphData:
     adj: list            # per level: (offsets[n+1] int64, neighbors u32)
 
 
+def load_usearch_index(data) -> "HnswGraphData":
+    """Parse a usearch-2.x index_dense serialization (the file format
+    MatrixOne persists per HNSW mini-model — hnsw/model.go chunks exactly
+    these bytes into the model table) into HnswGraphData, natively.
+
+    Layout (usearch index_dense.hpp save_to_stream:1143-1209 +
+    index.hpp save_to_stream:3703-3748, studied not copied):
+      [u32 rows][u32 bytes_per_vector][rows x bpv vector bytes]
+      [64 B dense head: "usearch", 3 x u16 version, 4 x u8 kinds,
+       u64 count_present, u64 count_deleted, u64 dimensions, u8 multi]
+      [index header: 5 x u64 size/connectivity/connectivity_base/
+       max_level/entry_slot]
+      [size x i16 node levels]
+      [per node: u64 key, i16 level, then level-0 block
+       (u32 count + connectivity_base x u32 slots) and `level` blocks of
+       (u32 count + connectivity x u32 slots)]
+    Scalar kinds f32/f16/bf16 are decoded to f32 (the metric evaluates in
+    f32 either way on our kernels). `data` is a path or bytes."""
+    if isinstance(data, (str, bytes)) and isinstance(data, str):
+        with open(data, "rb") as f:
+            raw = np.frombuffer(f.read(), dtype=np.uint8)
+    else:
+        raw = np.frombuffer(data, dtype=np.uint8)
+    off = 0
+    rows, bpv = raw[off:off + 8].view(np.uint32)
+    off += 8
+    vec_bytes = raw[off:off + rows * bpv]
+    off += int(rows) * int(bpv)
+    head = raw[off:off + 64]
+    off += 64
+    if bytes(head[:7]) != b"usearch":
+        raise ValueError("not a usearch index file")
+    kind_metric = int(head[13])
+    kind_scalar = int(head[14])
+    dims = int(head[33:41].view(np.uint64)[0])
+    metric = {ord("e"): "l2sq", ord("i"): "ip", ord("c"): "cos"}.get(
+        kind_metric)
+    if metric is None:
+        raise ValueError(f"unsupported usearch metric kind {kind_metric}")
+    hdr = raw[off:off + 40].view(np.uint64)
+    off += 40
+    size, conn, conn_base, max_level, entry_slot = (int(x) for x in hdr)
+    if size != rows:
+        raise ValueError("vector matrix and graph size disagree "
+                         f"({rows} vs {size})")
+    levels = raw[off:off + size * 2].view(np.int16).astype(np.int32)
+    off += size * 2
+
+    # vectors -> f32 [size][dims]
+    if kind_scalar == 11:    # f32
+        vecs = vec_bytes.view(np.float32).reshape(size, dims).copy()
+    elif kind_scalar == 12:  # f16
+        vecs = (vec_bytes.view(np.float16).reshape(size, dims)
+                .astype(np.float32))
+    elif kind_scalar == 4:   # bf16
+        u = vec_bytes.view(np.uint16).reshape(size, dims)
+        vecs = (u.astype(np.uint32) << 16).view(np.float32).copy()
+    else:
+        raise ValueError(f"unsupported usearch scalar kind {kind_scalar}")
+
+    # node offsets from the levels alone (fixed per-level block sizes)
+    base_block = 4 + 4 * conn_base
+    lvl_block = 4 + 4 * conn
+    node_bytes = 10 + base_block + levels.astype(np.int64) * lvl_block
+    node_off = np.zeros(size + 1, dtype=np.int64)
+    np.cumsum(node_bytes, out=node_off[1:])
+    node_off += off
+    if int(node_off[-1]) > raw.size:
+        raise ValueError("truncated usearch index file")
+
+    keys = np.empty(size, dtype=np.uint64)
+    k8 = node_off[:-1, None] + np.arange(8)
+    keys[:] = raw[k8].reshape(size, 8).copy().view(np.uint64).ravel()
+
+    adj = []
+    for lvl in range(max_level + 1):
+        has = np.nonzero(levels >= lvl)[0]
+        offs = np.zeros(size + 1, dtype=np.int64)
+        if len(has) == 0:
+            adj.append((offs, np.zeros(0, dtype=np.uint32)))
+            continue
+        block = base_block if lvl == 0 else lvl_block
+        boff = node_off[has] + 10 + (0 if lvl == 0 else
+                                     base_block + (lvl - 1) * lvl_block)
+        blocks = raw[boff[:, None] + np.arange(block)].reshape(len(has),
+                                                               block)
+        w = blocks.copy().view(np.uint32).reshape(len(has), block // 4)
+        counts = w[:, 0].astype(np.int64)
+        cap = block // 4 - 1
+        if (counts > cap).any():
+            raise ValueError("corrupt neighbor count in usearch file")
+        per = np.zeros(size, dtype=np.int64)
+        per[has] = counts
+        np.cumsum(per, out=offs[1:])
+        mask = np.arange(cap)[None, :] < counts[:, None]
+        nbrs = w[:, 1:][mask].astype(np.uint32)
+        if nbrs.size and int(nbrs.max()) >= size:
+            raise ValueError("corrupt neighbor slot in usearch file")
+        adj.append((offs, nbrs))
+
+    return HnswGraphData(dim=dims, metric=metric, entry_slot=entry_slot,
+                         max_level=max_level, levels=levels, keys=keys,
+                         vecs=vecs, adj=adj)
+
+
 class HnswIndex:
     """One HNSW model on one MI355X (ingested usearch graph)."""
 
@@ -151,6 +256,12 @@ class HnswIndex:
                             out_dists.ctypes.data, err.ref)
         err.check("hnsw_search")
         return out_ids, out_dists
+
+    @classmethod
+    def from_usearch_file(cls, path_or_bytes):
+        """Build the GPU index directly from a usearch-serialized
+        mini-model (the bytes MO's hnsw/model.go persists)."""
+        return cls(load_usearch_index(path_or_bytes))
 
     def search_docfiltered(self, queries, ef: int, k: int, kind: int,
                            payload: bytes):

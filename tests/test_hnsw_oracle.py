@@ -95,3 +95,50 @@ def test_filtered_oracle_matches_usearch():
     assert pos >= 118, pos
     valid = ref_ids[ref_ids >= 0]
     assert all(int(x) in aset for x in valid)
+
+
+def test_native_usearch_file_loader():
+    """The PRODUCT's usearch-file parser (matrixone_amd.hnsw.
+    load_usearch_index — the bytes MO's hnsw/model.go persists) must
+    reproduce exactly the graph the reference itself reports for the same
+    saved file (shim export_graph): keys, levels, vectors, adjacency,
+    entry point."""
+    import tempfile
+
+    from matrixone_amd.hnsw import load_usearch_index
+
+    for metric, mname in [(H.METRIC_L2SQ, "l2sq"), (H.METRIC_COS, "cos"),
+                          (H.METRIC_IP, "ip")]:
+        ix, _ = _mk(metric, n=3000, d=40, conn=12, seed=metric + 2)
+        g = ix.export_graph()
+        with tempfile.NamedTemporaryFile(suffix=".usearch") as f:
+            ix.save(f.name)
+            mine = load_usearch_index(f.name)
+        assert mine.metric == mname
+        assert mine.dim == g.dim
+        assert mine.entry_slot == g.entry_slot
+        assert mine.max_level == g.max_level
+        np.testing.assert_array_equal(mine.levels, g.levels)
+        np.testing.assert_array_equal(mine.keys.astype(np.uint64),
+                                      g.keys.astype(np.uint64))
+        np.testing.assert_allclose(mine.vecs, g.vecs)
+        assert len(mine.adj) == len(g.adj)
+        for lvl, ((mo, mn), (ro, rn)) in enumerate(zip(mine.adj, g.adj)):
+            np.testing.assert_array_equal(mo, ro, err_msg=f"offs l{lvl}")
+            np.testing.assert_array_equal(mn, rn, err_msg=f"nbrs l{lvl}")
+
+
+def test_native_loader_from_bytes():
+    """Same parse from an in-memory byte blob (model chunks concatenated)."""
+    import tempfile
+
+    from matrixone_amd.hnsw import load_usearch_index
+
+    ix, _ = _mk(H.METRIC_L2SQ, n=800, d=24, conn=8, seed=77)
+    with tempfile.NamedTemporaryFile(suffix=".usearch") as f:
+        ix.save(f.name)
+        blob = open(f.name, "rb").read()
+    g = ix.export_graph()
+    mine = load_usearch_index(blob)
+    np.testing.assert_array_equal(mine.levels, g.levels)
+    np.testing.assert_allclose(mine.vecs, g.vecs)
