@@ -101,9 +101,10 @@ def test_croaring_container_shapes_live():
             [rng.choice(5000, 500, replace=False) + (np.int64(b) << 32)
              for b in (0, 1, 7, 123)]),
     }
+    sets["no-runopt"] = np.arange(8000, 60000, 3)  # cookie 12346 + offsets
     for name, kk in sets.items():
         kk = kk.astype(np.int64)
-        pay = DF.build_croaring(kk)
+        pay = DF.build_croaring(kk, run_optimize=(name != "no-runopt"))
         probe = np.concatenate([kk[::3], kk + 1,
                                 rng.integers(0, 2**40, 3000)]).astype(np.int64)
         mine = engine.docfilter_eval(engine.FILTER_CROARING, pay, probe)
