@@ -180,10 +180,14 @@ def main():
     ap.add_argument("--recall-queries", type=int, default=256)
     ap.add_argument("--cpu-baseline-queries", type=int, default=48)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--refine", type=int, default=0,
+    ap.add_argument("--refine", type=int, default=128,
                     help="two-stage scan: byte first pass + exact f32 "
                          "re-rank of the top-R (0 = exact one-pass scan). "
-                         "recall@10 is still gated at 0.95 either way")
+                         "ADAPTIVE: if the measured recall@k gate (0.95) "
+                         "fails with refine on, the bench falls back to "
+                         "the exact one-pass scan and re-measures — the "
+                         "reported configuration always satisfies the "
+                         "gate if the exact path does")
     ap.add_argument("--qtype", default="f32",
                     choices=["f32", "f16", "bf16", "int8", "uint8"],
                     help="storage type; the BASELINE metric is f32 — narrow "
@@ -243,15 +247,22 @@ def main():
         hi = float(torch.quantile(sample, 0.999))
         ix.set_quantizer(lo, hi)
     ix.build()
-    if args.refine and args.qtype == "f32":
+    if args.refine and args.qtype == "f32" and not distributed:
         ix.enable_refine(args.refine)
+    else:
+        args.refine = 0
     log(f"[bench] build done in {time.perf_counter() - t_build0:.1f}s")
 
     # ---- recall check ------------------------------------------------------
     nrq = args.recall_queries
     gt_local = exact_gt(data, queries[:nrq], args.k)  # local-shard GT
-    ids_r, dists_r = ix.search_device(queries[:nrq].contiguous(), args.k,
-                                      args.nprobe)
+
+    def measure_recall():
+        ids_rr, dists_rr = ix.search_device(queries[:nrq].contiguous(),
+                                            args.k, args.nprobe)
+        return ids_rr, dists_rr
+
+    ids_r, dists_r = measure_recall()
     if not distributed:
         gt_ids = gt_local.cpu().numpy()
         recall = float(np.mean([
@@ -274,6 +285,21 @@ def main():
             len(set(m_ids[i]).intersection(g_ids[i])) / args.k
             for i in range(nrq)]))
     log(f"[bench] recall@{args.k} = {recall:.4f} (target >= 0.95)")
+    if (recall < 0.95 and args.refine and args.qtype == "f32"
+            and not distributed):
+        # two-stage scan missed the gate on this data: fall back to the
+        # exact one-pass scan and re-measure (the reported config is
+        # always the one that was actually benched)
+        log("[bench] refine missed the recall gate — falling back to the "
+            "exact one-pass scan")
+        ix.enable_refine(0)
+        args.refine = 0
+        ids_r, dists_r = measure_recall()
+        gt_ids = gt_local.cpu().numpy()
+        recall = float(np.mean([
+            len(set(ids_r[i]).intersection(gt_ids[i])) / args.k
+            for i in range(nrq)]))
+        log(f"[bench] recall@{args.k} = {recall:.4f} (exact path)")
 
     # ---- timed region ------------------------------------------------------
     nbatches = max(1, queries.shape[0] // args.batch)
@@ -409,6 +435,8 @@ def main():
                 "global_batch": args.batch,
                 "k": args.k,
                 "recall_at_k": round(recall, 4),
+                "scan": (f"two-stage-refine-R{args.refine}" if args.refine
+                         else "exact-one-pass"),
                 "parallelism": f"sharded-lists x{world}" if world > 1
                                else "single",
             },
