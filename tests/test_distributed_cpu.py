@@ -91,3 +91,28 @@ def test_sharded_allgather_merge_gloo(tmp_path):
     port = 29511
     mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
     assert os.path.exists(tmp_path / "ok.npy")
+
+
+def test_merge_edge_cases():
+    """Empty shard (all sentinels), duplicate ids across shards, k larger
+    than live entries — cpu_topk_merge_sharded edge semantics."""
+    fmax = np.finfo(np.float32).max
+    k = 6
+    a_i = np.array([[[5, 7, -1, -1, -1, -1]]], dtype=np.int64)
+    a_d = np.array([[[0.5, 0.9, fmax, fmax, fmax, fmax]]], dtype=np.float32)
+    empty_i = np.full((1, 1, k), -1, dtype=np.int64)
+    empty_d = np.full((1, 1, k), fmax, dtype=np.float32)
+    ids = np.concatenate([a_i, empty_i])
+    dd = np.concatenate([a_d, empty_d])
+    out_i, out_d = merge_topk_torch(torch.from_numpy(ids),
+                                    torch.from_numpy(dd), k)
+    np.testing.assert_array_equal(out_i.numpy()[0, :2], [5, 7])
+    assert (out_i.numpy()[0, 2:] == -1).all()
+    assert (out_d.numpy()[0, 2:] == fmax).all()
+    # duplicate id in two shards keeps both occurrences ordered by dist
+    b_i = np.array([[[5, 9, -1, -1, -1, -1]]], dtype=np.int64)
+    b_d = np.array([[[0.4, 1.2, fmax, fmax, fmax, fmax]]], dtype=np.float32)
+    out_i, out_d = merge_topk_torch(
+        torch.from_numpy(np.concatenate([a_i, b_i])),
+        torch.from_numpy(np.concatenate([a_d, b_d])), k)
+    np.testing.assert_array_equal(out_i.numpy()[0, :4], [5, 5, 7, 9])
