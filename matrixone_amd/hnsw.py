@@ -411,3 +411,48 @@ class MultiModelHnswSearch:
     def Destroy(self):
         for sch in self.searchers:
             sch.Destroy()
+
+
+class ShardedHnswSearch:
+    """Multi-GPU HNSW model sharding (HNSW side): the
+    reference already searches disjoint mini-models concurrently on one
+    host (hnsw/search.go:140-160); here each RANK owns the models assigned
+    round-robin to it (one process per GPU over torch.distributed) and one
+    RCCL all-gather of the per-rank top-k blocks merges the answer —
+    exactly the IVF exchange (matrixone_amd/distributed.py), since model
+    results carry globally-unique keys.
+
+    Construction: pass the FULL model list; rank r keeps models
+    [r::world]. Ranks with no models still participate in the collective
+    with sentinel blocks."""
+
+    def __init__(self, graphs, op_type: str = "vector_l2_ops",
+                 ef_search: int = 64, device: int = 0, rank: int = 0,
+                 world: int = 1):
+        self.rank, self.world = rank, world
+        mine = list(graphs)[rank::world]
+        self.local = (MultiModelHnswSearch(mine, op_type, ef_search, device)
+                      if mine else None)
+
+    def Load(self, proc=None):
+        if self.local:
+            self.local.Load()
+
+    def Search(self, proc, query, rt):
+        k = rt.limit
+        q = np.atleast_2d(np.asarray(query, dtype=np.float32))
+        nq = q.shape[0]
+        if self.local:
+            ids, d64 = self.local.Search(proc, q, rt)
+        else:
+            ids = np.full((nq, k), -1, dtype=np.int64)
+            d64 = np.full((nq, k), np.float64(np.finfo(np.float32).max))
+        if self.world == 1:
+            return ids, d64
+        from matrixone_amd.distributed import allgather_merge
+        m_ids, m_d = allgather_merge(ids, d64.astype(np.float32), k)
+        return m_ids, m_d.astype(np.float64)
+
+    def Destroy(self):
+        if self.local:
+            self.local.Destroy()

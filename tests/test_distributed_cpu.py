@@ -116,3 +116,78 @@ def test_merge_edge_cases():
         torch.from_numpy(np.concatenate([a_i, b_i])),
         torch.from_numpy(np.concatenate([a_d, b_d])), k)
     np.testing.assert_array_equal(out_i.numpy()[0, :4], [5, 5, 7, 9])
+
+
+def _hnsw_shard_worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from oracle import hnsw as H
+        from matrixone_amd.hnsw import ShardedHnswSearch
+        from matrixone_amd.ivfflat import RuntimeConfig
+
+        rng = np.random.Generator(np.random.PCG64(7))
+        d, k = 32, 8
+        models = []
+        for m in range(3):  # 3 models over 2 ranks (uneven round-robin)
+            vecs = rng.standard_normal((1200, d), dtype=np.float32)
+            keys = np.arange(1200, dtype=np.uint64) + m * 10_000
+            ix = H.RefHnsw(d, metric=H.METRIC_L2SQ, connectivity=8,
+                           expansion_add=32, expansion_search=32,
+                           capacity=1200, threads=1)
+            ix.add(vecs, keys=keys)
+            models.append(ix)
+        queries = rng.standard_normal((6, d), dtype=np.float32)
+
+        class _Stub:  # rank-local models searched by the REFERENCE engine
+            def __init__(self, mine):
+                self.mine = mine
+
+            def Search(self, proc, q, rt):
+                bi, bd = [], []
+                for ix in self.mine:
+                    ri, rd = ix.search(q, ef=32, k=rt.limit)
+                    bi.append(ri)
+                    bd.append(rd.astype(np.float64))
+                ids = np.stack(bi)
+                dd = np.stack(bd)
+                nm, nq, kk = ids.shape
+                fi = ids.transpose(1, 0, 2).reshape(nq, nm * kk)
+                fd = dd.transpose(1, 0, 2).reshape(nq, nm * kk)
+                order = np.argsort(fd, axis=1, kind="stable")[:, :rt.limit]
+                return (np.take_along_axis(fi, order, 1),
+                        np.take_along_axis(fd, order, 1))
+
+        sh = ShardedHnswSearch([], rank=rank, world=world)
+        mine = models[rank::world]
+        sh.local = _Stub(mine) if mine else None
+        rt = RuntimeConfig(limit=k)
+        ids, d64 = sh.Search(None, queries, rt)
+
+        # expectation: merge across ALL models (deterministic on all ranks)
+        allb_i, allb_d = [], []
+        for ix in models:
+            ri, rd = ix.search(queries, ef=32, k=k)
+            allb_i.append(ri)
+            allb_d.append(rd.astype(np.float32))
+        exp_i, exp_d = orc.topk_merge(np.stack(allb_i), np.stack(allb_d), k)
+        mism = ids != exp_i
+        assert (np.abs(d64[mism] - exp_d[mism].astype(np.float64))
+                < 1e-5).all()
+        np.testing.assert_allclose(d64, exp_d.astype(np.float64),
+                                   rtol=2e-5, atol=2e-5)
+        if rank == 0:
+            open(os.path.join(tmpdir, "hnsw_ok"), "w").write("1")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_hnsw_gloo(tmp_path):
+    """§8e HNSW side: models round-robined over 2 ranks, all-gather merge
+    equals the whole-model-set merge (rank-local search = the reference
+    usearch engine; the collective path is the one the GPU uses)."""
+    port = 29741 + os.getpid() % 500
+    mp.spawn(_hnsw_shard_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    assert (tmp_path / "hnsw_ok").exists()
