@@ -195,19 +195,3 @@ def test_multi_model_fanout():
     assert agree > 0.95, agree
     np.testing.assert_allclose(d64, md.astype(np.float64), rtol=3e-5, atol=3e-5)
     mm.Destroy()
-
-
-def test_gpu_from_usearch_file(tmp_path):
-    """End-to-end: MO-persisted usearch bytes -> native loader -> GPU beam
-    search == usearch's own results on the same file."""
-    from matrixone_amd.hnsw import HnswIndex
-    ix, qs = _mk(H.METRIC_COS, n=4000, d=64)
-    p = str(tmp_path / "m.usearch")
-    ix.save(p)
-    gpu = HnswIndex.from_usearch_file(p)
-    gids, gdists = gpu.search(qs, 48, 10)
-    ref_ids, ref_d = ix.search(qs, ef=48, k=10)
-    assert (gids == ref_ids).mean() > 0.95
-    np.testing.assert_allclose(np.sort(gdists, 1)[:, 0],
-                               np.sort(ref_d, 1)[:, 0], rtol=2e-5, atol=2e-5)
-    gpu.close()
