@@ -117,6 +117,12 @@ def load_usearch_index(data) -> "HnswGraphData":
         kind_metric)
     if metric is None:
         raise ValueError(f"unsupported usearch metric kind {kind_metric}")
+    count_deleted = int(head[25:33].view(np.uint64)[0])
+    if count_deleted:
+        # MO's sync.go rebuilds models rather than deleting in place; a
+        # file with soft-deleted members would surface them as live here
+        raise ValueError("usearch index holds deleted entries — rebuild/"
+                         "compact the model before ingestion")
     hdr = raw[off:off + 40].view(np.uint64)
     off += 40
     size, conn, conn_base, max_level, entry_slot = (int(x) for x in hdr)

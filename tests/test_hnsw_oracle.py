@@ -142,3 +142,21 @@ def test_native_loader_from_bytes():
     mine = load_usearch_index(blob)
     np.testing.assert_array_equal(mine.levels, g.levels)
     np.testing.assert_allclose(mine.vecs, g.vecs)
+
+
+def test_native_loader_rejects_garbage():
+    import tempfile
+
+    import pytest as _pytest
+
+    from matrixone_amd.hnsw import load_usearch_index
+
+    with _pytest.raises(ValueError):
+        load_usearch_index(b"\x00" * 256)
+    # truncated real file
+    ix, _ = _mk(H.METRIC_L2SQ, n=500, d=16, conn=8, seed=3)
+    with tempfile.NamedTemporaryFile(suffix=".usearch") as f:
+        ix.save(f.name)
+        blob = open(f.name, "rb").read()
+    with _pytest.raises(Exception):
+        load_usearch_index(blob[: len(blob) // 2])
