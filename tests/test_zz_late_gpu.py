@@ -12,9 +12,26 @@ import pytest
 
 from oracle import oracle as orc
 from oracle import hnsw as H
-from tests.test_parity_gpu import _assert_parity
 
 pytestmark = pytest.mark.gpu
+
+RTOL = 1e-5
+
+
+def _assert_parity(ids, dists, ref_ids, ref_dists, ctx=""):
+    ids = np.asarray(ids)
+    ref_ids = np.asarray(ref_ids)
+    d = np.asarray(dists, dtype=np.float64)
+    rd = np.asarray(ref_dists, dtype=np.float64)
+    np.testing.assert_allclose(d, rd, rtol=RTOL, atol=1e-6, err_msg=ctx)
+    mism = ids != ref_ids
+    if mism.any():
+        da, db = d[mism], rd[mism]
+        rel = np.abs(da - db) / np.maximum(
+            np.maximum(np.abs(da), np.abs(db)), 1e-30)
+        assert (rel < RTOL).all(), (
+            f"{ctx}: {mism.sum()} id mismatches beyond FP ties "
+            f"(worst rel dist err {rel.max():.2e})")
 
 
 def _mk(metric, n=6000, d=96, conn=16, seed=11, nq=24):
