@@ -625,14 +625,15 @@ def test_search_with_tail():
 
 
 @pytest.mark.parametrize("qtype,metric", [
-    ("f16", "l2sq"), ("f16", "cos"), ("bf16", "l2sq"), ("bf16", "ip")])
+    ("f16", "l2sq"), ("f16", "cos"), ("f16", "ip"), ("f16", "l1"),
+    ("bf16", "l2sq"), ("bf16", "ip"), ("bf16", "cos")])
 def test_half_storage(qtype, metric):
     """f16/bf16 plain-cast storage (quantizer.go:50-58): the HIP half scan
     == the oracle run over RTNE-decoded entries and query
     (relation_search.go:56-63, distance_func_narrow.go:27-52)."""
     from matrixone_amd import engine
     mmap = {"l2sq": orc.METRIC_L2SQ, "ip": orc.METRIC_IP,
-            "cos": orc.METRIC_COS}
+            "cos": orc.METRIC_COS, "l1": orc.METRIC_L1}
     rng = np.random.Generator(np.random.PCG64(1600 + len(qtype)))
     n, d, nlist, nq, probe, k = 12000, 96, 32, 12, 6, 10
     vecs = rng.standard_normal((n, d), dtype=np.float32)
