@@ -345,6 +345,15 @@ gpu_ivf_flat_search_res_t moann_ivf_pq_search(moann_ivf_pq_c h,
                                               uint32_t limit,
                                               ivf_flat_search_params_t search_params,
                                               void* errmsg);
+/* membership-filtered PQ search: same bitset contract as
+ * moann_ivf_flat_search_filtered (bit per GLOBAL slot, cleared = excluded;
+ * the ADC scan substitutes FLT_MAX inside the kernel). */
+gpu_ivf_flat_search_res_t moann_ivf_pq_search_filtered(
+    moann_ivf_pq_c h, const void* queries, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, const uint32_t* filter_bitset,
+    uint64_t bitset_words, void* errmsg);
+
 void moann_ivf_pq_perf(moann_ivf_pq_c h, moann_perf_t* out);
 void moann_ivf_pq_destroy(moann_ivf_pq_c h, void* errmsg);
 

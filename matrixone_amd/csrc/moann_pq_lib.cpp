@@ -31,7 +31,8 @@ void launch_pq_scan_q(const uint8_t* codes, const float* atab,
                       const float* csel_dists, int probe,
                       const int64_t* qoffs, const int64_t* probe_offs,
                       const int64_t* list_gbase, const int32_t* list_rows,
-                      const int64_t* list_slot_base, int nsub, int nq,
+                      const int64_t* list_slot_base,
+                      const uint32_t* filter_bitset, int nsub, int nq,
                       float* dists_out, hipStream_t stream);
 void launch_pq_rowb(bool w16, const uint8_t* codes, const float* btab,
                     const int32_t* g2list, const int64_t* g2slot,
@@ -83,7 +84,7 @@ struct PqIndex {
 
     MoannDevBuf w_queries, w_cdists, w_csel_slots, w_csel_dists, w_jobs;
     MoannDevBuf w_cand, w_sel_slots, w_sel_dists, w_probe_lists, w_probe_offs;
-    MoannDevBuf w_qoffs, w_out_ids, w_out_dists, w_atab, w_qn;
+    MoannDevBuf w_qoffs, w_out_ids, w_out_dists, w_atab, w_qn, w_filter;
 
     std::mutex mu;
     moann_perf_t perf {};
@@ -108,7 +109,9 @@ PqIndex* PX(moann_ivf_pq_c h) { return (PqIndex*)h; }
 
 std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
                                        bool on_device, uint64_t nq,
-                                       uint32_t limit, uint32_t n_probes) {
+                                       uint32_t limit, uint32_t n_probes,
+                                       const uint32_t* filter_words
+                                       = nullptr) {
     std::lock_guard<std::mutex> lk(ix->mu);
     MOANN_HIP_CHECK(hipSetDevice(ix->device));
     auto res = std::make_unique<MoannResult>();
@@ -253,6 +256,15 @@ std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
         ix->w_qn.ensure(nq * 4);
         launch_qnorms(d_q, (int)nq, dpad, ix->w_qn.as<float>(), s);
         MOANN_KCHECK("pq-qnorms");
+        const uint32_t* d_filter = nullptr;
+        if (filter_words) {
+            const size_t fw = (size_t)((ix->count + 31) / 32);
+            ix->w_filter.ensure(fw * 4);
+            MOANN_HIP_CHECK(hipMemcpyAsync(ix->w_filter.ptr, filter_words,
+                                           fw * 4, hipMemcpyHostToDevice,
+                                           s));
+            d_filter = ix->w_filter.as<uint32_t>();
+        }
         MOANN_HIP_CHECK(hipEventRecord(ix->ev[0], s));
         launch_pq_scan_q(ix->d_codes, ix->w_atab.as<float>(), ix->d_rowb,
                          ix->w_qn.as<float>(), ix->d_cnorm2,
@@ -260,7 +272,7 @@ std::unique_ptr<MoannResult> pq_search(PqIndex* ix, const float* queries,
                          ix->w_csel_dists.as<float>(), (int)probe,
                          ix->w_qoffs.as<int64_t>(),
                          ix->w_probe_offs.as<int64_t>(), ix->d_gbase_dev,
-                         ix->d_rows_dev, ix->d_list_slot_base,
+                         ix->d_rows_dev, ix->d_list_slot_base, d_filter,
                          (int)ix->nsub, (int)nq, ix->w_cand.as<float>(), s);
         MOANN_KCHECK("pq-scan");
         MOANN_HIP_CHECK(hipEventRecord(ix->ev[1], s));
@@ -629,6 +641,28 @@ gpu_ivf_flat_search_res_t moann_ivf_pq_search(moann_ivf_pq_c h,
                              .release();
     } catch (const std::exception& e) {
         moann_set_errmsg(errmsg, "moann_ivf_pq_search", e.what());
+    }
+    return out;
+}
+
+gpu_ivf_flat_search_res_t moann_ivf_pq_search_filtered(
+    moann_ivf_pq_c h, const void* queries, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t sp, const uint32_t* filter_bitset,
+    uint64_t bitset_words, void* errmsg) {
+    gpu_ivf_flat_search_res_t out {nullptr};
+    try {
+        auto ix = PX(h);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        if (bitset_words < (ix->count + 31) / 32)
+            throw std::runtime_error("filter bitset too small");
+        out.result_ptr = pq_search(ix, (const float*)queries, false,
+                                   num_queries, limit, sp.n_probes,
+                                   filter_bitset)
+                             .release();
+    } catch (const std::exception& e) {
+        moann_set_errmsg(errmsg, "moann_ivf_pq_search_filtered", e.what());
     }
     return out;
 }

@@ -161,6 +161,8 @@ __global__ __launch_bounds__(1024) void pq_scan_q_kernel(
     const int64_t* __restrict__ list_gbase,
     const int32_t* __restrict__ list_rows,
     const int64_t* __restrict__ list_slot_base,
+    const uint32_t* __restrict__ filter_bitset, /* bit per GLOBAL slot;
+        cleared bit excludes (same contract as the flat scans); null = none */
     int nsub, float* __restrict__ dists_out) {
     extern __shared__ float lut[]; /* [nsub*256] */
     const int q = blockIdx.x;
@@ -214,8 +216,14 @@ __global__ __launch_bounds__(1024) void pq_scan_q_kernel(
                 }
             }
             const int row = g * 64 + lane;
-            if (row < rows)
-                dists_out[ob + row] = acc + row_b[sbase + row] + dconst;
+            if (row < rows) {
+                const int64_t gs = sbase + row;
+                const bool pass =
+                    !filter_bitset ||
+                    ((filter_bitset[gs >> 5] >> (gs & 31)) & 1u);
+                dists_out[ob + row] =
+                    pass ? acc + row_b[gs] + dconst : FLT_MAX;
+            }
         }
     }
 }
@@ -226,7 +234,8 @@ void launch_pq_scan_q(const uint8_t* codes, const float* atab,
                       const float* csel_dists, int probe,
                       const int64_t* qoffs, const int64_t* probe_offs,
                       const int64_t* list_gbase, const int32_t* list_rows,
-                      const int64_t* list_slot_base, int nsub, int nq,
+                      const int64_t* list_slot_base,
+                      const uint32_t* filter_bitset, int nsub, int nq,
                       float* dists_out, hipStream_t stream) {
     if (!nq) return;
     const size_t shmem = (size_t)nsub * 256 * 4;
@@ -234,14 +243,14 @@ void launch_pq_scan_q(const uint8_t* codes, const float* atab,
         hipLaunchKernelGGL(pq_scan_q_kernel<16>, dim3(nq), dim3(1024), shmem,
                            stream, codes, atab, row_b, qnorms, cnorm2,
                            csel_slots, csel_dists, probe, qoffs, probe_offs,
-                           list_gbase, list_rows, list_slot_base, nsub,
-                           dists_out);
+                           list_gbase, list_rows, list_slot_base,
+                           filter_bitset, nsub, dists_out);
     else
         hipLaunchKernelGGL(pq_scan_q_kernel<4>, dim3(nq), dim3(1024), shmem,
                            stream, codes, atab, row_b, qnorms, cnorm2,
                            csel_slots, csel_dists, probe, qoffs, probe_offs,
-                           list_gbase, list_rows, list_slot_base, nsub,
-                           dists_out);
+                           list_gbase, list_rows, list_slot_base,
+                           filter_bitset, nsub, dists_out);
 }
 
 /* pack codes: thread per output uchar4 */

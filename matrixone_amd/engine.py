@@ -153,6 +153,10 @@ def lib() -> ct.CDLL:
                                                ct.c_uint64, ct.c_void_p]
     L.moann_ivf_pq_build.argtypes = [ct.c_void_p, ct.c_void_p]
     L.moann_ivf_pq_search.restype = SearchRes
+    L.moann_ivf_pq_search_filtered.restype = SearchRes
+    L.moann_ivf_pq_search_filtered.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
+        SearchParams, ct.c_void_p, ct.c_uint64, ct.c_void_p]
     L.moann_ivf_pq_search.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_uint64,
                                       ct.c_uint32, ct.c_uint32, SearchParams,
                                       ct.c_void_p]
@@ -665,6 +669,28 @@ class IvfPqIndex:
         err = _Err()
         lib().moann_ivf_pq_build(self._h, err.ref)
         err.check("pq_build")
+
+    def search_filtered(self, queries, k: int, nprobe: int, bitset_words):
+        """Membership-filtered ADC search (same slot-bitset contract as
+        IvfFlatIndex.search_filtered)."""
+        queries = _as_f32(queries)
+        bits = np.ascontiguousarray(bitset_words, dtype=np.uint32)
+        nq = queries.shape[0]
+        err = _Err()
+        res = lib().moann_ivf_pq_search_filtered(
+            self._h, queries.ctypes.data, nq, queries.shape[1], k,
+            SearchParams(nprobe), bits.ctypes.data, len(bits), err.ref)
+        err.check("pq_search_filtered")
+        if not res.result_ptr:
+            raise MoannError("pq_search_filtered returned NULL result")
+        ids = np.empty((nq, k), dtype=np.int64)
+        dists = np.empty((nq, k), dtype=np.float32)
+        L = lib()
+        L.gpu_ivf_flat_get_neighbors(res.result_ptr, nq * k, ids.ctypes.data)
+        L.gpu_ivf_flat_get_distances(res.result_ptr, nq * k,
+                                     dists.ctypes.data)
+        L.gpu_ivf_flat_free_result(res.result_ptr)
+        return ids, dists
 
     def search(self, queries, k: int, nprobe: int):
         queries = _as_f32(queries)

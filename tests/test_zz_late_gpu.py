@@ -108,3 +108,64 @@ def test_gpu_from_usearch_file(tmp_path):
     np.testing.assert_allclose(np.sort(gdists, 1)[:, 0],
                                np.sort(ref_d, 1)[:, 0], rtol=2e-5, atol=2e-5)
     gpu.close()
+
+
+def test_pq_filtered_search():
+    """Membership-filtered IVF-PQ search: results equal the PQ oracle's
+    full ADC ranking post-filtered to the allowed set."""
+    from matrixone_amd import engine
+    from oracle import pq as opq
+    rng = np.random.Generator(np.random.PCG64(131))
+    n, d, nsub, nlist, nq, probe, k = 6000, 32, 8, 16, 8, 6, 10
+    centers = rng.standard_normal((60, d), dtype=np.float32)
+    comp = rng.integers(0, 60, n)
+    vecs = (centers[comp] +
+            0.5 * rng.standard_normal((n, d))).astype(np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    cb = opq.train_pq(vecs, cents, assign, nsub, iters=5, seed=7)
+    codes = opq.encode_pq(vecs, cents, assign, cb)
+    ids64 = rng.permutation(n).astype(np.int64) + 21
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    allowed = ids64[rng.random(n) < 0.3]
+    aset = set(int(x) for x in allowed)
+
+    oidx = opq.PqIndex(cents, cb, codes, assign, ids=ids64)
+    all_ids, all_d = oidx.search(queries, probe=probe, k=2500)
+    exp_ids = np.full((nq, k), -1, dtype=np.int64)
+    exp_d = np.full((nq, k), np.finfo(np.float32).max, dtype=np.float32)
+    for i in range(nq):
+        keep = [(all_d[i, j], all_ids[i, j]) for j in range(all_ids.shape[1])
+                if all_ids[i, j] >= 0 and int(all_ids[i, j]) in aset]
+        for j, (dd, ii) in enumerate(keep[:k]):
+            exp_ids[i, j] = ii
+            exp_d[i, j] = dd
+
+    ix = engine.IvfPqIndex(d, nsub, nlist, capacity=n)
+    ix.set_centroids(cents)
+    ix.set_codebooks(cb)
+    ix.add_codes(codes, ids=ids64)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    # slot bitset over GLOBAL slots (slot order = stable sort by assign)
+    order = np.argsort(assign, kind="stable")
+    slot_of_row = np.empty(n, dtype=np.int64)
+    slot_of_row[order] = np.arange(n)
+    rows_allowed = np.nonzero(np.isin(ids64, allowed))[0]
+    slots = slot_of_row[rows_allowed]
+    words = np.zeros((n + 31) // 32, dtype=np.uint32)
+    np.bitwise_or.at(words, slots // 32,
+                     (np.uint32(1) << (slots % 32).astype(np.uint32)))
+    gids, gdists = ix.search_filtered(queries, k, probe, words)
+    # ADC distances: decomposition-form f32 vs oracle direct-form — allow
+    # the documented 1e-4 band, ids gated by dist equality at ties
+    d0 = gdists.astype(np.float64)
+    d1 = exp_d.astype(np.float64)
+    np.testing.assert_allclose(d0, d1, rtol=1e-4, atol=1e-4)
+    mism = gids != exp_ids
+    if mism.any():
+        rel = np.abs(d0[mism] - d1[mism]) / np.maximum(np.abs(d1[mism]),
+                                                       1e-30)
+        assert (rel < 1e-4).all()
+    ix.close()
