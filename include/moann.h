@@ -354,6 +354,11 @@ gpu_ivf_flat_search_res_t moann_ivf_pq_search_filtered(
     ivf_flat_search_params_t search_params, const uint32_t* filter_bitset,
     uint64_t bitset_words, void* errmsg);
 
+/* PQ persistence: same directory-manifest contract as
+ * gpu_ivf_flat_save_dir/load_dir. */
+void moann_ivf_pq_save_dir(moann_ivf_pq_c h, const char* dir, void* errmsg);
+void moann_ivf_pq_load_dir(moann_ivf_pq_c h, const char* dir, void* errmsg);
+
 void moann_ivf_pq_perf(moann_ivf_pq_c h, moann_perf_t* out);
 void moann_ivf_pq_destroy(moann_ivf_pq_c h, void* errmsg);
 

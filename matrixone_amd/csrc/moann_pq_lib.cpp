@@ -65,6 +65,7 @@ struct PqIndex {
     std::vector<float> h_centroids;
     std::vector<uint8_t> h_codes; /* staging, row-major [count][nsub] */
     std::vector<float> h_codebooks;
+    std::vector<int64_t> h_id_by_slot; /* slot -> id (persistence) */
 
     uint8_t* d_codes = nullptr;       /* packed groups */
     float* d_cents_rows = nullptr;    /* [nlist][dpad] */
@@ -618,6 +619,7 @@ void moann_ivf_pq_build(moann_ivf_pq_c h, void* errmsg) {
                                   hipMemcpyHostToDevice));
         ix->h_codes.clear();
         ix->h_codes.shrink_to_fit();
+        ix->h_id_by_slot = std::move(id_by_slot);
         ix->built = true;
     } catch (const std::exception& e) {
         moann_set_errmsg(errmsg, "moann_ivf_pq_build", e.what());
@@ -669,6 +671,142 @@ gpu_ivf_flat_search_res_t moann_ivf_pq_search_filtered(
 
 void moann_ivf_pq_perf(moann_ivf_pq_c h, moann_perf_t* out) {
     *out = PX(h)->perf;
+}
+
+/* PQ persistence — same directory-manifest contract as the IVF index
+ * (gpu_ivf_flat_save_dir/load_dir; index_base.hpp:239-249 precedent).
+ * Blobs are SLOT-ordered; load replays them through the normal ingestion
+ * (add_codes + set_assignments + build), which reproduces the identical
+ * slot layout (stable sort by list). */
+void moann_ivf_pq_save_dir(moann_ivf_pq_c h, const char* dir, void* errmsg) {
+    try {
+        auto ix = PX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (!ix->built) throw std::runtime_error("index not built");
+        MOANN_HIP_CHECK(hipSetDevice(ix->device));
+        std::string d(dir);
+        auto wfile = [&](const std::string& name, const void* p, size_t n) {
+            FILE* f = fopen((d + "/" + name).c_str(), "wb");
+            if (!f) throw std::runtime_error("cannot write " + name);
+            if (n && fwrite(p, 1, n, f) != n) {
+                fclose(f);
+                throw std::runtime_error("short write " + name);
+            }
+            fclose(f);
+        };
+        {
+            char buf[256];
+            snprintf(buf, sizeof buf,
+                     "{\n  \"format\": \"moann-ivf-pq-v1\",\n"
+                     "  \"dimension\": %u,\n  \"nsub\": %u,\n"
+                     "  \"n_lists\": %u,\n  \"count\": %llu\n}\n",
+                     ix->dim, ix->nsub, ix->nlist,
+                     (unsigned long long)ix->count);
+            wfile("manifest.json", buf, strlen(buf));
+        }
+        wfile("centroids.f32", ix->h_centroids.data(),
+              ix->h_centroids.size() * 4);
+        wfile("codebooks.f32", ix->h_codebooks.data(),
+              ix->h_codebooks.size() * 4);
+        wfile("ids.i64", ix->h_id_by_slot.data(), ix->count * 8);
+        {
+            std::vector<int32_t> a(ix->count);
+            for (uint32_t l = 0; l < ix->nlist; ++l)
+                for (int64_t s2 = ix->list_slot_base[l];
+                     s2 < ix->list_slot_base[l] + ix->list_rows[l]; ++s2)
+                    a[s2] = (int32_t)l;
+            wfile("assign.i32", a.data(), ix->count * 4);
+        }
+        {
+            /* codes.u8: unpack the interleaved image to slot-major rows */
+            std::vector<int32_t> g2list;
+            std::vector<int64_t> g2slot;
+            std::vector<int32_t> g2valid;
+            for (uint32_t l = 0; l < ix->nlist; ++l) {
+                const int64_t lg = (ix->list_rows[l] + 63) / 64;
+                for (int64_t gi = 0; gi < lg; ++gi) {
+                    g2slot.push_back(ix->list_slot_base[l] + gi * 64);
+                    g2valid.push_back((int32_t)std::min<int64_t>(
+                        64, ix->list_rows[l] - gi * 64));
+                }
+            }
+            MoannDevBuf d_gsb, d_gv, d_tmp;
+            d_gsb.ensure(std::max<size_t>(8, g2slot.size() * 8));
+            d_gv.ensure(std::max<size_t>(4, g2valid.size() * 4));
+            MOANN_HIP_CHECK(hipMemcpy(d_gsb.ptr, g2slot.data(),
+                                      g2slot.size() * 8,
+                                      hipMemcpyHostToDevice));
+            MOANN_HIP_CHECK(hipMemcpy(d_gv.ptr, g2valid.data(),
+                                      g2valid.size() * 4,
+                                      hipMemcpyHostToDevice));
+            d_tmp.ensure(std::max<uint64_t>(1, ix->count) * ix->nsub);
+            launch_unpack_bytes(ix->d_codes, (int)ix->nsub, (int)ix->nsub,
+                                d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
+                                (int64_t)g2slot.size(), 0,
+                                (int64_t)ix->count, d_tmp.as<uint8_t>(),
+                                ix->stream);
+            MOANN_HIP_CHECK(hipStreamSynchronize(ix->stream));
+            std::vector<uint8_t> rows((size_t)ix->count * ix->nsub);
+            MOANN_HIP_CHECK(hipMemcpy(rows.data(), d_tmp.ptr, rows.size(),
+                                      hipMemcpyDeviceToHost));
+            wfile("codes.u8", rows.data(), rows.size());
+        }
+    } catch (const std::exception& e) {
+        moann_set_errmsg(errmsg, "moann_ivf_pq_save_dir", e.what());
+    }
+}
+
+void moann_ivf_pq_load_dir(moann_ivf_pq_c h, const char* dir, void* errmsg) {
+    try {
+        auto ix = PX(h);
+        if (ix->built) throw std::runtime_error("index already built");
+        std::string d(dir);
+        auto rfile = [&](const std::string& name, void* p, size_t n) {
+            FILE* f = fopen((d + "/" + name).c_str(), "rb");
+            if (!f) throw std::runtime_error("cannot read " + name);
+            const size_t got = fread(p, 1, n, f);
+            fclose(f);
+            if (got != n) throw std::runtime_error("short read " + name);
+        };
+        char buf[512] = {0};
+        rfile("manifest.json", buf, 0); /* existence check */
+        {
+            FILE* f = fopen((d + "/manifest.json").c_str(), "rb");
+            size_t got = fread(buf, 1, sizeof buf - 1, f);
+            fclose(f);
+            (void)got;
+        }
+        unsigned dim = 0, nsub = 0, nlist = 0;
+        unsigned long long count = 0;
+        if (!strstr(buf, "moann-ivf-pq-v1"))
+            throw std::runtime_error("unknown manifest format");
+        sscanf(strstr(buf, "\"dimension\":"), "\"dimension\": %u", &dim);
+        sscanf(strstr(buf, "\"nsub\":"), "\"nsub\": %u", &nsub);
+        sscanf(strstr(buf, "\"n_lists\":"), "\"n_lists\": %u", &nlist);
+        sscanf(strstr(buf, "\"count\":"), "\"count\": %llu", &count);
+        if (dim != ix->dim || nsub != ix->nsub || nlist != ix->nlist)
+            throw std::runtime_error(
+                "manifest does not match the created index");
+        if (count > ix->cap && ix->cap)
+            throw std::runtime_error("manifest count exceeds capacity");
+        std::vector<float> cents((size_t)nlist * dim);
+        rfile("centroids.f32", cents.data(), cents.size() * 4);
+        moann_ivf_pq_set_centroids(h, cents.data(), nlist, errmsg);
+        std::vector<float> cb((size_t)nsub * 256 * (dim / nsub));
+        rfile("codebooks.f32", cb.data(), cb.size() * 4);
+        moann_ivf_pq_set_codebooks(h, cb.data(), errmsg);
+        std::vector<int64_t> ids(count);
+        rfile("ids.i64", ids.data(), count * 8);
+        std::vector<uint8_t> codes((size_t)count * nsub);
+        rfile("codes.u8", codes.data(), codes.size());
+        moann_ivf_pq_add_codes(h, codes.data(), count, ids.data(), errmsg);
+        std::vector<int32_t> assign(count);
+        rfile("assign.i32", assign.data(), count * 4);
+        moann_ivf_pq_set_assignments(h, assign.data(), count, errmsg);
+        moann_ivf_pq_build(h, errmsg);
+    } catch (const std::exception& e) {
+        moann_set_errmsg(errmsg, "moann_ivf_pq_load_dir", e.what());
+    }
 }
 
 void moann_ivf_pq_destroy(moann_ivf_pq_c h, void* errmsg) {

@@ -153,6 +153,10 @@ def lib() -> ct.CDLL:
                                                ct.c_uint64, ct.c_void_p]
     L.moann_ivf_pq_build.argtypes = [ct.c_void_p, ct.c_void_p]
     L.moann_ivf_pq_search.restype = SearchRes
+    L.moann_ivf_pq_save_dir.argtypes = [ct.c_void_p, ct.c_char_p,
+                                        ct.c_void_p]
+    L.moann_ivf_pq_load_dir.argtypes = [ct.c_void_p, ct.c_char_p,
+                                        ct.c_void_p]
     L.moann_ivf_pq_search_filtered.restype = SearchRes
     L.moann_ivf_pq_search_filtered.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
@@ -709,6 +713,16 @@ class IvfPqIndex:
         L.gpu_ivf_flat_get_distances(res.result_ptr, nq * k, dists.ctypes.data)
         L.gpu_ivf_flat_free_result(res.result_ptr)
         return ids, dists
+
+    def save_dir(self, path: str):
+        err = _Err()
+        lib().moann_ivf_pq_save_dir(self._h, path.encode(), err.ref)
+        err.check("pq_save_dir")
+
+    def load_dir(self, path: str):
+        err = _Err()
+        lib().moann_ivf_pq_load_dir(self._h, path.encode(), err.ref)
+        err.check("pq_load_dir")
 
     def perf(self) -> dict:
         p = PerfT()

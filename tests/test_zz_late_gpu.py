@@ -169,3 +169,37 @@ def test_pq_filtered_search():
                                                        1e-30)
         assert (rel < 1e-4).all()
     ix.close()
+
+
+def test_pq_save_load_dir(tmp_path):
+    """PQ persistence round trip: identical results after save -> fresh
+    handle -> load."""
+    from matrixone_amd import engine
+    from oracle import pq as opq
+    rng = np.random.Generator(np.random.PCG64(77))
+    n, d, nsub, nlist, nq, probe, k = 4000, 32, 8, 16, 8, 6, 10
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    cb = opq.train_pq(vecs, cents, assign, nsub, iters=4, seed=3)
+    codes = opq.encode_pq(vecs, cents, assign, cb)
+    ids64 = rng.permutation(n).astype(np.int64) + 9
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+
+    ix = engine.IvfPqIndex(d, nsub, nlist, capacity=n)
+    ix.set_centroids(cents)
+    ix.set_codebooks(cb)
+    ix.add_codes(codes, ids=ids64)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    ref_ids, ref_d = ix.search(queries, k, probe)
+    ix.save_dir(str(tmp_path))
+    ix.close()
+
+    ix2 = engine.IvfPqIndex(d, nsub, nlist, capacity=n)
+    ix2.load_dir(str(tmp_path))
+    ids2, d2 = ix2.search(queries, k, probe)
+    np.testing.assert_array_equal(ids2, ref_ids)
+    np.testing.assert_allclose(d2, ref_d)
+    ix2.close()
