@@ -242,6 +242,28 @@ void moann_pairwise_distance(distance_type_t metric, const float* query,
                              const float* rows, uint64_t n, uint32_t dim,
                              int device, float* out_dists, void* errmsg);
 
+/* moann extension — membership-filtered search: slot_bitset has one bit
+ * per GLOBAL slot (LSB-first u32 words; nbits = total bits covered, must
+ * span len(index)); a CLEARED bit excludes the entry (cuVS bitset_filter
+ * polarity, cgo/cuvs/ivf_flat.hpp:908-924). */
+gpu_ivf_flat_search_res_t moann_ivf_flat_search_filtered(
+    gpu_ivf_flat_c index_c, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, const uint32_t* slot_bitset,
+    uint64_t nbits, void* errmsg);
+
+/* moann extension — id -> global slot lookup for building slot bitsets
+ * host-side; out_slots[i] = -1 for unknown ids. */
+void moann_ivf_flat_slots_of_ids(gpu_ivf_flat_c index_c, const int64_t* ids,
+                                 uint64_t n, int64_t* out_slots,
+                                 void* errmsg);
+
+/* ivf_flat_c.h:91 shape: fuse concurrent gpu_ivf_flat_search_async calls
+ * arriving within window_us into one device batch (0 disables;
+ * dynamic_batching.hpp:19-60 precedent). */
+void gpu_ivf_flat_set_batch_window(gpu_ivf_flat_c index_c, int64_t window_us,
+                                   void* errmsg);
+
 /* --- docfilter membership evaluation (moann extension) -------------------
  * Turns a serialized MatrixOne membership filter into a key bitset for the
  * filtered-search kernels. Kinds match cgo/usearchex.h:30-33 (the tag bytes
