@@ -148,14 +148,14 @@ __global__ __launch_bounds__(256) void scan_kernel(
                  * Counted-wait discipline = nothing crosses the wait.      */\
                 __builtin_amdgcn_sched_barrier(0);                          \
                 const float4 x0 = slot;                                     \
-                const int q4 = (idx);                                       \
+                const int q4i = (idx);                                      \
                 if (METRIC == KM_COS)                                       \
                     rn0 += x0.x * x0.x + x0.y * x0.y + x0.z * x0.z +        \
                            x0.w * x0.w;                                     \
                 _Pragma("unroll")                                           \
                 for (int t = 0; t < QT; ++t) {                              \
                     const float4 qv =                                       \
-                        ((const float4*)(ldsq + t * dpad))[q4];             \
+                        ((const float4*)(ldsq + t * dpad))[q4i];             \
                     if (METRIC == KM_L2SQ) {                                \
                         float e;                                            \
                         e = x0.x - qv.x; acc0[t] = fmaf(e, e, acc0[t]);     \
