@@ -284,6 +284,7 @@ void pack_lists(IvfIndex* ix, const float* d_rows, int64_t nrows,
                 const std::vector<int64_t>& loffsets, /* [nlists+1] slots */
                 float** d_packed_out, std::vector<int64_t>* gbase_out,
                 int64_t* ngroups_out) {
+    (void)nrows; /* implicit in loffsets */
     const int64_t nlists = (int64_t)loffsets.size() - 1;
     std::vector<int64_t> gbase(nlists + 1, 0);
     for (int64_t l = 0; l < nlists; ++l) {
