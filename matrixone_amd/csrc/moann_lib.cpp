@@ -798,6 +798,17 @@ double collect_result(MoannResult* res) {
     const float* ho_dists = (const float*)(ho_ids + n);
     memcpy(res->neighbors.data(), ho_ids, n * 8);
     memcpy(res->distances.data(), ho_dists, n * 4);
+    /* Reference C-ABI contract (cgo/cuvs/index_base.hpp:355-375
+     * transform_distance): InnerProduct distances cross the C ABI as the
+     * TRUE +a.b (the internal/MO-convention -a.b is flipped at readout;
+     * ±FLT_MAX pad/filter sentinels preserved). The host mirror negates
+     * back to MO's -a.b convention, matching the reference's division of
+     * labor between cgo and Go. */
+    if (ix->kmetric == KM_IP) {
+        float* d = res->distances.data();
+        for (size_t i = 0; i < n; ++i)
+            if (d[i] != FLT_MAX && d[i] != -FLT_MAX) d[i] = -d[i];
+    }
     float r_ms = 0, sc_ms = 0, sel_ms = 0;
     (void)hipEventElapsedTime(&r_ms, cx->ev[0], cx->ev[1]);
     (void)hipEventElapsedTime(&sc_ms, cx->ev[2], cx->ev[3]);
@@ -1658,10 +1669,16 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
             int metric = 0;
             if (!strstr(buf, "moann-ivf-flat-v1"))
                 throw std::runtime_error("unknown manifest format");
-            sscanf(strstr(buf, "\"dimension\":"), "\"dimension\": %u", &dim);
-            sscanf(strstr(buf, "\"n_lists\":"), "\"n_lists\": %u", &nlist);
-            sscanf(strstr(buf, "\"metric\":"), "\"metric\": %d", &metric);
-            sscanf(strstr(buf, "\"count\":"), "\"count\": %llu", &count);
+            const char* fdim = strstr(buf, "\"dimension\":");
+            const char* fnl = strstr(buf, "\"n_lists\":");
+            const char* fme = strstr(buf, "\"metric\":");
+            const char* fct = strstr(buf, "\"count\":");
+            if (!fdim || !fnl || !fme || !fct)
+                throw std::runtime_error("malformed manifest.json");
+            sscanf(fdim, "\"dimension\": %u", &dim);
+            sscanf(fnl, "\"n_lists\": %u", &nlist);
+            sscanf(fme, "\"metric\": %d", &metric);
+            sscanf(fct, "\"count\": %llu", &count);
             if (dim != ix->dim || nlist != ix->nlist ||
                 metric != (int)ix->metric)
                 throw std::runtime_error(

@@ -354,6 +354,14 @@ class IvfFlatIndex:
         L.gpu_ivf_flat_get_neighbors(res.result_ptr, nq * k, ids.ctypes.data)
         L.gpu_ivf_flat_get_distances(res.result_ptr, nq * k, dists.ctypes.data)
         L.gpu_ivf_flat_free_result(res.result_ptr)
+        if self.metric == "ip":
+            # The C ABI emits the reference contract: true +a.b for IP
+            # (cgo/cuvs/index_base.hpp:355-375 transform_distance). This
+            # mirror plays the Go side and restores MO's -a.b convention
+            # (distance_func.go:174-207); ±FLT_MAX sentinels preserved.
+            sent = np.float32(np.finfo(np.float32).max)
+            m = (dists != sent) & (dists != -sent)
+            dists[m] = -dists[m]
         return ids, dists
 
     def search(self, queries, k: int, nprobe: int):

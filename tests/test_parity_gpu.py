@@ -730,3 +730,47 @@ def test_pred_filtered_search():
                                  ix.filter_bitset(np.array(sorted(sel))))
     np.testing.assert_array_equal(g2, e2)
     ix.close()
+
+
+def test_ip_sign_at_c_abi_matches_reference_contract():
+    """The raw C ABI must emit the TRUE inner product +a.b for IP results
+    (the reference flips cuvs' negated IP before it crosses the C ABI —
+    cgo/cuvs/index_base.hpp:355-375 transform_distance), while the host
+    mirror restores MO's -a.b convention (distance_func.go:174-207). So
+    raw-ctypes distances == -(engine-level distances) except sentinels."""
+    import ctypes as ct
+    from matrixone_amd import engine
+    rng = np.random.Generator(np.random.PCG64(4242))
+    n, d, nlist, nq, probe, k = 5000, 96, 32, 8, 4, 12
+    vecs = rng.standard_normal((n, d), dtype=np.float32)
+    cents = vecs[rng.choice(n, nlist, replace=False)].copy()
+    assign = ((vecs[:, None, :].astype(np.float64) -
+               cents[None].astype(np.float64)) ** 2).sum(-1).argmin(1)
+    queries = rng.standard_normal((nq, d), dtype=np.float32)
+    ix = engine.IvfFlatIndex(d, nlist, metric="ip", capacity=n)
+    ix.add(vecs)
+    ix.set_centroids(cents)
+    ix.set_assignments(assign.astype(np.int32))
+    ix.build()
+    eng_ids, eng_dists = ix.search(queries, k, probe)
+
+    L = engine.lib()
+    err = engine._Err()
+    res = L.gpu_ivf_flat_search(ix._h, queries.ctypes.data, nq, d, k,
+                                engine.SearchParams(probe), err.ref)
+    err.check("raw search")
+    raw_ids = np.empty((nq, k), dtype=np.int64)
+    raw_dists = np.empty((nq, k), dtype=np.float32)
+    L.gpu_ivf_flat_get_neighbors(res.result_ptr, nq * k, raw_ids.ctypes.data)
+    L.gpu_ivf_flat_get_distances(res.result_ptr, nq * k, raw_dists.ctypes.data)
+    L.gpu_ivf_flat_free_result(res.result_ptr)
+    np.testing.assert_array_equal(raw_ids, eng_ids)
+    sent = np.float32(np.finfo(np.float32).max)
+    live = (raw_dists != sent) & (raw_dists != -sent)
+    np.testing.assert_array_equal(live, (eng_dists != sent) & (eng_dists != -sent))
+    np.testing.assert_array_equal(raw_dists[live], -eng_dists[live])
+    # +a.b really is the inner product of the returned neighbor
+    q0n = int(raw_ids[0, 0])  # default ids are 0..n-1 (row order)
+    want = float(np.dot(queries[0].astype(np.float64), vecs[q0n].astype(np.float64)))
+    assert abs(raw_dists[0, 0] - want) <= 1e-4 * max(1.0, abs(want))
+    ix.close()
