@@ -1459,6 +1459,13 @@ void moann_brute_force_search(const float* dataset, uint64_t count,
                                    out_neighbors);
         gpu_ivf_flat_get_distances(r.result_ptr, num_queries * limit,
                                    out_distances);
+        /* moann_* entries speak MO conventions (IP = -a.b,
+         * distance_func.go:174-207); the cuVS-named readout above emits the
+         * reference C-ABI contract (+a.b) — negate back, skip sentinels. */
+        if (metric == DistanceType_InnerProduct)
+            for (uint64_t i = 0; i < num_queries * (uint64_t)limit; ++i)
+                if (out_distances[i] != FLT_MAX && out_distances[i] != -FLT_MAX)
+                    out_distances[i] = -out_distances[i];
         gpu_ivf_flat_free_result(r.result_ptr);
         gpu_ivf_flat_destroy(h, nullptr);
     } catch (const std::exception& e) {
