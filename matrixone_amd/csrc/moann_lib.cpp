@@ -384,8 +384,18 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
 
     /* 2. centroid rank: scan the centroid matrix (one "list" of nlist rows),
      * then top-probe per query (ivfflat/search.go:265-308 rankCentroids). */
+    /* The job tile width MUST match the kernel that scans the lists: the
+     * byte kernels (int8/uint8 storage AND the refine byte first pass) and
+     * the half kernels are QT=8; only the f32 scan has the QT=16 asm
+     * specialization. Round-1 latent bug: refine jobs were built at the f32
+     * QT=16 while scan_i8_dot_kernel processes t<8 — tile lanes 8..15 were
+     * never scanned and their candidates kept stale buffer contents (masked
+     * in tests by same-query buffer aliasing; recall 0.9414 at 10M). */
+    const bool refine_scan = ix->qtype == Quantization_F32 &&
+                             ix->refine_depth > 0 &&
+                             ix->d_packed_rq != nullptr;
     const int qt = pick_qt(ix->kmetric, ix->dpad,
-                           ix->qtype != Quantization_F32);
+                           ix->qtype != Quantization_F32 || refine_scan);
     if (use_mfma_rank) {
         cx->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         HIP_CHECK(hipEventRecord(cx->ev[0], s));
@@ -582,8 +592,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const bool uns = ix->qtype == Quantization_UINT8;
     const bool half = ix->qtype == Quantization_F16 ||
                       ix->qtype == Quantization_BF16;
-    const bool use_refine =
-        !narrow && ix->refine_depth > 0 && ix->d_packed_rq != nullptr;
+    const bool use_refine = refine_scan;
     const uint32_t* d_filter = nullptr;
     if (filter_words) {
         const size_t fw = (size_t)((ix->count + 31) / 32);

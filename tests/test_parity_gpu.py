@@ -565,11 +565,16 @@ def test_refine_exactness():
     ix.set_centroids(cents)
     ix.set_assignments(assign.astype(np.int32))
     ix.build()
-    exact_ids, exact_d = ix.search(queries, k, probe)
 
-    # R = 4096 > max candidates per query (~20000*4/64 ~ 1250): identical
+    # R = 4096 > max candidates per query (~20000*4/64 ~ 1250): identical.
+    # The refine search runs FIRST on the fresh index: a prior exact search
+    # over the same queries leaves correct f32 distances in the reused
+    # candidate buffer, which masked the round-1 QT-16/QT-8 job-geometry
+    # bug (unwritten tile lanes read stale-but-correct values).
     ix.enable_refine(4096)
     rid, rd = ix.search(queries, k, probe)
+    ix.enable_refine(0)
+    exact_ids, exact_d = ix.search(queries, k, probe)
     np.testing.assert_array_equal(rid, exact_ids)
     np.testing.assert_allclose(rd, exact_d, rtol=2e-5, atol=2e-5)
 
