@@ -180,11 +180,12 @@ def main():
     ap.add_argument("--recall-queries", type=int, default=256)
     ap.add_argument("--cpu-baseline-queries", type=int, default=48)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--refine", type=int, default=0,
+    ap.add_argument("--refine", type=int, default=128,
                     help="two-stage scan: byte first pass + exact f32 "
-                         "re-rank of the top-R (0 = exact one-pass scan; "
-                         "the default stays the GPU-validated exact path "
-                         "until a refine recall curve is measured at 10M). "
+                         "re-rank of the top-R (0 = exact one-pass scan). "
+                         "GPU-validated at 10M post the QT-8 job-geometry "
+                         "fix: R=64/128/256 all recall 1.0000 "
+                         "(profiles/r2_refine_sweep). "
                          "ADAPTIVE: if the measured recall@k gate (0.95) "
                          "fails with refine on, the bench falls back to "
                          "the exact one-pass scan and re-measures — the "

@@ -45,10 +45,14 @@ typedef enum {
     DistanceType_L2SqrtExpanded = 1, /* sqrt(sum((x-y)^2))                    */
     DistanceType_CosineExpanded = 2, /* 1 - x.y/(|x||y|)                      */
     DistanceType_L1 = 3,             /* sum|x-y|                              */
-    DistanceType_InnerProduct = 6,   /* returned in MO convention: -x.y
-                                      * (cgo/cuvs negates cuVS IP before Go —
-                                      * metric/types.go:219-243; we emit the MO
-                                      * convention directly)                  */
+    DistanceType_InnerProduct = 6,   /* distances cross the gpu_ivf_flat_*
+                                      * C ABI as the TRUE +x.y, matching the
+                                      * reference's transform_distance flip
+                                      * (cgo/cuvs/index_base.hpp:355-375);
+                                      * ordering is still by closeness
+                                      * (ascending -x.y). moann_* entries
+                                      * documented MO-convention return -x.y
+                                      * (distance_func.go:174-207)           */
     DistanceType_Unknown = 255
 } distance_type_t;
 
@@ -86,9 +90,39 @@ typedef struct {
 
 /* --- lifecycle (ivf_flat_c.h:39-106) ------------------------------------- */
 
+/* ivf_flat_c.h:41-45 gpu_ivf_flat_new: dataset constructor — equivalent to
+ * new_empty(count_vectors, ...) followed by one add_chunk of the whole
+ * dataset (storage-typed rows, exactly like the reference's build-from-
+ * dataset ctor). Call gpu_ivf_flat_build afterwards. */
+gpu_ivf_flat_c gpu_ivf_flat_new(const void* dataset_data,
+                                uint64_t count_vectors, uint32_t dimension,
+                                distance_type_t metric,
+                                ivf_flat_build_params_t build_params,
+                                const int* devices, int device_count,
+                                uint32_t nthread,
+                                distribution_mode_t dist_mode,
+                                quantization_t btype, quantization_t qtype,
+                                const int64_t* ids, void* errmsg);
+
+/* ivf_flat_c.h:48-51 gpu_ivf_flat_load_file: constructor that restores an
+ * index saved by gpu_ivf_flat_save. The container format is this library's
+ * own (a single-file wrap of the save_dir blobs) — the reference's is
+ * cuVS-serialize, equally library-internal; both sides write and read
+ * their own format, results are what must match. */
+gpu_ivf_flat_c gpu_ivf_flat_load_file(const char* filename,
+                                      uint32_t dimension,
+                                      distance_type_t metric,
+                                      ivf_flat_build_params_t build_params,
+                                      const int* devices, int device_count,
+                                      uint32_t nthread,
+                                      distribution_mode_t dist_mode,
+                                      quantization_t btype,
+                                      quantization_t qtype, void* errmsg);
+
 /* ivf_flat_c.h:62-67 gpu_ivf_flat_new_empty: pre-allocates for total_count
- * vectors of `dimension`.  btype/qtype: round 1 wires F32/F32 only; other
- * combos set errmsg and return NULL. */
+ * vectors of `dimension`.  Wired btype/qtype combos mirror the reference
+ * (ivf_flat_c.h:33-35): F32 base {F32,F16,BF16,INT8,UINT8}; F16 base
+ * {F16,INT8,UINT8}. Other combos set errmsg and return NULL. */
 gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
                                       distance_type_t metric,
                                       ivf_flat_build_params_t build_params,
@@ -98,11 +132,59 @@ gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
                                       quantization_t btype, quantization_t qtype,
                                       const int64_t* ids, void* errmsg);
 
-/* ivf_flat_c.h:69 gpu_ivf_flat_add_chunk: append chunk_count vectors (host
- * pointer, f32 row-major) with their ids (NULL = sequential). */
+/* ivf_flat_c.h:69 gpu_ivf_flat_add_chunk: append chunk_count vectors in the
+ * index's STORAGE type (f32 rows / f16·bf16 u16 rows / int8·uint8 bytes,
+ * row-major host pointer) with their ids (NULL = sequential). Narrow input
+ * is staged losslessly: the affine map round-trips bytes exactly
+ * (|b| <= 255 so dequant->requant re-lands on b), halves are exact casts. */
 void gpu_ivf_flat_add_chunk(gpu_ivf_flat_c index_c, const void* chunk_data,
                             uint64_t chunk_count, const int64_t* ids,
                             void* errmsg);
+
+/* ivf_flat_c.h:80 gpu_ivf_flat_add_chunk_float: f32 rows, quantized to the
+ * storage type at build (requires the quantizer set/trained for int8/uint8
+ * storage). */
+void gpu_ivf_flat_add_chunk_float(gpu_ivf_flat_c index_c,
+                                  const float* chunk_data,
+                                  uint64_t chunk_count, const int64_t* ids,
+                                  void* errmsg);
+
+/* ivf_flat_c.h:83 gpu_ivf_flat_add_chunk_quantize: rows in the BASE type B
+ * (f32 base: same as add_chunk_float; F16 base: u16 half rows). */
+void gpu_ivf_flat_add_chunk_quantize(gpu_ivf_flat_c index_c,
+                                     const void* base_data,
+                                     uint64_t chunk_count, const int64_t* ids,
+                                     void* errmsg);
+
+/* ivf_flat_c.h:71-78 extend: append vectors to an ALREADY-BUILT index.
+ * Rows are assigned to their nearest centroid under the index metric and
+ * become members of that list (the reference's cuvs::ivf_flat::extend
+ * semantics; search results are as if they had been in the build).
+ * new_ids NULL = sequential from the current length (ivf_flat_c.h:72).
+ * Internally extends land in a per-list tail sub-index rebuilt per call —
+ * O(tail) per extend, searched alongside the main lists and merged. */
+void gpu_ivf_flat_extend(gpu_ivf_flat_c index_c, const void* new_data,
+                         uint64_t n_rows, const int64_t* new_ids,
+                         void* errmsg);
+void gpu_ivf_flat_extend_float(gpu_ivf_flat_c index_c, const float* new_data,
+                               uint64_t n_rows, const int64_t* new_ids,
+                               void* errmsg);
+
+/* ivf_flat_c.h:86 train_quantizer: computes the scalar quantizer's
+ * [min,max] from a f32 training sample as the central-99% quantile range
+ * (the cuVS scalar-quantizer default the reference trains with,
+ * cgo/cuvs/quantize.hpp:60-66); the affine map itself keeps quantizer.go's
+ * Int8Params/Uint8Params formulas (the CPU-path parity anchor). No-op
+ * requirement: int8/uint8 storage only, before build. */
+void gpu_ivf_flat_train_quantizer(gpu_ivf_flat_c index_c,
+                                  const float* train_data,
+                                  uint64_t n_samples, void* errmsg);
+
+/* ivf_flat_c.h:102 delete_id: soft delete via the alive bitset
+ * (index_base.hpp:1142-1167: unknown id = silent no-op; repeated delete
+ * counted once). Honored by every search variant including tails; persisted
+ * by save/save_dir as bitset.bin (omitted when no deletions). */
+void gpu_ivf_flat_delete_id(gpu_ivf_flat_c index_c, int64_t id, void* errmsg);
 
 /* MatrixOne ivfflat ingestion (moann extension): the centroid hidden table
  * (pkg/vectorindex/ivfflat/search.go:71-161 LoadCentroids — nlist rows of
@@ -135,6 +217,17 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c index_c, const char* dir,
 void gpu_ivf_flat_load_dir(gpu_ivf_flat_c index_c, const char* dir,
                            distribution_mode_t target_mode, void* errmsg);
 
+/* ivf_flat_c.h:95 single-file save; restored by the gpu_ivf_flat_load_file
+ * constructor. Same blobs as save_dir in one container file. */
+void gpu_ivf_flat_save(gpu_ivf_flat_c index_c, const char* filename,
+                       void* errmsg);
+
+/* ivf_flat_c.h:159 info: heap-allocated NUL-terminated JSON (caller frees
+ * with free()). Field set mirrors index_base.hpp:2098-2117: element_size,
+ * dimension, metric, status, capacity, current_length, dist_mode, has_ids,
+ * devices, plus n_lists / qtype / deleted_count. */
+char* gpu_ivf_flat_info(gpu_ivf_flat_c index_c, void* errmsg);
+
 /* ivf_flat_c.h:53 */
 void gpu_ivf_flat_destroy(gpu_ivf_flat_c index_c, void* errmsg);
 
@@ -159,6 +252,21 @@ gpu_ivf_flat_search_res_t gpu_ivf_flat_search(gpu_ivf_flat_c index_c,
                                               uint32_t limit,
                                               ivf_flat_search_params_t search_params,
                                               void* errmsg);
+
+/* ivf_flat_c.h:119-122 search_quantize: queries in the BASE type B (f32
+ * base: identical to gpu_ivf_flat_search, which already narrows the query
+ * with the entry map for quantized storage — the search_quantize contract;
+ * F16 base: u16 half rows, cast to f32 then narrowed). */
+gpu_ivf_flat_search_res_t gpu_ivf_flat_search_quantize(
+    gpu_ivf_flat_c index_c, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, void* errmsg);
+
+/* ivf_flat_c.h:130-133 */
+uint64_t gpu_ivf_flat_search_quantize_async(
+    gpu_ivf_flat_c index_c, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, void* errmsg);
 
 /* moann extension — two-stage scan: byte-quantized first pass over an
  * INTERNAL clip-quantized image (0.1%/99.9% sampled quantiles; independent
@@ -263,6 +371,48 @@ void moann_ivf_flat_slots_of_ids(gpu_ivf_flat_c index_c, const int64_t* ids,
  * dynamic_batching.hpp:19-60 precedent). */
 void gpu_ivf_flat_set_batch_window(gpu_ivf_flat_c index_c, int64_t window_us,
                                    void* errmsg);
+
+/* ivf_flat_c.h:92: conservative dispatch waits the FULL window before
+ * dispatching a fused batch (default dispatches as soon as the window
+ * closes on the first request — identical here; the knob is accepted and
+ * recorded for contract completeness). */
+void gpu_ivf_flat_set_dynb_conservative_dispatch(gpu_ivf_flat_c index_c,
+                                                 bool enable, void* errmsg);
+
+/* --- pre-filter INCLUDE columns (ivf_flat_c.h:165-199) -------------------
+ * Register column metadata + column chunks on the index, then search with a
+ * JSON predicate evaluated index-side (the eval_filter_bitmap_cpu shape,
+ * cgo/cuvs/filter.hpp:855-960 — same evaluator as moann_preds_eval).
+ * col_meta_json: [{"name":"price","type":2},...] with type 0=i32, 1=i64,
+ * 2=f32, 3=f64, 4=u64 (VARCHAR hash) — cagra_c.h:166-167. Chunks are raw
+ * rows in ADD order; null_bitmap u32 LSB-first (1 = NULL), NULL = dense.
+ * Both must precede build. */
+void gpu_ivf_flat_set_filter_columns(gpu_ivf_flat_c index_c,
+                                     const char* col_meta_json,
+                                     uint64_t total_count, void* errmsg);
+void gpu_ivf_flat_add_filter_chunk(gpu_ivf_flat_c index_c, uint32_t col_idx,
+                                   const void* data,
+                                   const uint32_t* null_bitmap,
+                                   uint64_t nrows, void* errmsg);
+
+/* ivf_flat_c.h:179-199: filtered search over the registered columns.
+ * preds_json NULL or "" = unfiltered. The quantize variant takes base-typed
+ * queries; the async variant returns a job id for gpu_ivf_flat_search_wait. */
+gpu_ivf_flat_search_res_t gpu_ivf_flat_search_with_filter(
+    gpu_ivf_flat_c index_c, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, const char* preds_json,
+    void* errmsg);
+gpu_ivf_flat_search_res_t gpu_ivf_flat_search_quantize_with_filter(
+    gpu_ivf_flat_c index_c, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, const char* preds_json,
+    void* errmsg);
+uint64_t gpu_ivf_flat_search_quantize_with_filter_async(
+    gpu_ivf_flat_c index_c, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit,
+    ivf_flat_search_params_t search_params, const char* preds_json,
+    void* errmsg);
 
 /* --- docfilter membership evaluation (moann extension) -------------------
  * Turns a serialized MatrixOne membership filter into a key bitset for the
@@ -445,6 +595,46 @@ void moann_hnsw_search_filtered(moann_hnsw_c h, const float* queries,
 void moann_hnsw_perf(moann_hnsw_c h, moann_perf_t* out);
 void moann_hnsw_perf_reset(moann_hnsw_c h);
 void moann_hnsw_destroy(moann_hnsw_c h, void* errmsg);
+
+/* --- docfilter handles + the usearchex C entry ---------------------------
+ * The reference's Go bridge holds a C handle per docfilter
+ * (bloom/croaring/cbitmap/sorted64) and passes (handle, kind) into
+ * usearchex_filtered_search_with_membership per search
+ * (cgo/usearchex.h:38-42, pkg/vectorindex/usearchex/search.go:37-109).
+ * Mirror: parse the serialized payload once into a moann handle, then pass
+ * it to the same-named entry below. */
+typedef void* moann_docfilter_c;
+
+/* kind per MOANN_FILTER_* / cgo/usearchex.h:30-33 tag bytes; payload
+ * formats as documented at moann_docfilter_eval. Returns NULL + errmsg on
+ * malformed payloads (fail closed). */
+moann_docfilter_c moann_docfilter_parse(int kind, const void* payload,
+                                        uint64_t payload_len, void* errmsg);
+/* membership test of nkeys int64 keys against a parsed handle (bit i set
+ * iff keys[i] member; out_words = (nkeys+31)/32 u32 LSB-first). */
+int moann_docfilter_test(moann_docfilter_c f, const int64_t* keys,
+                         uint64_t nkeys, uint32_t* out_words, void* errmsg);
+void moann_docfilter_free(moann_docfilter_c f);
+
+/* cgo/usearchex.h:38-42 usearchex_filtered_search_with_membership, on a
+ * moann_hnsw_c instead of a usearch_index_t (same position, both opaque
+ * pointers; the Go bridge swaps the handle it passes, nothing else).
+ * query_kind must be usearch_scalar_f32_k (= 1, tarball c/usearch.h:57) —
+ * MO always searches f32 queries (usearchex/search.go:68). count = k.
+ * filter = a moann_docfilter_c (NULL passes all keys, usearchex.c:58);
+ * filter_kind as above. keys/distances are caller buffers of `count`;
+ * returns the number found; *error is set to a static message on failure
+ * (mo_error_t convention, usearchex.h:23). Distances are usearch
+ * conventions like moann_hnsw_search. ef = max(moann_hnsw_set_ef value,
+ * count). */
+size_t usearchex_filtered_search_with_membership(
+    moann_hnsw_c index, const void* query_vector, int query_kind,
+    size_t count, void* filter, int filter_kind, int64_t* keys,
+    float* distances, const char** error);
+
+/* default ef for the usearchex-shaped entry (MO passes efSearch through
+ * the usearch index config; our handle carries it here). Default 64. */
+void moann_hnsw_set_ef(moann_hnsw_c h, uint32_t ef, void* errmsg);
 
 /* --- misc ----------------------------------------------------------------- */
 

@@ -161,6 +161,7 @@ struct IvfIndex {
     distance_type_t metric = DistanceType_L2Expanded;
     int kmetric = KM_L2SQ;
     uint64_t cap = 0, count = 0;
+    quantization_t btype = Quantization_F32; /* base/query element type    */
     quantization_t qtype = Quantization_F32;
     double quant_mul = 1.0, quant_add = 0.0; /* identity until trained */
     double quant_min = 0.0, quant_max = 0.0;
@@ -229,6 +230,34 @@ struct IvfIndex {
     std::thread batcher;
     bool batcher_stop = false;
     bool batcher_started = false;
+    bool dynb_conservative = false; /* ivf_flat_c.h:92 knob (recorded) */
+
+    /* post-build extends (gpu_ivf_flat_extend/_float): a per-list TAIL
+     * sub-index over the extended rows, same centroids/metric/quantizer,
+     * rebuilt from host staging on each extend (O(tail) per call) and
+     * searched alongside the main lists with a top-k merge at collect —
+     * result-identical to the reference's in-list insertion
+     * (cuvs::ivf_flat::extend): a row participates exactly when its
+     * assigned list is probed. */
+    std::unique_ptr<IvfIndex> tail;
+    std::vector<float> tail_staging;   /* f32 rows, extend order */
+    std::vector<int64_t> tail_ids;
+    std::vector<int32_t> tail_assign;
+
+    /* soft-delete bitset (index_base.hpp:183-209): bit per SLOT, 1 = alive,
+     * lazily allocated on first delete. Applied to every search by ANDing
+     * into the membership filter. */
+    std::vector<uint32_t> alive;
+    uint64_t deleted_count = 0;
+
+    /* pre-filter INCLUDE columns (ivf_flat_c.h:165-177): raw column bytes
+     * in ADD order + null bitmaps, registered before build */
+    std::string filter_cols_json;
+    std::vector<int> filter_col_types;
+    std::vector<std::vector<uint8_t>> filter_col_data;
+    std::vector<std::vector<uint32_t>> filter_col_nulls;
+    std::vector<uint64_t> filter_col_rows;
+    std::vector<bool> filter_col_has_nulls;
 
     ~IvfIndex() {
         if (batcher_started) {
