@@ -394,3 +394,83 @@ extern "C" int moann_docfilter_eval(int kind, const void* payload,
         return -1;
     }
 }
+
+/* ---- docfilter handles (the Go bridge's per-filter C handle shape,
+ * cgo/usearchex.h:38-42 / pkg/vectorindex/usearchex/search.go:37-109) ----- */
+
+namespace {
+struct MoannDocFilter {
+    int kind;
+    std::vector<uint8_t> payload;
+};
+}  // namespace
+
+extern "C" moann_docfilter_c moann_docfilter_parse(int kind,
+                                                   const void* payload,
+                                                   uint64_t payload_len,
+                                                   void* errmsg) {
+    try {
+        if (!payload) {
+            set_err(errmsg, "moann_docfilter_parse", "null payload");
+            return nullptr;
+        }
+        /* validate by a trial evaluation over zero keys' parse step */
+        const uint8_t* p = (const uint8_t*)payload;
+        switch (kind) {
+        case MOANN_FILTER_BLOOM:
+            if (!bloom_view(p, payload_len).ok) {
+                set_err(errmsg, "moann_docfilter_parse", "bad bloom payload");
+                return nullptr;
+            }
+            break;
+        case MOANN_FILTER_CROARING:
+            if (!parse_r64(p, payload_len).ok) {
+                set_err(errmsg, "moann_docfilter_parse",
+                        "bad roaring64 payload");
+                return nullptr;
+            }
+            break;
+        case MOANN_FILTER_CBITMAP:
+            if (payload_len < 16) {
+                set_err(errmsg, "moann_docfilter_parse",
+                        "bad cbitmap payload");
+                return nullptr;
+            }
+            break;
+        case MOANN_FILTER_SORTED64:
+            if (payload_len < 8) {
+                set_err(errmsg, "moann_docfilter_parse",
+                        "bad sorted64 payload");
+                return nullptr;
+            }
+            break;
+        default:
+            set_err(errmsg, "moann_docfilter_parse", "unknown filter kind");
+            return nullptr;
+        }
+        auto f = new MoannDocFilter;
+        f->kind = kind;
+        f->payload.assign(p, p + payload_len);
+        return (moann_docfilter_c)f;
+    } catch (const std::exception& e) {
+        set_err(errmsg, "moann_docfilter_parse", e.what());
+        return nullptr;
+    }
+}
+
+extern "C" int moann_docfilter_test(moann_docfilter_c fh, const int64_t* keys,
+                                    uint64_t nkeys, uint32_t* out_words,
+                                    void* errmsg) {
+    if (!fh) {
+        set_err(errmsg, "moann_docfilter_test", "null filter");
+        return -1;
+    }
+    auto f = (MoannDocFilter*)fh;
+    return moann_docfilter_eval(f->kind, f->payload.data(),
+                                f->payload.size(), keys, nkeys, out_words,
+                                errmsg);
+}
+
+extern "C" void moann_docfilter_free(moann_docfilter_c fh) {
+    delete (MoannDocFilter*)fh;
+}

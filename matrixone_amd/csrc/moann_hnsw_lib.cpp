@@ -6,6 +6,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cfloat>
+#include <algorithm>
 #include <cstring>
 #include <memory>
 #include <mutex>
@@ -87,6 +88,8 @@ struct HnswIndex {
     int32_t* d_up_offs = nullptr;
     uint32_t* d_up_nbrs = nullptr;
     int64_t* d_keys = nullptr;
+    std::vector<int64_t> h_keys; /* slot -> key, for docfilter bitsets */
+    uint32_t default_ef = 64;    /* moann_hnsw_set_ef (usearchex entry) */
 
     HnswDevBuf w_queries, w_qnorms, w_visited, w_out_ids, w_out_dists,
         w_evals, w_filter;
@@ -253,6 +256,7 @@ void moann_hnsw_set_keys(moann_hnsw_c h, const int64_t* keys, void* errmsg) {
         HIP_CHECK(hipMalloc(&ix->d_keys, (size_t)ix->n * 8));
         HIP_CHECK(hipMemcpy(ix->d_keys, keys, (size_t)ix->n * 8,
                             hipMemcpyHostToDevice));
+        ix->h_keys.assign(keys, keys + ix->n);
     } catch (const std::exception& e) {
         hn_set_errmsg(errmsg, "moann_hnsw_set_keys", e.what());
     }
@@ -372,6 +376,76 @@ void moann_hnsw_search_filtered(moann_hnsw_c h, const float* queries,
                         slot_bitset);
     } catch (const std::exception& e) {
         hn_set_errmsg(errmsg, "moann_hnsw_search_filtered", e.what());
+    }
+}
+
+void moann_hnsw_set_ef(moann_hnsw_c h, uint32_t ef, void* errmsg) {
+    try {
+        if (ef == 0) throw std::runtime_error("ef must be positive");
+        HX(h)->default_ef = ef;
+    } catch (const std::exception& e) {
+        hn_set_errmsg(errmsg, "moann_hnsw_set_ef", e.what());
+    }
+}
+
+/* cgo/usearchex.h:38-42 / cgo/usearchex.c:51-70 — the filtered-search
+ * bridge the Go side binds, on a moann_hnsw_c. The membership handle is a
+ * moann_docfilter_c (moann_docfilter_parse of the serialized payload); it
+ * is evaluated over the slot->key map into a slot bitset (the GPU-side
+ * eval_filter_bitmap_cpu shape, cgo/cuvs/filter.hpp:855-960) and gates the
+ * REPORTABLE set exactly like usearch's predicate. Returns the number of
+ * results written; on failure writes a message into *error (mo_error_t
+ * convention: a NUL-terminated string, usearchex.h:23; storage is
+ * thread-local and stays valid until the next failing call on the
+ * thread). */
+size_t usearchex_filtered_search_with_membership(
+    moann_hnsw_c index, const void* query_vector, int query_kind,
+    size_t count, void* filter, int filter_kind, int64_t* keys,
+    float* distances, const char** error) {
+    static thread_local std::string err_store;
+    try {
+        auto ix = HX(index);
+        if (query_kind != 1 /* usearch_scalar_f32_k, c/usearch.h:57 */)
+            throw std::runtime_error(
+                "query kind must be usearch_scalar_f32_k (MO searches f32, "
+                "usearchex/search.go:68)");
+        if (count == 0) return 0;
+        const uint32_t ef =
+            std::max<uint32_t>(ix->default_ef, (uint32_t)count);
+        std::vector<uint32_t> bits;
+        if (filter) {
+            bits.assign((size_t)((ix->n + 31) / 32), 0);
+            if (ix->h_keys.empty())
+                throw std::runtime_error("keys not set on the index");
+            char* inner = nullptr;
+            if (moann_docfilter_test((moann_docfilter_c)filter,
+                                     ix->h_keys.data(), ix->n, bits.data(),
+                                     &inner) != 0) {
+                std::string m = inner ? inner : "filter evaluation failed";
+                if (inner) free(inner);
+                (void)filter_kind;
+                throw std::runtime_error(m);
+            }
+        }
+        std::vector<int64_t> okeys(count);
+        std::vector<float> odists(count);
+        run_hnsw_search(ix, (const float*)query_vector, false, 1, ef,
+                        (uint32_t)count, okeys.data(), odists.data(),
+                        filter ? bits.data() : nullptr);
+        size_t found = 0;
+        for (size_t i = 0; i < count; ++i) {
+            if (okeys[i] == -1) break;
+            keys[found] = okeys[i];
+            distances[found] = odists[i];
+            ++found;
+        }
+        return found;
+    } catch (const std::exception& e) {
+        if (error) {
+            err_store = e.what();
+            *error = err_store.c_str();
+        }
+        return 0;
     }
 }
 

@@ -7,6 +7,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstring>
+#include <memory>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -49,6 +50,12 @@ struct MoannResult {
     void* owner = nullptr;        /* SearchCtx* (IVF lib) or null          */
     hipEvent_t done_ev = nullptr; /* owned by the result; null => sync    */
     bool collected = true;
+    /* extend support: the tail sub-index's (already collected) result,
+     * top-k merged into this one at collect (gpu_ivf_flat_extend) */
+    std::unique_ptr<MoannResult> tail_res;
+    int merge_kmetric = 0;   /* KM_* of the owning index (IP merge order) */
+    uint32_t merge_limit = 0; /* row stride (limit) for the tail merge */
+    bool tail_merged = false;
 };
 
 /* grow-only device buffer (cuvs_worker.hpp:50-150 shape, minus RMM) */

@@ -128,6 +128,8 @@ struct SearchCtx {
     DevBuf w_rsel_slots, w_rsel_dists, w_refined, w_sel2; /* refine stage */
     void* h_out = nullptr;   /* pinned [ids int64][dists f32] bounce */
     size_t h_out_bytes = 0;
+    std::vector<uint32_t> h_filter; /* staged filter words: must outlive the
+        async H2D when the submit path returns before the copy runs */
     MoannResult* pending = nullptr; /* submitted, not yet collected */
     void* parent = nullptr;         /* IvfIndex*, for perf accumulation */
     void ensure(void* ix) {
@@ -243,6 +245,7 @@ struct IvfIndex {
     std::vector<float> tail_staging;   /* f32 rows, extend order */
     std::vector<int64_t> tail_ids;
     std::vector<int32_t> tail_assign;
+    std::vector<int64_t> tail_deleted; /* deletes re-applied on rebuild */
 
     /* soft-delete bitset (index_base.hpp:183-209): bit per SLOT, 1 = alive,
      * lazily allocated on first delete. Applied to every search by ANDing
@@ -358,6 +361,94 @@ void pack_lists(IvfIndex* ix, const float* d_rows, int64_t nrows,
     *d_packed_out = d_packed;
     if (gbase_out) *gbase_out = std::move(gbase);
     *ngroups_out = ngroups;
+}
+
+/* ---------------- host-side element conversions (ABI edges) -------------- */
+
+float half_to_f32(uint16_t b) {
+    const uint32_t s = (uint32_t)(b >> 15) << 31;
+    uint32_t e = (b >> 10) & 0x1f, m = b & 0x3ff;
+    uint32_t out;
+    if (e == 0) {
+        if (m == 0) out = s;
+        else { /* subnormal */
+            e = 127 - 14;
+            while (!(m & 0x400)) { m <<= 1; --e; }
+            m &= 0x3ff;
+            out = s | (e << 23) | (m << 13);
+        }
+    } else if (e == 31) {
+        out = s | 0x7f800000u | (m << 13);
+    } else {
+        out = s | ((e + 112) << 23) | (m << 13);
+    }
+    float f;
+    memcpy(&f, &out, 4);
+    return f;
+}
+
+float bf16_to_f32(uint16_t b) {
+    const uint32_t out = (uint32_t)b << 16;
+    float f;
+    memcpy(&f, &out, 4);
+    return f;
+}
+
+/* storage-typed rows -> f32 staging rows. The affine byte map round-trips
+ * exactly (|code| <= 255: dequant->requant errors are ~1e-5 code steps, far
+ * inside the round-to-nearest window), halves are exact casts — so staging
+ * narrow input as f32 and re-quantizing at build reproduces the caller's
+ * bytes bit-for-bit. */
+void storage_to_f32(quantization_t st, const void* src, size_t n,
+                    double mul, double add, float* dst) {
+    switch (st) {
+    case Quantization_F32:
+        memcpy(dst, src, n * 4);
+        break;
+    case Quantization_INT8: {
+        const int8_t* p = (const int8_t*)src;
+        const double m = mul == 0.0 ? 1.0 : mul;
+        for (size_t i = 0; i < n; ++i)
+            dst[i] = (float)(((double)p[i] - add) / m);
+        break;
+    }
+    case Quantization_UINT8: {
+        const uint8_t* p = (const uint8_t*)src;
+        const double m = mul == 0.0 ? 1.0 : mul;
+        for (size_t i = 0; i < n; ++i)
+            dst[i] = (float)(((double)p[i] - add) / m);
+        break;
+    }
+    case Quantization_F16: {
+        const uint16_t* p = (const uint16_t*)src;
+        for (size_t i = 0; i < n; ++i) dst[i] = half_to_f32(p[i]);
+        break;
+    }
+    case Quantization_BF16: {
+        const uint16_t* p = (const uint16_t*)src;
+        for (size_t i = 0; i < n; ++i) dst[i] = bf16_to_f32(p[i]);
+        break;
+    }
+    default:
+        throw std::runtime_error("unsupported storage type");
+    }
+}
+
+size_t storage_elem_bytes(quantization_t st) {
+    return st == Quantization_F32 ? 4
+           : (st == Quantization_F16 || st == Quantization_BF16) ? 2 : 1;
+}
+
+/* base-typed queries (B = f32 or f16, ivf_flat_c.h:33-35) -> f32 */
+std::vector<float> base_to_f32(quantization_t btype, const void* src,
+                               size_t n) {
+    std::vector<float> out(n);
+    if (btype == Quantization_F32) memcpy(out.data(), src, n * 4);
+    else if (btype == Quantization_F16) {
+        const uint16_t* p = (const uint16_t*)src;
+        for (size_t i = 0; i < n; ++i) out[i] = half_to_f32(p[i]);
+    } else throw std::runtime_error("base type must be F32 or F16");
+    return out;
 }
 
 /* ------------------------- the search pipeline -------------------------- */
@@ -625,9 +716,10 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const uint32_t* d_filter = nullptr;
     if (filter_words) {
         const size_t fw = (size_t)((ix->count + 31) / 32);
+        cx->h_filter.assign(filter_words, filter_words + fw);
         cx->w_filter.ensure(fw * 4);
-        HIP_CHECK(hipMemcpyAsync(cx->w_filter.ptr, filter_words, fw * 4,
-                                 hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipMemcpyAsync(cx->w_filter.ptr, cx->h_filter.data(),
+                                 fw * 4, hipMemcpyHostToDevice, s));
         d_filter = cx->w_filter.as<uint32_t>();
     }
     cx->w_cand.ensure(std::max<int64_t>(1, total_cand) * 4);
@@ -825,6 +917,51 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
  * vectors and fold the kernel event windows into the perf counters.
  * Serialized by the submit/wait discipline (the submit path collects a
  * ctx's pending result before reusing it). */
+/* Top-k merge of an extend tail's (collected) result into the main result.
+ * Both rows are ordered by closeness (cpu_topk_merge_sharded semantics,
+ * cgo/cuvs/helper.h:137-167): ascending distance, except post-flip IP where
+ * the C-ABI value is +a.b and closeness is DESCENDING value. Pads (-1,
+ * FLT_MAX) sort last. */
+void merge_tail_into(MoannResult* res) {
+    if (!res->tail_res || res->tail_merged) return;
+    MoannResult* t = res->tail_res.get();
+    const size_t total = res->neighbors.size();
+    if (t->neighbors.size() != total || total == 0) { res->tail_merged = true; return; }
+    /* limit is carried implicitly: both results are [nq][limit]; recover
+     * nq from the merge caller's perspective via the stored stride */
+    const uint32_t limit = res->merge_limit;
+    const size_t nq = limit ? total / limit : 0;
+    const bool ip = res->merge_kmetric == KM_IP;
+    auto closer = [&](float a, float b) { return ip ? a > b : a < b; };
+    std::vector<int64_t> mi(limit);
+    std::vector<float> md(limit);
+    for (size_t q = 0; q < nq; ++q) {
+        const int64_t* an = res->neighbors.data() + q * limit;
+        const float* ad = res->distances.data() + q * limit;
+        const int64_t* bn = t->neighbors.data() + q * limit;
+        const float* bd = t->distances.data() + q * limit;
+        uint32_t i = 0, j = 0, o = 0;
+        while (o < limit) {
+            const bool av = i < limit && an[i] != -1;
+            const bool bv = j < limit && bn[j] != -1;
+            if (!av && !bv) break;
+            if (av && (!bv || closer(ad[i], bd[j]) ||
+                       (ad[i] == bd[j] && true /* main-first tie order,
+                           matching helper.h's stable shard order */))) {
+                mi[o] = an[i]; md[o] = ad[i]; ++i;
+            } else {
+                mi[o] = bn[j]; md[o] = bd[j]; ++j;
+            }
+            ++o;
+        }
+        for (; o < limit; ++o) { mi[o] = -1; md[o] = FLT_MAX; }
+        memcpy(res->neighbors.data() + q * limit, mi.data(), limit * 8);
+        memcpy(res->distances.data() + q * limit, md.data(), limit * 4);
+    }
+    res->tail_merged = true;
+    res->tail_res.reset();
+}
+
 double collect_result(MoannResult* res) {
     if (res->collected) return 0.0;
     auto cx = (SearchCtx*)res->owner;
@@ -856,7 +993,73 @@ double collect_result(MoannResult* res) {
     ix->perf.c.select_ms += sel_ms;
     res->collected = true;
     if (cx->pending == res) cx->pending = nullptr;
+    merge_tail_into(res);
     return (double)r_ms + sc_ms + sel_ms;
+}
+
+uint64_t total_len(IvfIndex* ix) {
+    return ix->count + (ix->tail ? ix->tail->count : 0);
+}
+
+/* AND the caller's slot bitset (over [0, nbits) slots) with the alive
+ * bitset of `ix` over its own slot range starting at `slot0` in the global
+ * numbering; returns words covering ix->count bits (bit extraction handles
+ * the unaligned tail region). nullptr user = all-pass. */
+std::vector<uint32_t> combined_filter(IvfIndex* ix, uint64_t slot0,
+                                      const uint32_t* user, uint64_t nbits) {
+    const uint64_t n = ix->count;
+    std::vector<uint32_t> out((size_t)((n + 31) / 32), ~0u);
+    if (user) {
+        for (uint64_t i = 0; i < n; ++i) {
+            const uint64_t g = slot0 + i;
+            const bool bit = g < nbits && ((user[g / 32] >> (g % 32)) & 1u);
+            if (!bit) out[i / 32] &= ~(1u << (i % 32));
+        }
+    }
+    if (ix->deleted_count) {
+        const size_t aw = ix->alive.size();
+        for (size_t w = 0; w < out.size() && w < aw; ++w)
+            out[w] &= ix->alive[w];
+    }
+    return out;
+}
+
+/* Search main + (if extends happened) the tail sub-index, with the soft-
+ * delete bitset and any caller filter folded in; tail top-k is merged at
+ * collect (or immediately for the synchronous path). This is the single
+ * funnel every C search entry routes through. */
+std::unique_ptr<Result> run_search_full(IvfIndex* ix, const float* queries,
+                                        bool queries_on_device, uint64_t nq,
+                                        uint32_t limit, uint32_t n_probes,
+                                        const uint32_t* user_filter = nullptr,
+                                        uint64_t user_nbits = 0,
+                                        bool sync_end = true) {
+    const bool has_tail = ix->tail && ix->tail->count > 0;
+    const bool need_main_filter = user_filter || ix->deleted_count > 0;
+    std::vector<uint32_t> mf;
+    if (need_main_filter)
+        mf = combined_filter(ix, 0, user_filter, user_nbits);
+
+    /* tail first (synchronous, small), so the submit path can return after
+     * enqueueing the main scan */
+    std::unique_ptr<Result> tres;
+    if (has_tail) {
+        IvfIndex* t = ix->tail.get();
+        std::vector<uint32_t> tf;
+        const bool need_tf = user_filter || t->deleted_count > 0;
+        if (need_tf) tf = combined_filter(t, ix->count, user_filter, user_nbits);
+        tres = run_search(t, queries, queries_on_device, nq, limit, n_probes,
+                          need_tf ? tf.data() : nullptr, /*sync_end=*/true);
+    }
+    auto res = run_search(ix, queries, queries_on_device, nq, limit, n_probes,
+                          need_main_filter ? mf.data() : nullptr, sync_end);
+    res->merge_kmetric = ix->kmetric;
+    res->merge_limit = limit;
+    if (tres) {
+        res->tail_res = std::move(tres);
+        if (res->collected) merge_tail_into(res.get());
+    }
+    return res;
 }
 
 void batcher_loop(IvfIndex* ix) {
@@ -890,8 +1093,8 @@ void batcher_loop(IvfIndex* ix) {
             std::unique_ptr<Result> merged;
             std::string err;
             try {
-                merged = run_search(ix, qcat.data(), false, total, limit,
-                                    nprobe);
+                merged = run_search_full(ix, qcat.data(), false, total,
+                                         limit, nprobe);
             } catch (const std::exception& e) {
                 err = e.what();
             }
@@ -933,14 +1136,22 @@ gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
                                       quantization_t btype, quantization_t qtype,
                                       const int64_t* /*ids*/, void* errmsg) {
     try {
-        if (btype != Quantization_F32 ||
-            (qtype != Quantization_F32 && qtype != Quantization_INT8 &&
-             qtype != Quantization_UINT8 && qtype != Quantization_F16 &&
-             qtype != Quantization_BF16))
+        /* wired btype/qtype combos mirror ivf_flat_c.h:33-35 */
+        const bool ok_f32_base =
+            btype == Quantization_F32 &&
+            (qtype == Quantization_F32 || qtype == Quantization_INT8 ||
+             qtype == Quantization_UINT8 || qtype == Quantization_F16 ||
+             qtype == Quantization_BF16);
+        const bool ok_f16_base =
+            btype == Quantization_F16 &&
+            (qtype == Quantization_F16 || qtype == Quantization_INT8 ||
+             qtype == Quantization_UINT8);
+        if (!ok_f32_base && !ok_f16_base)
             throw std::runtime_error(
-                "wired combos: F32 base with F32/F16/BF16/INT8/UINT8 "
-                "storage");
+                "wired combos (ivf_flat_c.h:33-35): F32 base "
+                "{F32,F16,BF16,INT8,UINT8}; F16 base {F16,INT8,UINT8}");
         auto ix = std::make_unique<IvfIndex>();
+        ix->btype = btype;
         ix->device = (devices && device_count > 0) ? devices[0] : 0;
         HIP_CHECK(hipSetDevice(ix->device));
         HIP_CHECK(hipStreamCreate(&ix->stream));
@@ -965,25 +1176,72 @@ gpu_ivf_flat_c gpu_ivf_flat_new_empty(uint64_t total_count, uint32_t dimension,
     }
 }
 
+/* common staging path: f32 rows into d_staging (quantized at build) */
+void stage_rows_f32(IvfIndex* ix, const float* rows, uint64_t chunk_count,
+                    const int64_t* ids) {
+    if (ix->built) throw std::runtime_error("index already built");
+    if (ix->count + chunk_count > ix->cap)
+        throw std::runtime_error("chunk exceeds capacity");
+    HIP_CHECK(hipSetDevice(ix->device));
+    HIP_CHECK(hipMemcpyAsync(ix->d_staging + ix->count * ix->dim, rows,
+                             chunk_count * ix->dim * 4,
+                             hipMemcpyHostToDevice, ix->stream));
+    HIP_CHECK(hipStreamSynchronize(ix->stream));
+    for (uint64_t i = 0; i < chunk_count; ++i)
+        ix->h_ids.push_back(ids ? ids[i] : (int64_t)(ix->count + i));
+    ix->count += chunk_count;
+}
+
+/* ivf_flat_c.h:69: chunk in the STORAGE type T (widened losslessly to the
+ * f32 staging; build re-narrows to the identical codes). */
 void gpu_ivf_flat_add_chunk(gpu_ivf_flat_c h, const void* chunk_data,
                             uint64_t chunk_count, const int64_t* ids,
                             void* errmsg) {
     try {
         auto ix = IX(h);
         std::lock_guard<std::mutex> lk(ix->mu);
-        if (ix->built) throw std::runtime_error("index already built");
-        if (ix->count + chunk_count > ix->cap)
-            throw std::runtime_error("chunk exceeds capacity");
-        HIP_CHECK(hipSetDevice(ix->device));
-        HIP_CHECK(hipMemcpyAsync(ix->d_staging + ix->count * ix->dim,
-                                 chunk_data, chunk_count * ix->dim * 4,
-                                 hipMemcpyHostToDevice, ix->stream));
-        HIP_CHECK(hipStreamSynchronize(ix->stream));
-        for (uint64_t i = 0; i < chunk_count; ++i)
-            ix->h_ids.push_back(ids ? ids[i] : (int64_t)(ix->count + i));
-        ix->count += chunk_count;
+        if (ix->qtype == Quantization_F32) {
+            stage_rows_f32(ix, (const float*)chunk_data, chunk_count, ids);
+        } else {
+            std::vector<float> conv((size_t)chunk_count * ix->dim);
+            storage_to_f32(ix->qtype, chunk_data, conv.size(), ix->quant_mul,
+                           ix->quant_add, conv.data());
+            stage_rows_f32(ix, conv.data(), chunk_count, ids);
+        }
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_add_chunk", e.what());
+    }
+}
+
+/* ivf_flat_c.h:80: f32 chunk, narrowed to storage at build */
+void gpu_ivf_flat_add_chunk_float(gpu_ivf_flat_c h, const float* chunk_data,
+                                  uint64_t chunk_count, const int64_t* ids,
+                                  void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        stage_rows_f32(ix, chunk_data, chunk_count, ids);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_add_chunk_float", e.what());
+    }
+}
+
+/* ivf_flat_c.h:83: chunk in the BASE type B */
+void gpu_ivf_flat_add_chunk_quantize(gpu_ivf_flat_c h, const void* base_data,
+                                     uint64_t chunk_count,
+                                     const int64_t* ids, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (ix->btype == Quantization_F32) {
+            stage_rows_f32(ix, (const float*)base_data, chunk_count, ids);
+        } else {
+            auto conv = base_to_f32(ix->btype, base_data,
+                                    (size_t)chunk_count * ix->dim);
+            stage_rows_f32(ix, conv.data(), chunk_count, ids);
+        }
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_add_chunk_quantize", e.what());
     }
 }
 
@@ -1218,6 +1476,10 @@ void gpu_ivf_flat_destroy(gpu_ivf_flat_c h, void* errmsg) {
     }
 }
 
+/* ivf_flat_c.h:114-116: queries in the STORAGE type T (the reference's
+ * search<T> signature); narrow input is widened host-side and re-narrowed
+ * by the internal pipeline — an exact round trip, so results equal a
+ * native byte-domain scan of the caller's codes. */
 gpu_ivf_flat_search_res_t gpu_ivf_flat_search(gpu_ivf_flat_c h,
                                               const void* queries_data,
                                               uint64_t num_queries,
@@ -1230,11 +1492,45 @@ gpu_ivf_flat_search_res_t gpu_ivf_flat_search(gpu_ivf_flat_c h,
         auto ix = IX(h);
         if (query_dimension != ix->dim)
             throw std::runtime_error("query dimension mismatch");
-        out.result_ptr = run_search(ix, (const float*)queries_data, false,
-                                    num_queries, limit, sp.n_probes)
+        const float* q = (const float*)queries_data;
+        std::vector<float> conv;
+        if (ix->qtype != Quantization_F32) {
+            conv.resize((size_t)num_queries * ix->dim);
+            storage_to_f32(ix->qtype, queries_data, conv.size(),
+                           ix->quant_mul, ix->quant_add, conv.data());
+            q = conv.data();
+        }
+        out.result_ptr = run_search_full(ix, q, false, num_queries, limit,
+                                         sp.n_probes)
                              .release();
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_search", e.what());
+    }
+    return out;
+}
+
+/* ivf_flat_c.h:119-122: queries in the BASE type B. */
+gpu_ivf_flat_search_res_t gpu_ivf_flat_search_quantize(
+    gpu_ivf_flat_c h, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit, ivf_flat_search_params_t sp,
+    void* errmsg) {
+    gpu_ivf_flat_search_res_t out {nullptr};
+    try {
+        auto ix = IX(h);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        const float* q = (const float*)queries_data;
+        std::vector<float> conv;
+        if (ix->btype != Quantization_F32) {
+            conv = base_to_f32(ix->btype, queries_data,
+                               (size_t)num_queries * ix->dim);
+            q = conv.data();
+        }
+        out.result_ptr = run_search_full(ix, q, false, num_queries, limit,
+                                         sp.n_probes)
+                             .release();
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_search_quantize", e.what());
     }
     return out;
 }
@@ -1251,8 +1547,8 @@ gpu_ivf_flat_search_res_t moann_ivf_flat_search_device(gpu_ivf_flat_c h,
         auto ix = IX(h);
         if (query_dimension != ix->dim)
             throw std::runtime_error("query dimension mismatch");
-        out.result_ptr = run_search(ix, (const float*)queries_dev, true,
-                                    num_queries, limit, sp.n_probes)
+        out.result_ptr = run_search_full(ix, (const float*)queries_dev, true,
+                                         num_queries, limit, sp.n_probes)
                              .release();
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "moann_ivf_flat_search_device", e.what());
@@ -1275,10 +1571,10 @@ gpu_ivf_flat_search_res_t moann_ivf_flat_search_submit(
         auto ix = IX(h);
         if (query_dimension != ix->dim)
             throw std::runtime_error("query dimension mismatch");
-        out.result_ptr = run_search(ix, (const float*)queries,
-                                    queries_on_device != 0, num_queries,
-                                    limit, sp.n_probes, nullptr,
-                                    /*sync_end=*/false)
+        out.result_ptr = run_search_full(ix, (const float*)queries,
+                                         queries_on_device != 0, num_queries,
+                                         limit, sp.n_probes, nullptr, 0,
+                                         /*sync_end=*/false)
                              .release();
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "moann_ivf_flat_search_submit", e.what());
@@ -1296,14 +1592,11 @@ int moann_ivf_flat_result_wait(gpu_ivf_flat_result_c r, void* errmsg) {
     }
 }
 
-uint64_t gpu_ivf_flat_search_async(gpu_ivf_flat_c h, const void* queries_data,
-                                   uint64_t num_queries,
-                                   uint32_t query_dimension, uint32_t limit,
-                                   ivf_flat_search_params_t sp, void* errmsg) {
-    try {
-        auto ix = IX(h);
-        if (query_dimension != ix->dim)
-            throw std::runtime_error("query dimension mismatch");
+static uint64_t async_f32_impl(IvfIndex* ix, const float* queries_f32,
+                               uint64_t num_queries, uint32_t limit,
+                               ivf_flat_search_params_t sp) {
+    {
+        const void* queries_data = queries_f32;
         if (ix->batch_window_us.load() > 0) {
             /* dynamic batching: enqueue and let the batcher fuse requests
              * arriving within the window into one device search */
@@ -1326,14 +1619,62 @@ uint64_t gpu_ivf_flat_search_async(gpu_ivf_flat_c h, const void* queries_data,
             ix->bcv.notify_one();
             return id;
         }
-        auto res = run_search(ix, (const float*)queries_data, false,
-                              num_queries, limit, sp.n_probes);
+        auto res = run_search_full(ix, (const float*)queries_data, false,
+                                   num_queries, limit, sp.n_probes);
         std::lock_guard<std::mutex> lk(ix->mu);
         const uint64_t id = ix->next_job++;
         ix->jobs[id] = std::move(res);
         return id;
+    }
+}
+
+uint64_t gpu_ivf_flat_search_async(gpu_ivf_flat_c h, const void* queries_data,
+                                   uint64_t num_queries,
+                                   uint32_t query_dimension, uint32_t limit,
+                                   ivf_flat_search_params_t sp, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        /* storage-typed queries like gpu_ivf_flat_search */
+        std::vector<float> conv;
+        const float* q = (const float*)queries_data;
+        if (ix->qtype != Quantization_F32) {
+            conv.resize((size_t)num_queries * ix->dim);
+            storage_to_f32(ix->qtype, queries_data, conv.size(),
+                           ix->quant_mul, ix->quant_add, conv.data());
+            q = conv.data();
+        }
+        return async_f32_impl(ix, q, num_queries, limit, sp);
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_search_async", e.what());
+        return 0;
+    }
+}
+
+/* ivf_flat_c.h:130-133: async pair in the BASE type B (the entry behind
+ * pkg/cuvs/ivf_flat.go:772 SearchQuantizeAsync — call stack C). */
+uint64_t gpu_ivf_flat_search_quantize_async(gpu_ivf_flat_c h,
+                                            const void* queries_data,
+                                            uint64_t num_queries,
+                                            uint32_t query_dimension,
+                                            uint32_t limit,
+                                            ivf_flat_search_params_t sp,
+                                            void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        const float* q = (const float*)queries_data;
+        std::vector<float> conv;
+        if (ix->btype != Quantization_F32) {
+            conv = base_to_f32(ix->btype, queries_data,
+                               (size_t)num_queries * ix->dim);
+            q = conv.data();
+        }
+        return async_f32_impl(ix, q, num_queries, limit, sp);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_search_quantize_async", e.what());
         return 0;
     }
 }
@@ -1452,7 +1793,7 @@ int moann_ivf_flat_filter_from_docfilter(gpu_ivf_flat_c h, int kind,
 }
 
 uint64_t gpu_ivf_flat_cap(gpu_ivf_flat_c h) { return IX(h)->cap; }
-uint64_t gpu_ivf_flat_len(gpu_ivf_flat_c h) { return IX(h)->count; }
+uint64_t gpu_ivf_flat_len(gpu_ivf_flat_c h) { return total_len(IX(h)); }
 uint32_t gpu_ivf_flat_get_n_list(gpu_ivf_flat_c h) { return IX(h)->nlist; }
 
 void gpu_ivf_flat_get_centers(gpu_ivf_flat_c h, void* centers, void* errmsg) {
@@ -1526,11 +1867,11 @@ gpu_ivf_flat_search_res_t moann_ivf_flat_search_filtered(
         auto ix = IX(h);
         if (query_dimension != ix->dim)
             throw std::runtime_error("query dimension mismatch");
-        if (nbits < ix->count)
+        if (nbits < total_len(ix))
             throw std::runtime_error("filter bitset smaller than index");
-        out.result_ptr = run_search(ix, (const float*)queries_data, false,
-                                    num_queries, limit, sp.n_probes,
-                                    slot_bitset)
+        out.result_ptr = run_search_full(ix, (const float*)queries_data,
+                                         false, num_queries, limit,
+                                         sp.n_probes, slot_bitset, nbits)
                              .release();
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "moann_ivf_flat_search_filtered", e.what());
@@ -1552,12 +1893,501 @@ void moann_ivf_flat_slots_of_ids(gpu_ivf_flat_c h, const int64_t* ids,
             for (uint64_t sidx = 0; sidx < ix->count; ++sidx)
                 ix->id_to_slot.emplace(ix->h_id_by_slot[sidx], (int64_t)sidx);
         }
+        IvfIndex* t = ix->tail.get();
+        if (t && t->count && t->id_to_slot.empty()) { /* lazy tail map */
+            t->id_to_slot.reserve(t->count * 2);
+            for (uint64_t sidx = 0; sidx < t->count; ++sidx)
+                t->id_to_slot.emplace(t->h_id_by_slot[sidx], (int64_t)sidx);
+        }
         for (uint64_t i = 0; i < n; ++i) {
             auto it = ix->id_to_slot.find(ids[i]);
-            out_slots[i] = it == ix->id_to_slot.end() ? -1 : it->second;
+            if (it != ix->id_to_slot.end()) {
+                out_slots[i] = it->second;
+                continue;
+            }
+            out_slots[i] = -1;
+            if (t && t->count) {
+                auto jt = t->id_to_slot.find(ids[i]);
+                if (jt != t->id_to_slot.end())
+                    /* tail slots follow the main slots in the global
+                     * numbering (gpu_ivf_flat_extend) */
+                    out_slots[i] = (int64_t)ix->count + jt->second;
+            }
         }
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "moann_ivf_flat_slots_of_ids", e.what());
+    }
+}
+
+/* ---- dataset ctor (ivf_flat_c.h:41-45) ---------------------------------- */
+
+gpu_ivf_flat_c gpu_ivf_flat_new(const void* dataset_data,
+                                uint64_t count_vectors, uint32_t dimension,
+                                distance_type_t metric,
+                                ivf_flat_build_params_t build_params,
+                                const int* devices, int device_count,
+                                uint32_t nthread,
+                                distribution_mode_t dist_mode,
+                                quantization_t btype, quantization_t qtype,
+                                const int64_t* ids, void* errmsg) {
+    char* inner = nullptr;
+    gpu_ivf_flat_c h = gpu_ivf_flat_new_empty(
+        count_vectors, dimension, metric, build_params, devices,
+        device_count, nthread, dist_mode, btype, qtype, ids, &inner);
+    if (!h) {
+        set_errmsg(errmsg, "gpu_ivf_flat_new", inner ? inner : "new failed");
+        if (inner) free(inner);
+        return nullptr;
+    }
+    if (count_vectors)
+        gpu_ivf_flat_add_chunk(h, dataset_data, count_vectors, ids, &inner);
+    if (inner) {
+        set_errmsg(errmsg, "gpu_ivf_flat_new", inner);
+        free(inner);
+        gpu_ivf_flat_destroy(h, nullptr);
+        return nullptr;
+    }
+    return h;
+}
+
+/* ---- extend (ivf_flat_c.h:71-78) ---------------------------------------- */
+
+/* rebuild the tail sub-index from the accumulated extend staging (O(tail)
+ * per extend call; cheap while the tail is small relative to the corpus) */
+static void rebuild_tail(IvfIndex* ix) {
+    char* inner = nullptr;
+    auto chk = [&](const char* what) {
+        if (inner) {
+            std::string m = std::string(what) + ": " + inner;
+            free(inner);
+            inner = nullptr;
+            throw std::runtime_error(m);
+        }
+    };
+    const uint64_t n = ix->tail_ids.size();
+    ivf_flat_build_params_t bp {ix->nlist, true, 1.0};
+    int dev = ix->device;
+    gpu_ivf_flat_c th = gpu_ivf_flat_new_empty(
+        n, ix->dim, ix->metric, bp, &dev, 1, 0,
+        DistributionMode_SINGLE_GPU, ix->btype, ix->qtype, nullptr, &inner);
+    chk("tail new_empty");
+    if (!th) throw std::runtime_error("tail new_empty failed");
+    std::unique_ptr<IvfIndex> tix((IvfIndex*)th);
+    if (ix->qtype == Quantization_INT8 || ix->qtype == Quantization_UINT8) {
+        gpu_ivf_flat_set_quantizer(th, (float)ix->quant_min,
+                                   (float)ix->quant_max, &inner);
+        chk("tail set_quantizer");
+    }
+    gpu_ivf_flat_add_chunk_float(th, ix->tail_staging.data(), n,
+                                 ix->tail_ids.data(), &inner);
+    chk("tail add_chunk");
+    moann_ivf_flat_set_centroids(th, ix->h_centroids.data(), ix->nlist,
+                                 &inner);
+    chk("tail set_centroids");
+    moann_ivf_flat_set_assignments(th, ix->tail_assign.data(), n, &inner);
+    chk("tail set_assignments");
+    gpu_ivf_flat_build(th, &inner);
+    chk("tail build");
+    ix->tail = std::move(tix);
+    /* re-apply soft deletes that target tail rows */
+    for (int64_t id : ix->tail_deleted) {
+        IvfIndex* t = ix->tail.get();
+        if (t->id_to_slot.empty() && t->count) {
+            t->id_to_slot.reserve(t->count * 2);
+            for (uint64_t s2 = 0; s2 < t->count; ++s2)
+                t->id_to_slot.emplace(t->h_id_by_slot[s2], (int64_t)s2);
+        }
+        auto it = t->id_to_slot.find(id);
+        if (it == t->id_to_slot.end()) continue;
+        const uint64_t pos = (uint64_t)it->second;
+        const uint64_t nw = (t->count + 31) / 32;
+        if (t->alive.size() < nw) t->alive.resize(nw, ~0u);
+        if ((t->alive[pos / 32] >> (pos % 32)) & 1u) {
+            t->alive[pos / 32] &= ~(1u << (pos % 32));
+            ++t->deleted_count;
+        }
+    }
+}
+
+void gpu_ivf_flat_extend_float(gpu_ivf_flat_c h, const float* new_data,
+                               uint64_t n_rows, const int64_t* new_ids,
+                               void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (!ix->built)
+            throw std::runtime_error("extend: index not built"); /* the
+                reference throws too (index_base.hpp:2156) */
+        if (n_rows == 0) return;
+        /* assign to the nearest centroid under the index metric (the
+         * cuvs::ivf_flat::extend rule) via the one-shot brute force */
+        std::vector<int64_t> nearest(n_rows);
+        std::vector<float> ndist(n_rows);
+        char* inner = nullptr;
+        moann_brute_force_search(ix->h_centroids.data(), ix->nlist, ix->dim,
+                                 ix->metric, nullptr, new_data, n_rows, 1,
+                                 ix->device, nearest.data(), ndist.data(),
+                                 &inner);
+        if (inner) {
+            std::string m = std::string("extend assign: ") + inner;
+            free(inner);
+            throw std::runtime_error(m);
+        }
+        const uint64_t base = total_len(ix);
+        ix->tail_staging.insert(ix->tail_staging.end(), new_data,
+                                new_data + n_rows * ix->dim);
+        for (uint64_t i = 0; i < n_rows; ++i) {
+            ix->tail_ids.push_back(new_ids ? new_ids[i]
+                                           : (int64_t)(base + i));
+            ix->tail_assign.push_back((int32_t)nearest[i]);
+        }
+        rebuild_tail(ix);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_extend_float", e.what());
+    }
+}
+
+void gpu_ivf_flat_extend(gpu_ivf_flat_c h, const void* new_data,
+                         uint64_t n_rows, const int64_t* new_ids,
+                         void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (ix->qtype == Quantization_F32) {
+            gpu_ivf_flat_extend_float(h, (const float*)new_data, n_rows,
+                                      new_ids, errmsg);
+            return;
+        }
+        std::vector<float> conv((size_t)n_rows * ix->dim);
+        storage_to_f32(ix->qtype, new_data, conv.size(), ix->quant_mul,
+                       ix->quant_add, conv.data());
+        gpu_ivf_flat_extend_float(h, conv.data(), n_rows, new_ids, errmsg);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_extend", e.what());
+    }
+}
+
+/* ---- soft delete (ivf_flat_c.h:102; index_base.hpp:1142-1167) ----------- */
+
+void gpu_ivf_flat_delete_id(gpu_ivf_flat_c h, int64_t id, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (!ix->built) throw std::runtime_error("delete_id: index not built");
+        if (ix->id_to_slot.empty() && ix->count) {
+            ix->id_to_slot.reserve(ix->count * 2);
+            for (uint64_t sidx = 0; sidx < ix->count; ++sidx)
+                ix->id_to_slot.emplace(ix->h_id_by_slot[sidx],
+                                       (int64_t)sidx);
+        }
+        auto it = ix->id_to_slot.find(id);
+        if (it != ix->id_to_slot.end()) {
+            const uint64_t pos = (uint64_t)it->second;
+            const uint64_t nw = (ix->count + 31) / 32;
+            if (ix->alive.size() < nw) ix->alive.resize(nw, ~0u);
+            if ((ix->alive[pos / 32] >> (pos % 32)) & 1u) {
+                ix->alive[pos / 32] &= ~(1u << (pos % 32));
+                ++ix->deleted_count;
+            }
+            return;
+        }
+        IvfIndex* t = ix->tail.get();
+        if (t && t->count) {
+            if (t->id_to_slot.empty()) {
+                t->id_to_slot.reserve(t->count * 2);
+                for (uint64_t s2 = 0; s2 < t->count; ++s2)
+                    t->id_to_slot.emplace(t->h_id_by_slot[s2], (int64_t)s2);
+            }
+            auto jt = t->id_to_slot.find(id);
+            if (jt != t->id_to_slot.end()) {
+                const uint64_t pos = (uint64_t)jt->second;
+                const uint64_t nw = (t->count + 31) / 32;
+                if (t->alive.size() < nw) t->alive.resize(nw, ~0u);
+                if ((t->alive[pos / 32] >> (pos % 32)) & 1u) {
+                    t->alive[pos / 32] &= ~(1u << (pos % 32));
+                    ++t->deleted_count;
+                    ix->tail_deleted.push_back(id);
+                }
+                return;
+            }
+        }
+        /* unknown id: silent no-op (index_base.hpp:1147) */
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_delete_id", e.what());
+    }
+}
+
+/* ---- quantizer training (ivf_flat_c.h:86) ------------------------------- */
+
+void gpu_ivf_flat_train_quantizer(gpu_ivf_flat_c h, const float* train_data,
+                                  uint64_t n_samples, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (ix->qtype != Quantization_INT8 &&
+            ix->qtype != Quantization_UINT8)
+            throw std::runtime_error(
+                "storage has no affine quantizer (int8/uint8 only)");
+        if (n_samples == 0) throw std::runtime_error("empty training set");
+        /* central-99% quantile range, the cuVS scalar-quantizer default
+         * the reference trains with (quantize.hpp:60-66, cuvs
+         * sq_params.quantile = 0.99); the affine map behind
+         * set_quantizer keeps quantizer.go's formulas (the CPU-path
+         * parity anchor). Bounded sample for large trainsets. */
+        size_t n = (size_t)n_samples * ix->dim;
+        size_t stride = 1;
+        const size_t cap = 1u << 26;
+        if (n > cap) stride = (n + cap - 1) / cap;
+        std::vector<float> samp;
+        samp.reserve(n / stride + 1);
+        for (size_t i = 0; i < n; i += stride) samp.push_back(train_data[i]);
+        const size_t k = samp.size() / 200; /* 0.5% per side */
+        std::nth_element(samp.begin(), samp.begin() + k, samp.end());
+        const float lo = samp[k];
+        std::nth_element(samp.begin(), samp.end() - 1 - k, samp.end());
+        const float hi = samp[samp.size() - 1 - k];
+        gpu_ivf_flat_set_quantizer(h, lo, hi, errmsg);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_train_quantizer", e.what());
+    }
+}
+
+/* ---- info (ivf_flat_c.h:159; field set of index_base.hpp:2098-2117) ----- */
+
+char* gpu_ivf_flat_info(gpu_ivf_flat_c h, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        char buf[512];
+        snprintf(buf, sizeof buf,
+                 "{\"element_size\": %zu, \"dimension\": %u, \"metric\": %d, "
+                 "\"status\": \"%s\", \"capacity\": %llu, "
+                 "\"current_length\": %llu, \"dist_mode\": 0, "
+                 "\"has_ids\": true, \"devices\": [%d], \"n_lists\": %u, "
+                 "\"qtype\": %d, \"btype\": %d, \"deleted_count\": %llu}",
+                 storage_elem_bytes(ix->qtype), ix->dim, (int)ix->metric,
+                 ix->built ? "Loaded" : "Empty",
+                 (unsigned long long)ix->cap,
+                 (unsigned long long)total_len(ix), ix->device, ix->nlist,
+                 (int)ix->qtype, (int)ix->btype,
+                 (unsigned long long)(ix->deleted_count +
+                                      (ix->tail ? ix->tail->deleted_count
+                                                : 0)));
+        return strdup(buf);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_info", e.what());
+        return nullptr;
+    }
+}
+
+/* ---- dynamic-batching knob (ivf_flat_c.h:92) ---------------------------- */
+
+void gpu_ivf_flat_set_dynb_conservative_dispatch(gpu_ivf_flat_c h,
+                                                 bool enable, void* errmsg) {
+    try {
+        IX(h)->dynb_conservative = enable;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_set_dynb_conservative_dispatch",
+                   e.what());
+    }
+}
+
+/* ---- pre-filter INCLUDE columns (ivf_flat_c.h:165-199) ------------------ */
+
+void gpu_ivf_flat_set_filter_columns(gpu_ivf_flat_c h,
+                                     const char* col_meta_json,
+                                     uint64_t total_count, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (ix->built)
+            throw std::runtime_error("set_filter_columns must precede build");
+        if (!col_meta_json) throw std::runtime_error("null col_meta_json");
+        (void)total_count;
+        /* minimal parse: per column object, the "type" field
+         * (cagra_c.h:166-167: 0=i32, 1=i64, 2=f32, 3=f64, 4=u64) */
+        std::vector<int> types;
+        const char* p = col_meta_json;
+        while ((p = strstr(p, "\"type\""))) {
+            p += 6;
+            while (*p == ':' || *p == ' ') ++p;
+            if (*p < '0' || *p > '9')
+                throw std::runtime_error("malformed col_meta_json");
+            types.push_back(atoi(p));
+        }
+        if (types.empty())
+            throw std::runtime_error("col_meta_json declares no columns");
+        for (int t : types)
+            if (t < 0 || t > 4)
+                throw std::runtime_error("unknown column type");
+        ix->filter_cols_json = col_meta_json;
+        ix->filter_col_types = std::move(types);
+        ix->filter_col_data.assign(ix->filter_col_types.size(), {});
+        ix->filter_col_nulls.assign(ix->filter_col_types.size(), {});
+        ix->filter_col_rows.assign(ix->filter_col_types.size(), 0);
+        ix->filter_col_has_nulls.assign(ix->filter_col_types.size(), false);
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_set_filter_columns", e.what());
+    }
+}
+
+void gpu_ivf_flat_add_filter_chunk(gpu_ivf_flat_c h, uint32_t col_idx,
+                                   const void* data,
+                                   const uint32_t* null_bitmap,
+                                   uint64_t nrows, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        std::lock_guard<std::mutex> lk(ix->mu);
+        if (ix->built)
+            throw std::runtime_error("add_filter_chunk must precede build");
+        if (col_idx >= ix->filter_col_types.size())
+            throw std::runtime_error("column index out of range");
+        static const size_t esz[5] = {4, 8, 4, 8, 8};
+        const size_t eb = esz[ix->filter_col_types[col_idx]];
+        auto& d = ix->filter_col_data[col_idx];
+        auto& nl = ix->filter_col_nulls[col_idx];
+        const uint64_t base = ix->filter_col_rows[col_idx];
+        d.insert(d.end(), (const uint8_t*)data,
+                 (const uint8_t*)data + nrows * eb);
+        /* null words are re-packed from the chunk base so chunks of any
+         * size concatenate (stored LSB-first over the FULL column) */
+        nl.resize((size_t)((base + nrows + 31) / 32), 0);
+        if (null_bitmap) {
+            ix->filter_col_has_nulls[col_idx] = true;
+            for (uint64_t i = 0; i < nrows; ++i)
+                if ((null_bitmap[i / 32] >> (i % 32)) & 1u) {
+                    const uint64_t g = base + i;
+                    nl[g / 32] |= (1u << (g % 32));
+                }
+        }
+        ix->filter_col_rows[col_idx] = base + nrows;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_add_filter_chunk", e.what());
+    }
+}
+
+/* slot bitset over [main + tail] from a preds_json over the registered
+ * columns; tail rows (post-build extends have no column rows) fail closed */
+static std::vector<uint32_t> preds_bitset(IvfIndex* ix,
+                                          const char* preds_json) {
+    if (ix->filter_col_types.empty())
+        throw std::runtime_error(
+            "no filter columns registered (gpu_ivf_flat_set_filter_columns)");
+    for (size_t c = 0; c < ix->filter_col_types.size(); ++c)
+        if (ix->filter_col_rows[c] != ix->count)
+            throw std::runtime_error("filter column rows != index rows");
+    std::vector<const void*> cdata(ix->filter_col_types.size());
+    std::vector<const uint32_t*> cnulls(ix->filter_col_types.size());
+    for (size_t c = 0; c < ix->filter_col_types.size(); ++c) {
+        cdata[c] = ix->filter_col_data[c].data();
+        cnulls[c] = ix->filter_col_has_nulls[c]
+                        ? ix->filter_col_nulls[c].data()
+                        : nullptr;
+    }
+    const uint64_t total = total_len(ix);
+    std::vector<uint32_t> words((size_t)((total + 31) / 32), 0);
+    char* inner = nullptr;
+    const int rc = moann_ivf_flat_filter_from_preds(
+        (gpu_ivf_flat_c)ix, ix->filter_cols_json.c_str(), preds_json,
+        cdata.data(), cnulls.data(), words.data(), &inner);
+    if (rc != 0 || inner) {
+        std::string m = inner ? inner : "predicate evaluation failed";
+        if (inner) free(inner);
+        throw std::runtime_error(m);
+    }
+    /* tail region (bits >= main count) stays 0: fail closed */
+    if (ix->count % 32) {
+        const uint64_t w = ix->count / 32;
+        words[w] &= (1u << (ix->count % 32)) - 1u;
+    }
+    return words;
+}
+
+gpu_ivf_flat_search_res_t gpu_ivf_flat_search_with_filter(
+    gpu_ivf_flat_c h, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit, ivf_flat_search_params_t sp,
+    const char* preds_json, void* errmsg) {
+    gpu_ivf_flat_search_res_t out {nullptr};
+    try {
+        auto ix = IX(h);
+        if (!preds_json || !*preds_json)
+            return gpu_ivf_flat_search(h, queries_data, num_queries,
+                                       query_dimension, limit, sp, errmsg);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        const float* q = (const float*)queries_data;
+        std::vector<float> conv;
+        if (ix->qtype != Quantization_F32) {
+            conv.resize((size_t)num_queries * ix->dim);
+            storage_to_f32(ix->qtype, queries_data, conv.size(),
+                           ix->quant_mul, ix->quant_add, conv.data());
+            q = conv.data();
+        }
+        auto words = preds_bitset(ix, preds_json);
+        out.result_ptr = run_search_full(ix, q, false, num_queries, limit,
+                                         sp.n_probes, words.data(),
+                                         total_len(ix))
+                             .release();
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_search_with_filter", e.what());
+    }
+    return out;
+}
+
+gpu_ivf_flat_search_res_t gpu_ivf_flat_search_quantize_with_filter(
+    gpu_ivf_flat_c h, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit, ivf_flat_search_params_t sp,
+    const char* preds_json, void* errmsg) {
+    gpu_ivf_flat_search_res_t out {nullptr};
+    try {
+        auto ix = IX(h);
+        if (!preds_json || !*preds_json)
+            return gpu_ivf_flat_search_quantize(h, queries_data, num_queries,
+                                                query_dimension, limit, sp,
+                                                errmsg);
+        if (query_dimension != ix->dim)
+            throw std::runtime_error("query dimension mismatch");
+        const float* q = (const float*)queries_data;
+        std::vector<float> conv;
+        if (ix->btype != Quantization_F32) {
+            conv = base_to_f32(ix->btype, queries_data,
+                               (size_t)num_queries * ix->dim);
+            q = conv.data();
+        }
+        auto words = preds_bitset(ix, preds_json);
+        out.result_ptr = run_search_full(ix, q, false, num_queries, limit,
+                                         sp.n_probes, words.data(),
+                                         total_len(ix))
+                             .release();
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_search_quantize_with_filter",
+                   e.what());
+    }
+    return out;
+}
+
+uint64_t gpu_ivf_flat_search_quantize_with_filter_async(
+    gpu_ivf_flat_c h, const void* queries_data, uint64_t num_queries,
+    uint32_t query_dimension, uint32_t limit, ivf_flat_search_params_t sp,
+    const char* preds_json, void* errmsg) {
+    try {
+        auto ix = IX(h);
+        if (!preds_json || !*preds_json)
+            return gpu_ivf_flat_search_quantize_async(
+                h, queries_data, num_queries, query_dimension, limit, sp,
+                errmsg);
+        /* filtered async executes at submit and parks the result for
+         * search_wait, like the unbatched async path */
+        auto res = gpu_ivf_flat_search_quantize_with_filter(
+            h, queries_data, num_queries, query_dimension, limit, sp,
+            preds_json, errmsg);
+        if (!res.result_ptr) return 0;
+        std::lock_guard<std::mutex> lk(ix->mu);
+        const uint64_t id = ix->next_job++;
+        ix->jobs[id] = std::unique_ptr<Result>((Result*)res.result_ptr);
+        return id;
+    } catch (const std::exception& e) {
+        set_errmsg(errmsg, "gpu_ivf_flat_search_quantize_with_filter_async",
+                   e.what());
+        return 0;
     }
 }
 
@@ -1573,6 +2403,11 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
         std::lock_guard<std::mutex> lk(ix->mu);
         if (!ix->built) throw std::runtime_error("index not built");
         HIP_CHECK(hipSetDevice(ix->device));
+        IvfIndex* tl = (ix->tail && ix->tail->count) ? ix->tail.get()
+                                                     : nullptr;
+        const uint64_t total = total_len(ix);
+        const uint64_t ndel =
+            ix->deleted_count + (tl ? tl->deleted_count : 0);
         std::string d(dir);
         auto wfile = [&](const std::string& name, const void* p, size_t n) {
             FILE* f = fopen((d + "/" + name).c_str(), "wb");
@@ -1591,23 +2426,60 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
                      "  \"metric\": %d,\n  \"count\": %llu,\n"
                      "  \"qtype\": %d,\n"
                      "  \"quant_min\": %.17g,\n  \"quant_max\": %.17g,\n"
-                     "  \"quant_mul\": %.17g\n}\n",
+                     "  \"quant_mul\": %.17g,\n"
+                     "  \"deleted_count\": %llu\n}\n",
                      ix->dim, ix->nlist, (int)ix->metric,
-                     (unsigned long long)ix->count, (int)ix->qtype,
-                     ix->quant_min, ix->quant_max, ix->quant_mul);
+                     (unsigned long long)total, (int)ix->qtype,
+                     ix->quant_min, ix->quant_max, ix->quant_mul,
+                     (unsigned long long)ndel);
             wfile("manifest.json", buf, strlen(buf));
         }
         wfile("centroids.f32", ix->h_centroids.data(),
               ix->h_centroids.size() * 4);
-        wfile("ids.i64", ix->h_id_by_slot.data(), ix->count * 8);
+        /* rows are saved main-slot-order then tail-slot-order; extends fold
+         * back into an ordinary (tail-free) index on load */
+        {
+            std::vector<int64_t> idsv(ix->h_id_by_slot.begin(),
+                                      ix->h_id_by_slot.begin() + ix->count);
+            if (tl)
+                idsv.insert(idsv.end(), tl->h_id_by_slot.begin(),
+                            tl->h_id_by_slot.begin() + tl->count);
+            wfile("ids.i64", idsv.data(), idsv.size() * 8);
+        }
         /* per-slot list id (assignments in slot order) */
         {
-            std::vector<int32_t> a(ix->count);
+            std::vector<int32_t> a(total);
             for (uint32_t l = 0; l < ix->nlist; ++l)
                 for (int64_t sidx = ix->list_slot_base[l];
                      sidx < ix->list_slot_base[l] + ix->list_rows[l]; ++sidx)
                     a[sidx] = (int32_t)l;
-            wfile("assign.i32", a.data(), ix->count * 4);
+            if (tl)
+                for (uint32_t l = 0; l < tl->nlist; ++l)
+                    for (int64_t sidx = tl->list_slot_base[l];
+                         sidx < tl->list_slot_base[l] + tl->list_rows[l];
+                         ++sidx)
+                        a[ix->count + sidx] = (int32_t)l;
+            wfile("assign.i32", a.data(), total * 4);
+        }
+        /* soft-delete bitset in SAVED row order (bit = 1 alive), only when
+         * deletions exist (index_base.hpp:246 "omitted if no deletions") */
+        if (ndel) {
+            std::vector<uint32_t> bits((size_t)((total + 31) / 32), ~0u);
+            auto fold = [&](IvfIndex* s, uint64_t base) {
+                for (size_t w = 0; w < s->alive.size(); ++w) {
+                    const uint32_t inv = ~s->alive[w];
+                    if (!inv) continue;
+                    for (int b = 0; b < 32; ++b)
+                        if ((inv >> b) & 1u) {
+                            const uint64_t g = base + w * 32 + b;
+                            if (g - base < s->count)
+                                bits[g / 32] &= ~(1u << (g % 32));
+                        }
+                }
+            };
+            fold(ix, 0);
+            if (tl) fold(tl, ix->count);
+            wfile("bitset.bin", bits.data(), bits.size() * 4);
         }
         if (ix->qtype != Quantization_F32) {
             /* narrow storage: the f32 staging was dropped at build; save the
@@ -1620,67 +2492,81 @@ void gpu_ivf_flat_save_dir(gpu_ivf_flat_c h, const char* dir, void* errmsg) {
             const size_t s_eb = (ix->qtype == Quantization_F16 ||
                                  ix->qtype == Quantization_BF16)
                                     ? 2 : 1; /* bytes per element */
-            DevBuf d_gsb, d_gv, d_tmp;
-            d_gsb.ensure(ix->total_groups * 8);
-            d_gv.ensure(ix->total_groups * 4);
-            HIP_CHECK(hipMemcpy(d_gsb.ptr, ix->group_slotbase.data(),
-                                ix->total_groups * 8, hipMemcpyHostToDevice));
-            HIP_CHECK(hipMemcpy(d_gv.ptr, ix->group_valid.data(),
-                                ix->total_groups * 4, hipMemcpyHostToDevice));
-            const int64_t CH = 1 << 21;
-            d_tmp.ensure((size_t)CH * ix->dim * s_eb);
-            std::vector<uint8_t> hbuf((size_t)CH * ix->dim * s_eb);
-            for (int64_t lo = 0; lo < (int64_t)ix->count; lo += CH) {
-                const int64_t hi = std::min<int64_t>(lo + CH, ix->count);
-                launch_unpack_bytes(ix->d_packed_q, (int)(ix->dim * s_eb),
-                                    (int)(ix->dpad * s_eb),
-                                    d_gsb.as<int64_t>(),
-                                    d_gv.as<int32_t>(), ix->total_groups, lo,
-                                    hi, d_tmp.as<uint8_t>(), ix->stream);
-                HIP_CHECK(hipMemcpyAsync(hbuf.data(), d_tmp.ptr,
-                                         (size_t)(hi - lo) * ix->dim * s_eb,
-                                         hipMemcpyDeviceToHost, ix->stream));
-                HIP_CHECK(hipStreamSynchronize(ix->stream));
-                HIP_CHECK(hipGetLastError());
-                const size_t n = (size_t)(hi - lo) * ix->dim * s_eb;
-                if (fwrite(hbuf.data(), 1, n, f) != n) {
-                    fclose(f);
-                    throw std::runtime_error("short write vectors.u8");
+            auto dump_bytes = [&](IvfIndex* s) {
+                DevBuf d_gsb, d_gv, d_tmp;
+                d_gsb.ensure(s->total_groups * 8);
+                d_gv.ensure(s->total_groups * 4);
+                HIP_CHECK(hipMemcpy(d_gsb.ptr, s->group_slotbase.data(),
+                                    s->total_groups * 8,
+                                    hipMemcpyHostToDevice));
+                HIP_CHECK(hipMemcpy(d_gv.ptr, s->group_valid.data(),
+                                    s->total_groups * 4,
+                                    hipMemcpyHostToDevice));
+                const int64_t CH = 1 << 21;
+                d_tmp.ensure((size_t)CH * s->dim * s_eb);
+                std::vector<uint8_t> hbuf((size_t)CH * s->dim * s_eb);
+                for (int64_t lo = 0; lo < (int64_t)s->count; lo += CH) {
+                    const int64_t hi = std::min<int64_t>(lo + CH, s->count);
+                    launch_unpack_bytes(
+                        s->d_packed_q, (int)(s->dim * s_eb),
+                        (int)(s->dpad * s_eb), d_gsb.as<int64_t>(),
+                        d_gv.as<int32_t>(), s->total_groups, lo, hi,
+                        d_tmp.as<uint8_t>(), s->stream);
+                    HIP_CHECK(hipMemcpyAsync(
+                        hbuf.data(), d_tmp.ptr,
+                        (size_t)(hi - lo) * s->dim * s_eb,
+                        hipMemcpyDeviceToHost, s->stream));
+                    HIP_CHECK(hipStreamSynchronize(s->stream));
+                    HIP_CHECK(hipGetLastError());
+                    const size_t n = (size_t)(hi - lo) * s->dim * s_eb;
+                    if (fwrite(hbuf.data(), 1, n, f) != n) {
+                        fclose(f);
+                        throw std::runtime_error("short write vectors.u8");
+                    }
                 }
-            }
+            };
+            dump_bytes(ix);
+            if (tl) dump_bytes(tl);
             fclose(f);
         } else
         /* vectors.f32: unpack in chunks of groups */
         {
             FILE* f = fopen((d + "/vectors.f32").c_str(), "wb");
             if (!f) throw std::runtime_error("cannot write vectors.f32");
-            DevBuf d_gsb, d_gv, d_tmp;
-            d_gsb.ensure(ix->total_groups * 8);
-            d_gv.ensure(ix->total_groups * 4);
-            HIP_CHECK(hipMemcpy(d_gsb.ptr, ix->group_slotbase.data(),
-                                ix->total_groups * 8, hipMemcpyHostToDevice));
-            HIP_CHECK(hipMemcpy(d_gv.ptr, ix->group_valid.data(),
-                                ix->total_groups * 4, hipMemcpyHostToDevice));
-            const int64_t CH = 1 << 20; /* slots per chunk */
-            d_tmp.ensure((size_t)CH * ix->dim * 4);
-            std::vector<float> hbuf((size_t)CH * ix->dim);
-            for (int64_t lo = 0; lo < (int64_t)ix->count; lo += CH) {
-                const int64_t hi = std::min<int64_t>(lo + CH, ix->count);
-                launch_unpack(ix->d_packed, (int)ix->dim, (int)ix->dpad,
-                              d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
-                              ix->total_groups, lo, hi, d_tmp.as<float>(),
-                              ix->stream);
-                HIP_CHECK(hipMemcpyAsync(hbuf.data(), d_tmp.ptr,
-                                         (size_t)(hi - lo) * ix->dim * 4,
-                                         hipMemcpyDeviceToHost, ix->stream));
-                HIP_CHECK(hipStreamSynchronize(ix->stream));
-                HIP_CHECK(hipGetLastError());
-                const size_t n = (size_t)(hi - lo) * ix->dim * 4;
-                if (fwrite(hbuf.data(), 1, n, f) != n) {
-                    fclose(f);
-                    throw std::runtime_error("short write vectors.f32");
+            auto dump_f32 = [&](IvfIndex* s) {
+                DevBuf d_gsb, d_gv, d_tmp;
+                d_gsb.ensure(s->total_groups * 8);
+                d_gv.ensure(s->total_groups * 4);
+                HIP_CHECK(hipMemcpy(d_gsb.ptr, s->group_slotbase.data(),
+                                    s->total_groups * 8,
+                                    hipMemcpyHostToDevice));
+                HIP_CHECK(hipMemcpy(d_gv.ptr, s->group_valid.data(),
+                                    s->total_groups * 4,
+                                    hipMemcpyHostToDevice));
+                const int64_t CH = 1 << 20; /* slots per chunk */
+                d_tmp.ensure((size_t)CH * s->dim * 4);
+                std::vector<float> hbuf((size_t)CH * s->dim);
+                for (int64_t lo = 0; lo < (int64_t)s->count; lo += CH) {
+                    const int64_t hi = std::min<int64_t>(lo + CH, s->count);
+                    launch_unpack(s->d_packed, (int)s->dim, (int)s->dpad,
+                                  d_gsb.as<int64_t>(), d_gv.as<int32_t>(),
+                                  s->total_groups, lo, hi,
+                                  d_tmp.as<float>(), s->stream);
+                    HIP_CHECK(hipMemcpyAsync(
+                        hbuf.data(), d_tmp.ptr,
+                        (size_t)(hi - lo) * s->dim * 4,
+                        hipMemcpyDeviceToHost, s->stream));
+                    HIP_CHECK(hipStreamSynchronize(s->stream));
+                    HIP_CHECK(hipGetLastError());
+                    const size_t n = (size_t)(hi - lo) * s->dim * 4;
+                    if (fwrite(hbuf.data(), 1, n, f) != n) {
+                        fclose(f);
+                        throw std::runtime_error("short write vectors.f32");
+                    }
                 }
-            }
+            };
+            dump_f32(ix);
+            if (tl) dump_f32(tl);
             fclose(f);
         }
     } catch (const std::exception& e) {
@@ -1825,16 +2711,207 @@ void gpu_ivf_flat_load_dir(gpu_ivf_flat_c h, const char* dir,
                     fclose(vf);
                     throw std::runtime_error("short read vectors.f32");
                 }
-                gpu_ivf_flat_add_chunk(h, vbuf.data(), n, ids.data() + done,
-                                       errmsg);
+                gpu_ivf_flat_add_chunk_float(h, vbuf.data(), n,
+                                             ids.data() + done, errmsg);
                 done += n;
             }
             fclose(vf);
             moann_ivf_flat_set_assignments(h, assign.data(), count, errmsg);
             gpu_ivf_flat_build(h, errmsg);
+            /* soft-delete bitset (saved in ROW order, bit = 1 alive):
+             * re-key to the rebuilt slot order via slot_rows */
+            FILE* bf = fopen((d + "/bitset.bin").c_str(), "rb");
+            if (bf) {
+                std::vector<uint32_t> bits((size_t)((count + 31) / 32));
+                const size_t got = fread(bits.data(), 4, bits.size(), bf);
+                fclose(bf);
+                if (got != bits.size())
+                    throw std::runtime_error("short read bitset.bin");
+                ix->alive.assign((size_t)((ix->count + 31) / 32), ~0u);
+                ix->deleted_count = 0;
+                for (uint64_t slot = 0; slot < ix->count; ++slot) {
+                    const uint64_t row = (uint64_t)ix->h_slot_rows[slot];
+                    const bool al = (bits[row / 32] >> (row % 32)) & 1u;
+                    if (!al) {
+                        ix->alive[slot / 32] &= ~(1u << (slot % 32));
+                        ++ix->deleted_count;
+                    }
+                }
+            }
         }
     } catch (const std::exception& e) {
         set_errmsg(errmsg, "gpu_ivf_flat_load_dir", e.what());
+    }
+}
+
+/* ---- single-file save/load (ivf_flat_c.h:95, 48-51) ---------------------
+ * Container: "MOANNPK1\n" then per blob [u32 name_len][name][u64 size]
+ * [payload], wrapping the save_dir blob set. Written/read only by this
+ * library (the reference's counterpart is cuVS-serialize, equally
+ * internal). */
+
+static const char* kPackMagic = "MOANNPK1\n";
+
+static std::string pack_tmpdir() {
+    char tmpl[] = "/tmp/moannpk.XXXXXX";
+    char* p = mkdtemp(tmpl);
+    if (!p) throw std::runtime_error("mkdtemp failed");
+    return std::string(p);
+}
+
+static void rm_rf_dir(const std::string& d,
+                      const std::vector<std::string>& names) {
+    for (const auto& n : names) (void)remove((d + "/" + n).c_str());
+    (void)remove(d.c_str());
+}
+
+static const std::vector<std::string> kPackNames = {
+    "manifest.json", "centroids.f32", "ids.i64", "assign.i32",
+    "bitset.bin", "vectors.f32", "vectors.u8"};
+
+void gpu_ivf_flat_save(gpu_ivf_flat_c h, const char* filename,
+                       void* errmsg) {
+    std::string tmp;
+    try {
+        tmp = pack_tmpdir();
+        char* inner = nullptr;
+        gpu_ivf_flat_save_dir(h, tmp.c_str(), &inner);
+        if (inner) {
+            std::string m(inner);
+            free(inner);
+            throw std::runtime_error(m);
+        }
+        FILE* out = fopen(filename, "wb");
+        if (!out) throw std::runtime_error("cannot write file");
+        fwrite(kPackMagic, 1, strlen(kPackMagic), out);
+        std::vector<uint8_t> buf(1 << 20);
+        for (const auto& n : kPackNames) {
+            FILE* f = fopen((tmp + "/" + n).c_str(), "rb");
+            if (!f) continue; /* optional blobs (bitset, one of vectors.*) */
+            fseek(f, 0, SEEK_END);
+            const uint64_t sz = (uint64_t)ftell(f);
+            fseek(f, 0, SEEK_SET);
+            const uint32_t nl = (uint32_t)n.size();
+            fwrite(&nl, 4, 1, out);
+            fwrite(n.data(), 1, nl, out);
+            fwrite(&sz, 8, 1, out);
+            uint64_t left = sz;
+            while (left) {
+                const size_t c = (size_t)std::min<uint64_t>(left, buf.size());
+                if (fread(buf.data(), 1, c, f) != c) {
+                    fclose(f);
+                    fclose(out);
+                    throw std::runtime_error("short read packing " + n);
+                }
+                fwrite(buf.data(), 1, c, out);
+                left -= c;
+            }
+            fclose(f);
+        }
+        fclose(out);
+        rm_rf_dir(tmp, kPackNames);
+    } catch (const std::exception& e) {
+        if (!tmp.empty()) rm_rf_dir(tmp, kPackNames);
+        set_errmsg(errmsg, "gpu_ivf_flat_save", e.what());
+    }
+}
+
+gpu_ivf_flat_c gpu_ivf_flat_load_file(const char* filename,
+                                      uint32_t dimension,
+                                      distance_type_t metric,
+                                      ivf_flat_build_params_t build_params,
+                                      const int* devices, int device_count,
+                                      uint32_t nthread,
+                                      distribution_mode_t dist_mode,
+                                      quantization_t btype,
+                                      quantization_t qtype, void* errmsg) {
+    std::string tmp;
+    gpu_ivf_flat_c h = nullptr;
+    try {
+        FILE* in = fopen(filename, "rb");
+        if (!in) throw std::runtime_error("cannot read file");
+        char magic[16] = {0};
+        if (fread(magic, 1, strlen(kPackMagic), in) != strlen(kPackMagic) ||
+            strncmp(magic, kPackMagic, strlen(kPackMagic)) != 0) {
+            fclose(in);
+            throw std::runtime_error("not a moann pack file");
+        }
+        tmp = pack_tmpdir();
+        std::vector<uint8_t> buf(1 << 20);
+        uint64_t count = 0;
+        for (;;) {
+            uint32_t nl = 0;
+            if (fread(&nl, 4, 1, in) != 1) break; /* EOF */
+            if (nl == 0 || nl > 64) {
+                fclose(in);
+                throw std::runtime_error("corrupt pack header");
+            }
+            std::string name(nl, 0);
+            uint64_t sz = 0;
+            if (fread(&name[0], 1, nl, in) != nl ||
+                fread(&sz, 8, 1, in) != 1) {
+                fclose(in);
+                throw std::runtime_error("corrupt pack entry");
+            }
+            if (name.find('/') != std::string::npos) {
+                fclose(in);
+                throw std::runtime_error("bad blob name");
+            }
+            FILE* f = fopen((tmp + "/" + name).c_str(), "wb");
+            if (!f) {
+                fclose(in);
+                throw std::runtime_error("cannot write temp blob");
+            }
+            uint64_t left = sz;
+            while (left) {
+                const size_t c = (size_t)std::min<uint64_t>(left, buf.size());
+                if (fread(buf.data(), 1, c, in) != c ||
+                    fwrite(buf.data(), 1, c, f) != c) {
+                    fclose(f);
+                    fclose(in);
+                    throw std::runtime_error("short pack payload");
+                }
+                left -= c;
+            }
+            fclose(f);
+            if (name == "manifest.json") {
+                /* capacity for the new_empty below */
+                FILE* mf = fopen((tmp + "/manifest.json").c_str(), "rb");
+                char mb[1024] = {0};
+                if (mf) {
+                    size_t r = fread(mb, 1, sizeof mb - 1, mf);
+                    (void)r;
+                    fclose(mf);
+                }
+                const char* fc = strstr(mb, "\"count\":");
+                if (fc) sscanf(fc, "\"count\": %llu",
+                               (unsigned long long*)&count);
+            }
+        }
+        fclose(in);
+        char* inner = nullptr;
+        h = gpu_ivf_flat_new_empty(count, dimension, metric, build_params,
+                                   devices, device_count, nthread,
+                                   dist_mode, btype, qtype, nullptr, &inner);
+        if (!h) {
+            std::string m = inner ? inner : "new_empty failed";
+            if (inner) free(inner);
+            throw std::runtime_error(m);
+        }
+        gpu_ivf_flat_start(h, &inner);
+        gpu_ivf_flat_load_dir(h, tmp.c_str(), dist_mode, &inner);
+        if (inner) {
+            std::string m(inner);
+            free(inner);
+            throw std::runtime_error(m);
+        }
+        rm_rf_dir(tmp, kPackNames);
+        return h;
+    } catch (const std::exception& e) {
+        if (h) gpu_ivf_flat_destroy(h, nullptr);
+        if (!tmp.empty()) rm_rf_dir(tmp, kPackNames);
+        set_errmsg(errmsg, "gpu_ivf_flat_load_file", e.what());
+        return nullptr;
     }
 }
 

@@ -68,8 +68,55 @@ def lib() -> ct.CDLL:
         ct.c_uint64, ct.c_uint32, ct.c_int, BuildParams,
         ct.POINTER(ct.c_int), ct.c_int, ct.c_uint32, ct.c_int, ct.c_int,
         ct.c_int, ct.c_void_p, ct.c_void_p]
-    L.gpu_ivf_flat_add_chunk.argtypes = [
-        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
+    L.gpu_ivf_flat_new.restype = ct.c_void_p
+    L.gpu_ivf_flat_new.argtypes = [
+        ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_int, BuildParams,
+        ct.POINTER(ct.c_int), ct.c_int, ct.c_uint32, ct.c_int, ct.c_int,
+        ct.c_int, ct.c_void_p, ct.c_void_p]
+    L.gpu_ivf_flat_load_file.restype = ct.c_void_p
+    L.gpu_ivf_flat_load_file.argtypes = [
+        ct.c_char_p, ct.c_uint32, ct.c_int, BuildParams,
+        ct.POINTER(ct.c_int), ct.c_int, ct.c_uint32, ct.c_int, ct.c_int,
+        ct.c_int, ct.c_void_p]
+    for f in (L.gpu_ivf_flat_add_chunk, L.gpu_ivf_flat_add_chunk_float,
+              L.gpu_ivf_flat_add_chunk_quantize, L.gpu_ivf_flat_extend,
+              L.gpu_ivf_flat_extend_float):
+        f.argtypes = [
+            ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p, ct.c_void_p]
+    L.gpu_ivf_flat_delete_id.argtypes = [ct.c_void_p, ct.c_int64, ct.c_void_p]
+    L.gpu_ivf_flat_train_quantizer.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_void_p]
+    L.gpu_ivf_flat_info.restype = ct.c_void_p  # char*, caller frees
+    L.gpu_ivf_flat_info.argtypes = [ct.c_void_p, ct.c_void_p]
+    L.gpu_ivf_flat_save.argtypes = [ct.c_void_p, ct.c_char_p, ct.c_void_p]
+    L.gpu_ivf_flat_set_dynb_conservative_dispatch.argtypes = [
+        ct.c_void_p, ct.c_bool, ct.c_void_p]
+    L.gpu_ivf_flat_set_filter_columns.argtypes = [
+        ct.c_void_p, ct.c_char_p, ct.c_uint64, ct.c_void_p]
+    L.gpu_ivf_flat_add_filter_chunk.argtypes = [
+        ct.c_void_p, ct.c_uint32, ct.c_void_p, ct.c_void_p, ct.c_uint64,
+        ct.c_void_p]
+    for f in (L.gpu_ivf_flat_search_with_filter,
+              L.gpu_ivf_flat_search_quantize_with_filter):
+        f.restype = SearchRes
+        f.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32,
+                      ct.c_uint32, SearchParams, ct.c_char_p, ct.c_void_p]
+    L.gpu_ivf_flat_search_quantize_with_filter_async.restype = ct.c_uint64
+    L.gpu_ivf_flat_search_quantize_with_filter_async.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
+        SearchParams, ct.c_char_p, ct.c_void_p]
+    L.moann_docfilter_parse.restype = ct.c_void_p
+    L.moann_docfilter_parse.argtypes = [ct.c_int, ct.c_void_p, ct.c_uint64,
+                                        ct.c_void_p]
+    L.moann_docfilter_test.restype = ct.c_int
+    L.moann_docfilter_test.argtypes = [ct.c_void_p, ct.c_void_p,
+                                       ct.c_uint64, ct.c_void_p, ct.c_void_p]
+    L.moann_docfilter_free.argtypes = [ct.c_void_p]
+    L.moann_hnsw_set_ef.argtypes = [ct.c_void_p, ct.c_uint32, ct.c_void_p]
+    L.usearchex_filtered_search_with_membership.restype = ct.c_size_t
+    L.usearchex_filtered_search_with_membership.argtypes = [
+        ct.c_void_p, ct.c_void_p, ct.c_int, ct.c_size_t, ct.c_void_p,
+        ct.c_int, ct.c_void_p, ct.c_void_p, ct.POINTER(ct.c_char_p)]
     L.moann_ivf_flat_set_centroids.argtypes = [
         ct.c_void_p, ct.c_void_p, ct.c_uint32, ct.c_void_p]
     L.moann_ivf_flat_set_assignments.argtypes = [
@@ -77,14 +124,17 @@ def lib() -> ct.CDLL:
     for f in (L.gpu_ivf_flat_build, L.gpu_ivf_flat_start):
         f.argtypes = [ct.c_void_p, ct.c_void_p]
     L.gpu_ivf_flat_destroy.argtypes = [ct.c_void_p, ct.c_void_p]
-    for f in (L.gpu_ivf_flat_search, L.moann_ivf_flat_search_device):
+    for f in (L.gpu_ivf_flat_search, L.gpu_ivf_flat_search_quantize,
+              L.moann_ivf_flat_search_device):
         f.restype = SearchRes
         f.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32,
                       ct.c_uint32, SearchParams, ct.c_void_p]
-    L.gpu_ivf_flat_search_async.restype = ct.c_uint64
-    L.gpu_ivf_flat_search_async.argtypes = [
-        ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
-        SearchParams, ct.c_void_p]
+    for f in (L.gpu_ivf_flat_search_async,
+              L.gpu_ivf_flat_search_quantize_async):
+        f.restype = ct.c_uint64
+        f.argtypes = [
+            ct.c_void_p, ct.c_void_p, ct.c_uint64, ct.c_uint32, ct.c_uint32,
+            SearchParams, ct.c_void_p]
     L.gpu_ivf_flat_search_wait.restype = SearchRes
     L.gpu_ivf_flat_search_wait.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p]
     L.gpu_ivf_flat_get_neighbors.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_void_p]
@@ -295,31 +345,106 @@ class IvfFlatIndex:
     QTYPES = {"f32": 0, "f16": 1, "int8": 2, "uint8": 3, "bf16": 4}  # quantization_t
 
     def __init__(self, dim: int, nlist: int, metric: str = "l2sq",
-                 capacity: int = 0, device: int = 0, qtype: str = "f32"):
+                 capacity: int = 0, device: int = 0, qtype: str = "f32",
+                 btype: str = "f32"):
         if metric not in METRICS:
             raise ValueError(f"metric {metric!r} not in {sorted(METRICS)}")
         self.dim, self.nlist, self.metric = dim, nlist, metric
         self.device = device
+        self.btype = btype
         err = _Err()
         bp = BuildParams(nlist, True, 0.5)
         dev = (ct.c_int * 1)(device)
         self._h = lib().gpu_ivf_flat_new_empty(
-            capacity, dim, METRICS[metric], bp, dev, 1, 0, 0, 0,
-            self.QTYPES[qtype], None, err.ref)
+            capacity, dim, METRICS[metric], bp, dev, 1, 0, 0,
+            self.QTYPES[btype], self.QTYPES[qtype], None, err.ref)
         err.check("new_empty")
         if not self._h:
             raise MoannError("gpu_ivf_flat_new_empty returned NULL")
 
     def add(self, vecs, ids=None):
+        """f32 rows -> gpu_ivf_flat_add_chunk_float (the storage-typed
+        gpu_ivf_flat_add_chunk takes T-typed rows per the reference)."""
         vecs = _as_f32(vecs)
         idp = None
         if ids is not None:
             ids = np.ascontiguousarray(ids, dtype=np.int64)
             idp = ids.ctypes.data
         err = _Err()
-        lib().gpu_ivf_flat_add_chunk(self._h, vecs.ctypes.data, vecs.shape[0],
-                                     idp, err.ref)
-        err.check("add_chunk")
+        lib().gpu_ivf_flat_add_chunk_float(self._h, vecs.ctypes.data,
+                                           vecs.shape[0], idp, err.ref)
+        err.check("add_chunk_float")
+
+    def extend(self, vecs, ids=None):
+        """Post-build append (gpu_ivf_flat_extend_float)."""
+        vecs = _as_f32(vecs)
+        idp = None
+        if ids is not None:
+            ids = np.ascontiguousarray(ids, dtype=np.int64)
+            idp = ids.ctypes.data
+        err = _Err()
+        lib().gpu_ivf_flat_extend_float(self._h, vecs.ctypes.data,
+                                        vecs.shape[0], idp, err.ref)
+        err.check("extend_float")
+
+    def delete_id(self, id_: int):
+        err = _Err()
+        lib().gpu_ivf_flat_delete_id(self._h, int(id_), err.ref)
+        err.check("delete_id")
+
+    def train_quantizer(self, train):
+        train = _as_f32(train)
+        err = _Err()
+        lib().gpu_ivf_flat_train_quantizer(self._h, train.ctypes.data,
+                                           train.shape[0], err.ref)
+        err.check("train_quantizer")
+
+    def info(self) -> dict:
+        import json as _json
+        err = _Err()
+        p = lib().gpu_ivf_flat_info(self._h, err.ref)
+        err.check("info")
+        if not p:
+            raise MoannError("info returned NULL")
+        try:
+            s = ct.string_at(p).decode()
+        finally:
+            ct.CDLL(None).free(ct.c_void_p(p))
+        return _json.loads(s)
+
+    def save_file(self, path: str):
+        err = _Err()
+        lib().gpu_ivf_flat_save(self._h, path.encode(), err.ref)
+        err.check("save")
+
+    def set_filter_columns(self, col_meta_json: str, total_count: int):
+        err = _Err()
+        lib().gpu_ivf_flat_set_filter_columns(
+            self._h, col_meta_json.encode(), total_count, err.ref)
+        err.check("set_filter_columns")
+
+    def add_filter_chunk(self, col_idx: int, data, null_bitmap=None):
+        data = np.ascontiguousarray(data)
+        nbp = None
+        if null_bitmap is not None:
+            null_bitmap = np.ascontiguousarray(null_bitmap, dtype=np.uint32)
+            nbp = null_bitmap.ctypes.data
+        err = _Err()
+        lib().gpu_ivf_flat_add_filter_chunk(
+            self._h, col_idx, data.ctypes.data, nbp, data.shape[0], err.ref)
+        err.check("add_filter_chunk")
+
+    def search_with_filter(self, queries, k: int, nprobe: int,
+                           preds_json: str):
+        queries = _as_f32(queries)
+        nq = queries.shape[0]
+        err = _Err()
+        res = lib().gpu_ivf_flat_search_quantize_with_filter(
+            self._h, queries.ctypes.data, nq, queries.shape[1], k,
+            SearchParams(nprobe),
+            preds_json.encode() if preds_json else None, err.ref)
+        err.check("search_quantize_with_filter")
+        return self._read_result(res, nq, k, "search_with_filter")
 
     def set_centroids(self, centroids):
         centroids = _as_f32(centroids)
@@ -365,13 +490,15 @@ class IvfFlatIndex:
         return ids, dists
 
     def search(self, queries, k: int, nprobe: int):
+        """f32 queries -> gpu_ivf_flat_search_quantize (base-typed; the
+        storage-typed gpu_ivf_flat_search takes T-typed queries)."""
         queries = _as_f32(queries)
         nq = queries.shape[0]
         err = _Err()
-        res = lib().gpu_ivf_flat_search(self._h, queries.ctypes.data, nq,
-                                        queries.shape[1], k,
-                                        SearchParams(nprobe), err.ref)
-        err.check("search")
+        res = lib().gpu_ivf_flat_search_quantize(
+            self._h, queries.ctypes.data, nq, queries.shape[1], k,
+            SearchParams(nprobe), err.ref)
+        err.check("search_quantize")
         return self._read_result(res, nq, k, "search")
 
     def set_quantizer(self, vmin: float, vmax: float):
@@ -527,10 +654,10 @@ class IvfFlatIndex:
         queries = _as_f32(queries)
         self._async_shape = (queries.shape[0], k)
         err = _Err()
-        job = lib().gpu_ivf_flat_search_async(
+        job = lib().gpu_ivf_flat_search_quantize_async(
             self._h, queries.ctypes.data, queries.shape[0], queries.shape[1],
             k, SearchParams(nprobe), err.ref)
-        err.check("search_async")
+        err.check("search_quantize_async")
         return job
 
     def search_wait(self, job: int):
