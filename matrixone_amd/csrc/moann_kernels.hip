@@ -516,7 +516,12 @@ void launch_topk(const float* dists, const int64_t* qoffs, int64_t uniform_n,
     if (nq == 0) return;
     int sortP = 1;
     while (sortP < k) sortP <<= 1;
-    if (sortP < 256) sortP = 256;
+    /* A larger sort buffer makes the group-collect exit fire at level 0
+     * (strict-below <= k plus a ~count/256 boundary group): 2 full-array
+     * reads instead of 3 for the big per-query candidate selections. The
+     * extra bitonic work (1024 u64 keys) is microseconds per block; LDS
+     * 2048+8*1024 = 10 KB still keeps several blocks per CU. */
+    if (sortP < 1024) sortP = 1024;
     const size_t shmem = 2048 + (size_t)sortP * 8;
     hipLaunchKernelGGL(topk_kernel, dim3(nq), dim3(256), shmem, stream,
                        dists, qoffs, uniform_n, k, sortP, out_slots, out_dists);
