@@ -392,14 +392,36 @@ def main():
                                   "bf16": 2}.get(args.qtype, 1)
     scan_gbs = (perf["scan_bytes"] / 4 * elem / 1e9) / max(
         perf["scan_ms"] / 1e3, 1e-9)
+    # PMC traffic: per-launch HBM bytes measured with rocprofv3 --pmc
+    # FETCH_SIZE against THIS workload+scan mode, committed under profiles/
+    # (collected by tools/collect_traffic.sh; x2 gfx950 counter correction
+    # per MI355X_MICROARCH.md). Emitted only when the keyed entry matches
+    # the exact configuration being run; otherwise null.
+    traffic = None
+    traffic_src = None
+    try:
+        tpath = os.path.join(REPO, "profiles", "r02_traffic.json")
+        with open(tpath) as f:
+            tdb = json.load(f)
+        key = (f"{args.rows}x{args.dim}-nlist{args.nlist}-"
+               f"nprobe{args.nprobe}-batch{args.batch}-"
+               f"{'refine' if args.refine else 'exact'}-{args.qtype}")
+        ent = tdb.get(key)
+        if ent and world == 1:
+            traffic = ent["bytes_per_launch"]
+            traffic_src = ent.get("provenance")
+    except Exception:
+        pass
     roofline = {
         "bound": "hbm",
         "achieved": round(scan_gbs, 1),
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
         "frac": round(scan_gbs / HBM_PEAK_GBS, 4),
-        "traffic": None,  # PMC bytes come from the committed rocprof runs
+        "traffic": traffic,
     }
+    if traffic_src:
+        roofline["traffic_provenance"] = traffic_src
 
     # ---- CPU baseline (rank 0, N=1 only) -----------------------------------
     cpu_baseline = None
