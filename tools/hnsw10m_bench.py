@@ -170,8 +170,12 @@ def main():
                      args.workdir, args.build_threads))
     t0 = time.perf_counter()
     ctx = mp.get_context("spawn")
+    res = []
     with ctx.Pool(nmodels) as pool:
-        res = pool.map(build_worker, jobs)
+        for r in pool.imap_unordered(build_worker, jobs):
+            res.append(r)
+            print(f"[h10] model {r[0]} built in {r[1]:.1f}s "
+                  f"({len(res)}/{nmodels})", flush=True)
     build_s = time.perf_counter() - t0
     per_model_cpu = {i: c for i, _, c in res}
     print(f"[h10] {nmodels} mini-model builds {build_s:.1f}s "
