@@ -41,7 +41,7 @@ def main():
     ap.add_argument("--k", type=int, default=10)
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--gt-queries", type=int, default=256)
-    ap.add_argument("--chunk", type=int, default=1 << 21)
+    ap.add_argument("--chunk", type=int, default=1 << 20)
     args = ap.parse_args()
     dev = torch.device("cuda:0")
     dsub = args.dim // args.nsub
@@ -134,11 +134,16 @@ def main():
         hi = min(lo + args.chunk, rows)
         blk = gen_chunk(lo, hi)
         assign_blk(blk, assign[lo:hi])
-        r = (blk - cent[assign[lo:hi].long()]) \
-            .view(hi - lo, args.nsub, dsub).permute(1, 0, 2).contiguous()
-        d2 = -2.0 * torch.bmm(r, cb.transpose(1, 2)) + cb2
-        codes[lo:hi] = d2.argmin(-1).permute(1, 0).to(torch.uint8)
-        del r, d2
+        # encode in sub-chunks: the [nsub, m, 256] bmm distance tensor is
+        # nsub*m*256*4 B (192 GB at m=2M) — keep m <= 256k (~25 GB)
+        EC = 1 << 18
+        for e0 in range(lo, hi, EC):
+            e1 = min(e0 + EC, hi)
+            r = (blk[e0 - lo:e1 - lo] - cent[assign[e0:e1].long()]) \
+                .view(e1 - e0, args.nsub, dsub).permute(1, 0, 2).contiguous()
+            d2 = -2.0 * torch.bmm(r, cb.transpose(1, 2)) + cb2
+            codes[e0:e1] = d2.argmin(-1).permute(1, 0).to(torch.uint8)
+            del r, d2
         # GT update (f32 GEMM: exact enough for a recall reference)
         ipq = gtq @ blk.T
         dq = qn2 - 2.0 * ipq + (blk * blk).sum(1)[None, :]
