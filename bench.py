@@ -180,7 +180,7 @@ def main():
     ap.add_argument("--recall-queries", type=int, default=256)
     ap.add_argument("--cpu-baseline-queries", type=int, default=48)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--refine", type=int, default=128,
+    ap.add_argument("--refine", type=int, default=64,
                     help="two-stage scan: byte first pass + exact f32 "
                          "re-rank of the top-R (0 = exact one-pass scan). "
                          "GPU-validated at 10M post the QT-8 job-geometry "

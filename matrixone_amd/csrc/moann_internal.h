@@ -103,7 +103,7 @@ void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
                     const uint8_t* queries_q, const int32_t* qnorms,
                     const int32_t* rownorms, int dpad,
                     const ScanJobs& jb, float* dists_out, hipStream_t stream,
-                    const uint32_t* filter_bitset = nullptr);
+                    const uint32_t* filter_bitset = nullptr, int qt = 8);
 /* per-(group,lane) sum-of-squares over the packed byte rows (build-time,
  * feeds the dot-form scan's rn + qn - 2*dot L2) */
 void launch_bytes_pack16(const uint8_t* rows_q, int dpad,

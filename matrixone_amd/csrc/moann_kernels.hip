@@ -1338,7 +1338,7 @@ __global__ __launch_bounds__(256) void scan_i8_kernel(
  * from launch_qnorms_i8. Results are bit-identical to scan_i8_kernel
  * (exact integer arithmetic both ways); that kernel stays the path for
  * dpad % 16 != 0. */
-template <int METRIC, bool UNSIGNED>
+template <int METRIC, bool UNSIGNED, int QT = 8>
 __global__ __launch_bounds__(256) void scan_i8_dot_kernel(
     const uint8_t* __restrict__ packed, const uint8_t* __restrict__ queries_q,
     const int32_t* __restrict__ qnorms_i, /* [nq] sum sq (l2/cos) */
@@ -1353,7 +1353,6 @@ __global__ __launch_bounds__(256) void scan_i8_dot_kernel(
     const int64_t* __restrict__ j_slot_base,
     const uint32_t* __restrict__ filter_bitset,
     float* __restrict__ dists_out) {
-    constexpr int QT = 8;
     constexpr bool NEED_N = METRIC == KM_L2SQ || METRIC == KM_COS;
     extern __shared__ float ldsf[];
     uint8_t* ldsq = (uint8_t*)ldsf;            /* [QT][dpad] bytes */
@@ -1601,11 +1600,12 @@ static void launch_scan_i8_m(bool uns, const uint8_t* packed,
                              const int32_t* rownorms, int dpad,
                              const ScanJobs& jb, float* dists_out,
                              hipStream_t stream,
-                             const uint32_t* filter_bitset) {
+                             const uint32_t* filter_bitset, int qt) {
     const dim3 grid(jb.njobs), block(256);
-    const size_t shmem = 8 * dpad + 8 * 4 + 16;
+    const size_t shmem = (size_t)qt * dpad + (size_t)qt * 4 + 16;
     /* dot-form kernel needs 16-dim-aligned rows and the precomputed row
-     * norms; falls back to the uchar4 kernel otherwise (odd dims) */
+     * norms; falls back to the uchar4 kernel otherwise (odd dims).
+     * qt MUST match the job builder's tile width (the r01 refine bug). */
     const bool dot = (dpad & 15) == 0 && rownorms != nullptr;
 #define I8ARGS packed, queries_q, qnorms, dpad, jb.databaseg, jb.gstart, \
     jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, jb.qslot_outbase, \
@@ -1613,13 +1613,26 @@ static void launch_scan_i8_m(bool uns, const uint8_t* packed,
 #define I8DARGS packed, queries_q, qnorms, rownorms, dpad, jb.databaseg, \
     jb.gstart, jb.gcount, jb.rows, jb.nq, jb.qbase, jb.qslot_query, \
     jb.qslot_outbase, jb.slot_base, filter_bitset, dists_out
-    if (dot) {
+    if (dot && qt == 16) {
         if (uns)
-            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, true>), grid,
+            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, true, 16>), grid,
                                block, shmem, stream, I8DARGS);
         else
-            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, false>), grid,
+            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, false, 16>), grid,
                                block, shmem, stream, I8DARGS);
+    } else if (dot && qt == 8) {
+        if (uns)
+            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, true, 8>), grid,
+                               block, shmem, stream, I8DARGS);
+        else
+            hipLaunchKernelGGL((scan_i8_dot_kernel<METRIC, false, 8>), grid,
+                               block, shmem, stream, I8DARGS);
+    } else if (qt != 8) {
+        /* no QT!=8 instantiation of the uchar4 kernel: loud geometry error
+         * instead of silently unscanned tile lanes */
+        (void)hipErrorInvalidValue;
+        hipLaunchKernelGGL((scan_i8_kernel<METRIC, true>), dim3(0), block,
+                           shmem, stream, I8ARGS); /* grid 0 -> launch err */
     } else if (uns)
         hipLaunchKernelGGL((scan_i8_kernel<METRIC, true>), grid, block, shmem,
                            stream, I8ARGS);
@@ -1634,13 +1647,13 @@ void launch_scan_i8(int metric, bool uns, const uint8_t* packed,
                     const uint8_t* queries_q, const int32_t* qnorms,
                     const int32_t* rownorms, int dpad,
                     const ScanJobs& jb, float* dists_out, hipStream_t stream,
-                    const uint32_t* filter_bitset) {
+                    const uint32_t* filter_bitset, int qt) {
     if (!jb.njobs) return;
     switch (metric) {
-    case KM_L2SQ: launch_scan_i8_m<KM_L2SQ>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
-    case KM_IP:   launch_scan_i8_m<KM_IP>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
-    case KM_COS:  launch_scan_i8_m<KM_COS>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
-    default:      launch_scan_i8_m<KM_L1>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset); break;
+    case KM_L2SQ: launch_scan_i8_m<KM_L2SQ>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset, qt); break;
+    case KM_IP:   launch_scan_i8_m<KM_IP>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset, qt); break;
+    case KM_COS:  launch_scan_i8_m<KM_COS>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset, qt); break;
+    default:      launch_scan_i8_m<KM_L1>(uns, packed, queries_q, qnorms, rownorms, dpad, jb, dists_out, stream, filter_bitset, qt); break;
     }
 }
 

@@ -514,8 +514,26 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     const bool refine_scan = ix->qtype == Quantization_F32 &&
                              ix->refine_depth > 0 &&
                              ix->d_packed_rq != nullptr;
-    const int qt = pick_qt(ix->kmetric, ix->dpad,
-                           ix->qtype != Quantization_F32 || refine_scan);
+    /* The byte DOT kernel also has a QT=16 instantiation — use the wide
+     * tile whenever the byte path will take the dot form (halves the
+     * query-tile re-read multiplicity, measured 1.14x compulsory at QT=8);
+     * the uchar4 fallback and the half kernels stay QT=8. */
+    const bool half_st = ix->qtype == Quantization_F16 ||
+                         ix->qtype == Quantization_BF16;
+    bool byte_dot16 = false;
+    if (refine_scan)
+        byte_dot16 = (ix->dpad & 15) == 0 && ix->d_rownorm_rq &&
+                     (size_t)16 * ix->dpad + 80 <= 48 * 1024;
+    else if (ix->qtype != Quantization_F32 && !half_st)
+        byte_dot16 = (ix->dpad & 15) == 0 && ix->d_rownorm_q &&
+                     (size_t)16 * ix->dpad + 80 <= 48 * 1024;
+    if (const char* e = getenv("MOANN_BYTE_QT"))
+        if (atoi(e) == 8) byte_dot16 = false; /* A/B override */
+    const int qt = byte_dot16
+                       ? 16
+                       : pick_qt(ix->kmetric, ix->dpad,
+                                 ix->qtype != Quantization_F32 ||
+                                     refine_scan);
     if (use_mfma_rank) {
         cx->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         HIP_CHECK(hipEventRecord(cx->ev[0], s));
@@ -767,7 +785,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
             launch_scan_i8(ix->kmetric, false, ix->d_packed_rq,
                            cx->w_queries_q.as<uint8_t>(), d_qni,
                            ix->d_rownorm_rq, dpad, jb,
-                           cx->w_cand.as<float>(), s, d_filter);
+                           cx->w_cand.as<float>(), s, d_filter, qt);
         } else if (half) {
             /* plain-cast narrow search: query cast RTNE like the entries
              * (relation_search.go:56-63), decoded-f32 distance */
@@ -809,7 +827,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
             launch_scan_i8(ix->kmetric, uns, ix->d_packed_q,
                            cx->w_queries_q.as<uint8_t>(), d_qni,
                            ix->d_rownorm_q, dpad, jb,
-                           cx->w_cand.as<float>(), s, d_filter);
+                           cx->w_cand.as<float>(), s, d_filter, qt);
         } else {
             launch_scan(ix->kmetric, qt, ix->d_packed, d_q, d_qn, dpad, jb,
                         cx->w_cand.as<float>(), s, d_filter);
