@@ -185,7 +185,7 @@ def main():
                          "re-rank of the top-R (0 = exact one-pass scan). "
                          "GPU-validated at 10M post the QT-8 job-geometry "
                          "fix: R=64/128/256 all recall 1.0000 "
-                         "(profiles/r2_refine_sweep). "
+                         "(profiles/r02_refine_sweep.log; R=64 best at steps 20). "
                          "ADAPTIVE: if the measured recall@k gate (0.95) "
                          "fails with refine on, the bench falls back to "
                          "the exact one-pass scan and re-measures — the "
