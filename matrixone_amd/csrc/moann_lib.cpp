@@ -529,11 +529,13 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
                      (size_t)16 * ix->dpad + 80 <= 48 * 1024;
     if (const char* e = getenv("MOANN_BYTE_QT"))
         if (atoi(e) == 8) byte_dot16 = false; /* A/B override */
-    const int qt = byte_dot16
-                       ? 16
-                       : pick_qt(ix->kmetric, ix->dpad,
-                                 ix->qtype != Quantization_F32 ||
-                                     refine_scan);
+    /* rank (f32 centroid scan) keeps its own tile width: only the flagship
+     * (L2SQ, 768) has a vetted QT=16 f32 path (the asm kernel); the byte
+     * list scan may still go wide independently */
+    const int qt_rank = pick_qt(ix->kmetric, ix->dpad,
+                                ix->qtype != Quantization_F32 ||
+                                    refine_scan);
+    const int qt = byte_dot16 ? 16 : qt_rank;
     if (use_mfma_rank) {
         cx->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         HIP_CHECK(hipEventRecord(cx->ev[0], s));
@@ -551,6 +553,7 @@ std::unique_ptr<Result> run_search(IvfIndex* ix, const float* queries,
     } else {
         cx->w_cdists.ensure((size_t)nq * ix->nlist * 4);
         /* tile queries, chunk centroid groups to fill the chip */
+        const int qt = qt_rank; /* shadow: the rank stage's tile width */
         const int ntiles = (int)((nq + qt - 1) / qt);
         const int64_t cgroups = ix->cent_groups;
         /* target ~1024 jobs: enough to fill 256 CUs without paying the
