@@ -283,10 +283,19 @@ struct IvfIndex {
 };
 
 int pick_qt(int kmetric, uint32_t dpad, bool narrow) {
-    /* MOANN_QT overrides for A/B tuning */
+    /* MOANN_QT overrides for A/B tuning. QT=16 is honored only for the
+     * vetted flagship shape (L2SQ, dpad 768 — the asm kernel, or
+     * MOANN_SCAN=generic measured there in r01): the generic f32 QT16
+     * instantiation miscomputed at small dpad in the r02 byte-QT16
+     * triage (7 parity failures when the rank stage ran it at d=64/96),
+     * so other shapes clamp to 8 instead of silently going wrong. */
     if (const char* e = getenv("MOANN_QT")) {
         const int v = atoi(e);
-        if (v == 1 || v == 2 || v == 4 || v == 8 || v == 12 || v == 16) return v;
+        if (v == 16) {
+            if (kmetric == KM_L2SQ && dpad == 768) return 16;
+        } else if (v == 1 || v == 2 || v == 4 || v == 8 || v == 12) {
+            return v;
+        }
     }
     /* flagship f32 shape: the hand-scheduled QT=16 asm kernel
      * (scan_asm768.hip) — lowest tile re-read multiplicity AND a pipelined
