@@ -241,7 +241,8 @@ struct IvfIndex {
      * result-identical to the reference's in-list insertion
      * (cuvs::ivf_flat::extend): a row participates exactly when its
      * assigned list is probed. */
-    std::unique_ptr<IvfIndex> tail;
+    std::shared_ptr<IvfIndex> tail; /* shared: searches snapshot it under
+        mu so a concurrent extend's rebuild cannot free it mid-search */
     std::vector<float> tail_staging;   /* f32 rows, extend order */
     std::vector<int64_t> tail_ids;
     std::vector<int32_t> tail_assign;
@@ -1040,10 +1041,19 @@ std::vector<uint32_t> combined_filter(IvfIndex* ix, uint64_t slot0,
     const uint64_t n = ix->count;
     std::vector<uint32_t> out((size_t)((n + 31) / 32), ~0u);
     if (user) {
-        for (uint64_t i = 0; i < n; ++i) {
-            const uint64_t g = slot0 + i;
-            const bool bit = g < nbits && ((user[g / 32] >> (g % 32)) & 1u);
-            if (!bit) out[i / 32] &= ~(1u << (i % 32));
+        if (slot0 % 32 == 0 && slot0 + n <= nbits) {
+            /* word-aligned range (the main index at slot0 = 0): straight
+             * word copy — no per-bit work on the 10M-slot hot path */
+            memcpy(out.data(), user + slot0 / 32, out.size() * 4);
+            if (n % 32) /* mask stray bits beyond count in the last word */
+                out.back() |= ~((1u << (n % 32)) - 1u);
+        } else {
+            for (uint64_t i = 0; i < n; ++i) {
+                const uint64_t g = slot0 + i;
+                const bool bit =
+                    g < nbits && ((user[g / 32] >> (g % 32)) & 1u);
+                if (!bit) out[i / 32] &= ~(1u << (i % 32));
+            }
         }
     }
     if (ix->deleted_count) {
@@ -1064,22 +1074,35 @@ std::unique_ptr<Result> run_search_full(IvfIndex* ix, const float* queries,
                                         const uint32_t* user_filter = nullptr,
                                         uint64_t user_nbits = 0,
                                         bool sync_end = true) {
-    const bool has_tail = ix->tail && ix->tail->count > 0;
-    const bool need_main_filter = user_filter || ix->deleted_count > 0;
-    std::vector<uint32_t> mf;
-    if (need_main_filter)
-        mf = combined_filter(ix, 0, user_filter, user_nbits);
+    /* snapshot mutable host state under the index mutex: ix->tail and the
+     * alive bitsets are written by extend/delete_id (which hold mu); the
+     * shared_ptr copy keeps a replaced tail alive through this search */
+    std::shared_ptr<IvfIndex> tl;
+    std::vector<uint32_t> mf, tf;
+    bool need_main_filter = false, need_tf = false;
+    {
+        std::lock_guard<std::mutex> lk(ix->mu);
+        tl = ix->tail;
+        need_main_filter = user_filter || ix->deleted_count > 0;
+        if (need_main_filter)
+            mf = combined_filter(ix, 0, user_filter, user_nbits);
+        if (tl && tl->count > 0) {
+            need_tf = user_filter || tl->deleted_count > 0;
+            if (need_tf)
+                tf = combined_filter(tl.get(), ix->count, user_filter,
+                                     user_nbits);
+        } else {
+            tl.reset();
+        }
+    }
 
     /* tail first (synchronous, small), so the submit path can return after
      * enqueueing the main scan */
     std::unique_ptr<Result> tres;
-    if (has_tail) {
-        IvfIndex* t = ix->tail.get();
-        std::vector<uint32_t> tf;
-        const bool need_tf = user_filter || t->deleted_count > 0;
-        if (need_tf) tf = combined_filter(t, ix->count, user_filter, user_nbits);
-        tres = run_search(t, queries, queries_on_device, nq, limit, n_probes,
-                          need_tf ? tf.data() : nullptr, /*sync_end=*/true);
+    if (tl) {
+        tres = run_search(tl.get(), queries, queries_on_device, nq, limit,
+                          n_probes, need_tf ? tf.data() : nullptr,
+                          /*sync_end=*/true);
     }
     auto res = run_search(ix, queries, queries_on_device, nq, limit, n_probes,
                           need_main_filter ? mf.data() : nullptr, sync_end);

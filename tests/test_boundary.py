@@ -201,3 +201,44 @@ def test_info_and_dynb_knob():
         ix._h, True, err.ref)
     err.check("dynb")
     ix.close()
+
+
+def test_traffic_and_mfma_parsers_pin_committed_evidence():
+    """The committed PMC evidence is reproducible from the committed CSVs
+    by the committed parsers (provenance chain for roofline.traffic and
+    the MFMA ratio — CPU-only)."""
+    import json
+    import subprocess
+    import sys as _sys
+    import tempfile
+    repo = REPO
+    csvp = os.path.join(repo, "profiles", "r02_fetch_refine_qt16.csv")
+    if not os.path.exists(csvp):
+        pytest.skip("evidence csv not present")
+    with tempfile.TemporaryDirectory() as td:
+        outp = os.path.join(td, "t.json")
+        r = subprocess.run(
+            [_sys.executable, os.path.join(repo, "tools", "collect_traffic.py"),
+             csvp, "--key", "k", "--kernel", "scan_i8_dot", "--out", outp],
+            capture_output=True, text=True, timeout=120)
+        assert r.returncode == 0, r.stderr
+        got = json.load(open(outp))["k"]["bytes_per_launch"]
+    want = json.load(open(os.path.join(repo, "profiles", "r02_traffic.json")))
+    want = want["10000000x768-nlist4096-nprobe32-batch1024-refine-f32"][
+        "bytes_per_launch"]
+    assert got == want, (got, want)
+
+    busy = os.path.join(repo, "profiles", "r02_mfma_busy.csv")
+    wave = os.path.join(repo, "profiles", "r02_mfma_wave.csv")
+    if os.path.exists(busy) and os.path.exists(wave):
+        import json as _json
+        r = subprocess.run(
+            [_sys.executable, os.path.join(repo, "tools", "mfma_util.py"),
+             busy, wave, "rank_gemm"], capture_output=True, text=True,
+            timeout=120)
+        assert r.returncode == 0, r.stderr
+        got = _json.loads(r.stdout)
+        want = _json.load(open(os.path.join(repo, "profiles",
+                                            "r02_mfma_util.json")))
+        assert got["mfma_busy_over_wave_cycles"] == \
+            want["mfma_busy_over_wave_cycles"]
