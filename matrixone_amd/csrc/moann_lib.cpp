@@ -2189,6 +2189,12 @@ void gpu_ivf_flat_train_quantizer(gpu_ivf_flat_c h, const float* train_data,
         size_t stride = 1;
         const size_t cap = 1u << 26;
         if (n > cap) stride = (n + cap - 1) / cap;
+        /* avoid column-aliased sampling of the row-major buffer: keep the
+         * stride coprime with small dim factors (dims are 2^a*3^b-ish) */
+        if (stride > 1) {
+            stride |= 1;
+            if (stride % 3 == 0) stride += 2;
+        }
         std::vector<float> samp;
         samp.reserve(n / stride + 1);
         for (size_t i = 0; i < n; i += stride) samp.push_back(train_data[i]);

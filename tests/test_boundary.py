@@ -242,3 +242,27 @@ def test_traffic_and_mfma_parsers_pin_committed_evidence():
                                             "r02_mfma_util.json")))
         assert got["mfma_busy_over_wave_cycles"] == \
             want["mfma_busy_over_wave_cycles"]
+
+
+def test_load_file_rejects_garbage_no_gpu(tmp_path):
+    """gpu_ivf_flat_load_file fails loudly on missing/corrupt containers
+    BEFORE touching the GPU (magic check precedes device work)."""
+    from matrixone_amd import engine
+    L = ct.CDLL(_built())
+    L.gpu_ivf_flat_load_file.restype = ct.c_void_p
+    bp = engine.BuildParams(4, True, 0.5)
+    dev = (ct.c_int * 1)(0)
+    err = engine._Err()
+    h = L.gpu_ivf_flat_load_file(str(tmp_path / "missing").encode(), 8,
+                                 0, bp, dev, 1, 0, 0, 0, 0, err.ref)
+    assert not h
+    with pytest.raises(engine.MoannError):
+        err.check("load_file")
+    bad = tmp_path / "bad.moann"
+    bad.write_bytes(b"NOTAPACK" + b"\x00" * 64)
+    err2 = engine._Err()
+    h2 = L.gpu_ivf_flat_load_file(str(bad).encode(), 8, 0, bp, dev, 1, 0,
+                                  0, 0, 0, err2.ref)
+    assert not h2
+    with pytest.raises(engine.MoannError):
+        err2.check("load_file")
