@@ -266,3 +266,39 @@ def test_load_file_rejects_garbage_no_gpu(tmp_path):
     assert not h2
     with pytest.raises(engine.MoannError):
         err2.check("load_file")
+
+
+def test_load_file_fuzz_no_crash(tmp_path):
+    """Randomized corrupt pack containers: every parse must fail with an
+    error (NULL handle + errmsg), never crash or succeed (CPU-only: the
+    parse precedes device work)."""
+    import numpy as np
+    from matrixone_amd import engine
+    L = ct.CDLL(_built())
+    L.gpu_ivf_flat_load_file.restype = ct.c_void_p
+    bp = engine.BuildParams(4, True, 0.5)
+    dev = (ct.c_int * 1)(0)
+    rng = np.random.Generator(np.random.PCG64(5150))
+    magic = b"MOANNPK1\n"
+    for i in range(24):
+        kind = i % 4
+        if kind == 0:  # pure noise
+            blob = rng.integers(0, 256, rng.integers(0, 256),
+                                dtype=np.uint8).tobytes()
+        elif kind == 1:  # right magic, noise body
+            blob = magic + rng.integers(0, 256, rng.integers(0, 128),
+                                        dtype=np.uint8).tobytes()
+        elif kind == 2:  # magic + plausible header, truncated payload
+            name = b"manifest.json"
+            blob = (magic + len(name).to_bytes(4, "little") + name +
+                    (1 << 30).to_bytes(8, "little") + b"x" * 16)
+        else:  # magic + oversized name length
+            blob = magic + (4096).to_bytes(4, "little") + b"y" * 8
+        f = tmp_path / f"fz{i}.moann"
+        f.write_bytes(blob)
+        err = engine._Err()
+        h = L.gpu_ivf_flat_load_file(str(f).encode(), 8, 0, bp, dev, 1, 0,
+                                     0, 0, 0, err.ref)
+        assert not h, f"fuzz case {i} unexpectedly parsed"
+        with pytest.raises(engine.MoannError):
+            err.check("fuzz")
