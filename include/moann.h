@@ -300,7 +300,7 @@ gpu_ivf_flat_search_res_t moann_ivf_flat_search_device(gpu_ivf_flat_c index_c,
                                                        ivf_flat_search_params_t search_params,
                                                        void* errmsg);
 
-/* ivf_flat_c.h:128-135: async pair.  Round 1 executes on the caller thread at
+/* ivf_flat_c.h:128-135: async pair.  Executes on the caller thread at
  * submit and parks the result for search_wait (the reference's per-device
  * worker pool collapses to one device per process here). */
 uint64_t gpu_ivf_flat_search_async(gpu_ivf_flat_c index_c,
